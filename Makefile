@@ -1,0 +1,46 @@
+# binder-amd: native-first rebuild of TritonDataCenter/binder.
+# Everything is plain C++20 + POSIX; no external libraries.
+
+CXX ?= g++
+PYTHON ?= python3
+CXXFLAGS ?= -O2 -g -std=c++20 -fPIC -fno-omit-frame-pointer \
+	-Wall -Wextra -Werror -Wno-unused-parameter
+LDFLAGS ?=
+
+BUILD := build
+
+CORE_SRCS := \
+	native/common/json.cpp \
+	native/common/log.cpp \
+	native/common/loop.cpp \
+	native/dns/codec.cpp \
+	native/engine/store.cpp \
+	native/engine/engine.cpp
+
+CORE_OBJS := $(CORE_SRCS:%.cpp=$(BUILD)/%.o)
+
+PY_EXT_SUFFIX := $(shell $(PYTHON)-config --extension-suffix 2>/dev/null || echo .so)
+PY_INCLUDES := $(shell $(PYTHON) -m pybind11 --includes)
+PYMOD := binder_amd/_native$(PY_EXT_SUFFIX)
+
+BINARIES :=
+
+all: $(PYMOD) $(BINARIES)
+
+$(BUILD)/%.o: %.cpp
+	@mkdir -p $(dir $@)
+	$(CXX) $(CXXFLAGS) -c $< -o $@
+
+# pybind11 module (compiled separately: needs Python includes, and
+# -Wno-error for pybind's warnings under -Wextra)
+$(BUILD)/native/pybind/module.o: native/pybind/module.cpp
+	@mkdir -p $(dir $@)
+	$(CXX) $(CXXFLAGS) -Wno-error $(PY_INCLUDES) -c $< -o $@
+
+$(PYMOD): $(CORE_OBJS) $(BUILD)/native/pybind/module.o
+	$(CXX) -shared $(CXXFLAGS) $^ -o $@ $(LDFLAGS)
+
+clean:
+	rm -rf $(BUILD) $(PYMOD) $(BINARIES)
+
+.PHONY: all clean

@@ -1,0 +1,142 @@
+/*
+ * binder-amd: record store interface + record compilation.
+ *
+ * The reference keeps raw JSON-parsed objects in its ZK mirror and walks
+ * them on every query (/root/reference/lib/server.js:249-424). We instead
+ * compile each znode's JSON payload into a flat CompiledRecord ONCE at
+ * update time (mirror watch delivery), so the query hot path touches only
+ * PODs and interned strings. Two Store implementations exist:
+ *   - StubStore: in-memory, fed directly (tests + BASELINE config 1);
+ *   - ZkMirror: watch-driven full mirror of a ZooKeeper tree (native/zk/).
+ */
+#pragma once
+
+#include <cstdint>
+#include <memory>
+#include <optional>
+#include <string>
+#include <unordered_map>
+#include <vector>
+
+#include "../common/json.hpp"
+
+namespace bamd {
+
+/*
+ * Registration record types understood by binder
+ * (lib/server.js:302-308, 352-360; lib/zk.js:172-189).
+ */
+enum class RecType : uint8_t {
+    Unknown = 0,
+    Host,
+    DbHost,
+    LoadBalancer,
+    MorayHost,
+    RedisHost,
+    OpsHost,
+    RrHost,
+    Database,
+    Service,
+};
+
+bool recTypeIsHostLike(RecType t);       // the 7 *_host/load_balancer types
+bool recTypeServesUnderService(RecType t);  // server.js:352-360 filter set
+RecType recTypeFromString(const std::string& s);
+const char* recTypeName(RecType t);
+
+struct CompiledRecord {
+    bool hasData = false;   // znode payload parsed to a JSON object
+    bool valid = false;     // .type is a string AND [.type] is an object
+    RecType type = RecType::Unknown;
+    std::string typeName;
+
+    /* host-like + database (URL-parsed hostname): */
+    std::string address;    // empty = null/absent address
+    /* TTL via the precedence chain: 30 -> record.ttl -> record[type].ttl
+     * -> (service, nested) record.service.service.ttl
+     * (lib/server.js:262-274, 324-332). */
+    uint32_t ttl = 30;
+    /* krec.ttl / krec[type].ttl override when this node is a service
+     * member (server.js:389-393). */
+    std::optional<uint32_t> memberTtlOverride;
+
+    /* service-only (after the optional service.service dive): */
+    std::string srvce;      // e.g. "_http"
+    std::string proto;      // e.g. "_tcp"
+    uint16_t defaultPort = 0;
+    bool hasDefaultPort = false;
+
+    /* member-only: per-member port list (server.js:383-385) */
+    std::vector<uint16_t> ports;
+};
+
+/* Compile a znode JSON payload. `parsed`=false when payload was not valid
+ * JSON (such nodes are kept but data is ignored; lib/zk.js:139-154). */
+CompiledRecord compileRecord(const Json& data);
+
+class StoreNode {
+  public:
+    virtual ~StoreNode() = default;
+    virtual const CompiledRecord& rec() const = 0;
+    virtual const std::string& domain() const = 0;   // fqdn, lowercase
+    virtual const std::string& name() const = 0;     // leftmost label
+    virtual std::vector<const StoreNode*> children() const = 0;
+};
+
+class Store {
+  public:
+    virtual ~Store() = default;
+    virtual const StoreNode* lookup(const std::string& domain) const = 0;
+    virtual const StoreNode* reverseLookup(const std::string& ip) const = 0;
+    virtual bool ready() const = 0;
+};
+
+/*
+ * StubStore: hash-map store fed directly with (domain -> JSON payload)
+ * pairs; maintains the same reverse IP index the mirror does.
+ */
+class StubStore : public Store {
+  public:
+    class Node : public StoreNode {
+      public:
+        Node(StubStore* store, std::string domain);
+        const CompiledRecord& rec() const override { return rec_; }
+        const std::string& domain() const override { return domain_; }
+        const std::string& name() const override { return name_; }
+        std::vector<const StoreNode*> children() const override;
+
+        StubStore* store_;
+        std::string domain_;
+        std::string name_;
+        CompiledRecord rec_;
+        std::vector<std::string> childDomains_;
+    };
+
+    /* Set/replace a node's payload; creates intermediate parents. */
+    void put(const std::string& domain, const Json& data);
+    void remove(const std::string& domain);
+    void setReady(bool r) { ready_ = r; }
+
+    const StoreNode* lookup(const std::string& domain) const override;
+    const StoreNode* reverseLookup(const std::string& ip) const override;
+    bool ready() const override { return ready_; }
+
+  private:
+    friend class Node;
+    Node* ensure(const std::string& domain);
+    void reindex(Node* n, const std::string& oldAddr);
+
+    std::unordered_map<std::string, std::unique_ptr<Node>> nodes_;
+    std::unordered_map<std::string, Node*> rev_;
+    bool ready_ = true;
+};
+
+/* "foo.com" -> "/com/foo" and back (lib/zk.js:225-228). */
+std::string domainToPath(const std::string& domain);
+std::string pathToDomain(const std::string& path);
+
+/* Extract hostname from a URL like "tcp://user@host:123/db"
+ * (lib/server.js:296-300 uses node's url.parse().hostname). */
+std::string urlHostname(const std::string& url);
+
+}  // namespace bamd
