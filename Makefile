@@ -17,15 +17,27 @@ CORE_SRCS := \
 	native/engine/store.cpp \
 	native/engine/engine.cpp
 
+SERVER_SRCS := \
+	native/server/metrics.cpp \
+	native/server/server.cpp \
+	native/server/recursion.cpp \
+	native/zk/client.cpp \
+	native/zk/mirror.cpp
+
 CORE_OBJS := $(CORE_SRCS:%.cpp=$(BUILD)/%.o)
+SERVER_OBJS := $(SERVER_SRCS:%.cpp=$(BUILD)/%.o)
 
 PY_EXT_SUFFIX := $(shell $(PYTHON)-config --extension-suffix 2>/dev/null || echo .so)
 PY_INCLUDES := $(shell $(PYTHON) -m pybind11 --includes)
 PYMOD := binder_amd/_native$(PY_EXT_SUFFIX)
 
-BINARIES :=
+BINARIES := bin/binderd
 
 all: $(PYMOD) $(BINARIES)
+
+bin/binderd: $(CORE_OBJS) $(SERVER_OBJS) $(BUILD)/native/server/binderd_main.o
+	@mkdir -p bin
+	$(CXX) $(CXXFLAGS) $^ -o $@ $(LDFLAGS)
 
 $(BUILD)/%.o: %.cpp
 	@mkdir -p $(dir $@)

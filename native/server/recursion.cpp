@@ -1,0 +1,342 @@
+#include "recursion.hpp"
+
+#include <arpa/inet.h>
+#include <ifaddrs.h>
+#include <netinet/in.h>
+#include <sys/epoll.h>
+#include <sys/socket.h>
+#include <unistd.h>
+
+#include <cstring>
+
+namespace bamd {
+
+using namespace dns;
+
+static constexpr int64_t kRefreshIntervalMs = 5 * 60 * 1000;
+static constexpr int64_t kRetryInitMs = 15 * 1000;
+static constexpr int64_t kUpstreamTimeoutMs = 3000;
+
+Recursion::Recursion(EventLoop* loop, Logger log, RecursionOptions opts,
+                     const Store* store)
+    : loop_(loop),
+      log_(log.child({{"component", Json("Recursion")}})),
+      opts_(std::move(opts)), store_(store) {}
+
+Recursion::~Recursion() {
+    if (fd_ >= 0) {
+        loop_->delFd(fd_);
+        close(fd_);
+    }
+    if (refreshTimer_) loop_->cancelTimer(refreshTimer_);
+}
+
+bool Recursion::openSocket() {
+    if (fd_ >= 0) return true;
+    fd_ = socket(AF_INET, SOCK_DGRAM | SOCK_NONBLOCK | SOCK_CLOEXEC, 0);
+    if (fd_ < 0) return false;
+    loop_->addFd(fd_, EPOLLIN, [this](uint32_t) { onSockReadable(); });
+    return true;
+}
+
+void Recursion::emitReady() {
+    if (ready_) return;
+    ready_ = true;
+    if (readyCb_) readyCb_();
+}
+
+void Recursion::init() {
+    log_.info("Initing Clients...");
+    openSocket();
+    refresh();
+}
+
+void Recursion::scheduleRefresh(int64_t ms) {
+    if (refreshTimer_) loop_->cancelTimer(refreshTimer_);
+    refreshTimer_ = loop_->addTimer(ms, [this]() {
+        refreshTimer_ = 0;
+        refresh();
+    });
+}
+
+void Recursion::refresh() {
+    const Json& cfg = opts_.config;
+    std::string source = cfg.get("source").asString();
+    if (source.empty())
+        source = cfg.get("dcs").isObject() ? "static" : "zk";
+
+    if (source == "static") {
+        std::map<std::string, std::vector<std::string>> dcs;
+        for (const auto& [dc, ips] : cfg.get("dcs").fields()) {
+            for (const auto& ip : ips.items())
+                if (ip.isString())
+                    dcs[dc].push_back(ip.asString());
+        }
+        dcs_ = std::move(dcs);
+        log_.info({{"dcs", Json((int64_t)dcs_.size())}},
+                  "setting recursion resolvers (static)");
+        emitReady();
+        scheduleRefresh(kRefreshIntervalMs);
+        return;
+    }
+
+    if (source == "zk") {
+        /* Bootstrap from the mirror itself: the resolver registry lives
+         * at a domain inside our own tree (pluggable replacement for the
+         * reference's UFDS-over-LDAP listResolvers; recursion.js:104-127
+         * similarly bootstraps UFDS's address from the ZK cache). */
+        std::string domain = cfg.get("registryDomain").asString();
+        if (domain.empty() || !store_->ready()) {
+            log_.warn("unable to refresh resolvers yet; will try again "
+                      "in 15 seconds (best effort)");
+            emitReady();  // 'ready' fires regardless (recursion.js:174-197)
+            scheduleRefresh(kRetryInitMs);
+            return;
+        }
+        const StoreNode* node = store_->lookup(domain);
+        if (node == nullptr) {
+            log_.warn({{"domain", Json(domain)}},
+                      "resolver registry domain not found yet");
+            emitReady();
+            scheduleRefresh(kRetryInitMs);
+            return;
+        }
+        /* registry node children: each child's record is a host-like
+         * entry whose name is "<dc>-<n>" or whose record carries a
+         * "datacenter" — we accept {"type":"resolver","resolver":
+         * {"datacenter":...,"address":...}} via raw compile too. For the
+         * common case the registry node's own JSON payload is the map,
+         * so prefer children of type host with dc prefix. */
+        std::map<std::string, std::vector<std::string>> dcs;
+        for (const StoreNode* kid : node->children()) {
+            const CompiledRecord& r = kid->rec();
+            if (!r.valid || r.address.empty()) continue;
+            std::string dc = kid->name();
+            size_t dash = dc.rfind('-');
+            if (dash != std::string::npos) dc = dc.substr(0, dash);
+            auto& v = dcs[dc];
+            bool dup = false;
+            for (const auto& e : v) dup = dup || e == r.address;
+            if (!dup) v.push_back(r.address);
+        }
+        dcs_ = std::move(dcs);
+        log_.info({{"dcs", Json((int64_t)dcs_.size())}},
+                  "setting recursion resolvers (zk)");
+        emitReady();
+        scheduleRefresh(kRefreshIntervalMs);
+        return;
+    }
+
+    log_.warn({{"source", Json(source)}},
+              "unknown resolver source; recursion disabled "
+              "(best effort)");
+    emitReady();
+    scheduleRefresh(kRetryInitMs);
+}
+
+std::vector<std::string> Recursion::ownAddrs() {
+    int64_t now = monotonicMillis();
+    if (!nicCache_.empty() && now - nicCacheAtMs_ <= 30000)
+        return nicCache_;
+    std::vector<std::string> out;
+    struct ifaddrs* ifap = nullptr;
+    if (getifaddrs(&ifap) == 0) {
+        for (struct ifaddrs* ifa = ifap; ifa; ifa = ifa->ifa_next) {
+            if (ifa->ifa_addr == nullptr) continue;
+            char buf[INET6_ADDRSTRLEN] = {0};
+            if (ifa->ifa_addr->sa_family == AF_INET) {
+                auto* sa = (struct sockaddr_in*)ifa->ifa_addr;
+                inet_ntop(AF_INET, &sa->sin_addr, buf, sizeof(buf));
+            } else if (ifa->ifa_addr->sa_family == AF_INET6) {
+                auto* sa = (struct sockaddr_in6*)ifa->ifa_addr;
+                inet_ntop(AF_INET6, &sa->sin6_addr, buf, sizeof(buf));
+            } else {
+                continue;
+            }
+            out.push_back(buf);
+        }
+        freeifaddrs(ifap);
+    }
+    nicCache_ = out;
+    nicCacheAtMs_ = now;
+    return out;
+}
+
+void Recursion::resolve(const Message& query, Message& resp,
+                        std::function<void()> done) {
+    const std::string domain =
+        query.questions.empty() ? "" : query.questions[0].name;
+    uint16_t qtype =
+        query.questions.empty() ? (uint16_t)TYPE_A
+                                : query.questions[0].qtype;
+    bool isPtr = qtype == TYPE_PTR;
+
+    auto refuse = [&]() {
+        resp.header.rcode = RCODE_REFUSED;
+        done();
+    };
+
+    /* Right dns domain? (recursion.js:330-333) */
+    if (!isPtr && !opts_.dnsDomain.empty()) {
+        if (domain.size() < opts_.dnsDomain.size() ||
+            domain.compare(domain.size() - opts_.dnsDomain.size(),
+                           opts_.dnsDomain.size(), opts_.dnsDomain) != 0) {
+            refuse();
+            return;
+        }
+    }
+
+    std::vector<std::string> upstreams;
+    if (!isPtr) {
+        /* DC label immediately left of dnsDomain (recursion.js:335-345) */
+        if (domain.size() <= opts_.dnsDomain.size()) {
+            refuse();
+            return;
+        }
+        std::string p = domain.substr(
+            0, domain.size() - opts_.dnsDomain.size() - 1);
+        size_t dot = p.rfind('.');
+        std::string dc = dot == std::string::npos ? p : p.substr(dot + 1);
+        auto it = dcs_.find(dc);
+        if (it == dcs_.end()) {
+            refuse();
+            return;
+        }
+        upstreams = it->second;
+    } else {
+        for (const auto& [dc, ips] : dcs_)
+            for (const auto& ip : ips) upstreams.push_back(ip);
+    }
+
+    /* Filter own NICs (recursion.js:356-376). */
+    std::vector<std::string> mine = ownAddrs();
+    std::vector<std::string> filtered;
+    for (const auto& u : upstreams) {
+        bool self = false;
+        for (const auto& m : mine) self = self || m == u;
+        if (!self) filtered.push_back(u);
+    }
+    if (filtered.empty() || fd_ < 0) {
+        refuse();
+        return;
+    }
+
+    auto up = std::make_shared<Upstream>();
+    up->qid = nextQid_ == 0 ? ++nextQid_ : nextQid_;
+    nextQid_ += 1;
+    up->hosts = std::move(filtered);
+    up->maxConcurrency = isPtr ? 100 : 2;  // recursion.js:64-78
+    up->resp = &resp;
+    up->done = std::move(done);
+    up->qname = domain;
+    pendingByQid_[up->qid] = up;
+
+    /* outgoing query: same question, rd cleared (recursion.js:258-261) */
+    Message out;
+    out.header.id = up->qid;
+    out.header.rd = false;
+    out.questions = query.questions;
+    auto wire = out.encode(0);
+
+    up->timeoutTimer = loop_->addTimer(kUpstreamTimeoutMs, [this, up]() {
+        up->timeoutTimer = 0;
+        finish(up, nullptr);
+    });
+
+    sendNext(up, wire);
+}
+
+void Recursion::sendNext(const std::shared_ptr<Upstream>& up,
+                         const std::vector<uint8_t>& wire) {
+    while (up->inFlight < up->maxConcurrency && !up->hosts.empty()) {
+        std::string host = up->hosts.front();
+        up->hosts.erase(up->hosts.begin());
+        struct sockaddr_in sa {};
+        sa.sin_family = AF_INET;
+        sa.sin_port = htons(53);
+        if (inet_pton(AF_INET, host.c_str(), &sa.sin_addr) != 1) {
+            up->errors++;
+            continue;
+        }
+        ssize_t rv = sendto(fd_, wire.data(), wire.size(), 0,
+                            (struct sockaddr*)&sa, sizeof(sa));
+        if (rv < 0) {
+            up->errors++;
+            continue;
+        }
+        up->inFlight++;
+    }
+    if (up->inFlight == 0) finish(up, nullptr);
+}
+
+void Recursion::onSockReadable() {
+    uint8_t buf[4096];
+    while (true) {
+        ssize_t nr = recv(fd_, buf, sizeof(buf), 0);
+        if (nr <= 0) return;
+        auto msg = Message::decode(buf, (size_t)nr);
+        if (!msg || !msg->header.qr) continue;
+        auto it = pendingByQid_.find(msg->header.id);
+        if (it == pendingByQid_.end()) continue;
+        auto up = it->second;
+        up->inFlight--;
+        if (msg->header.rcode == RCODE_NOERROR && !msg->answers.empty()) {
+            finish(up, &*msg);
+        } else {
+            up->errors++;
+            if (up->inFlight == 0 && up->hosts.empty())
+                finish(up, nullptr);
+            else if (!up->hosts.empty()) {
+                /* keep trying further resolvers */
+                Message out;
+                out.header.id = up->qid;
+                out.header.rd = false;
+                /* reconstruct question from resp (same as original) */
+                out.questions = up->resp->questions;
+                auto wire = out.encode(0);
+                sendNext(up, wire);
+            }
+        }
+    }
+}
+
+void Recursion::finish(const std::shared_ptr<Upstream>& up,
+                       const Message* answer) {
+    if (up->finished) return;
+    up->finished = true;
+    if (up->timeoutTimer) {
+        loop_->cancelTimer(up->timeoutTimer);
+        up->timeoutTimer = 0;
+    }
+    pendingByQid_.erase(up->qid);
+
+    Message& resp = *up->resp;
+    size_t accepted = 0;
+    if (answer != nullptr) {
+        for (const auto& rec : answer->answers) {
+            /* accepted types only (recursion.js:299-323) */
+            switch (rec.type) {
+            case TYPE_A:
+            case TYPE_AAAA:
+            case TYPE_TXT:
+            case TYPE_PTR:
+            case TYPE_CNAME:
+            case TYPE_SRV: {
+                Record r = rec;
+                r.name = up->qname;  // answers under the original name
+                resp.answers.push_back(std::move(r));
+                accepted++;
+                break;
+            }
+            default:
+                log_.warn("upstream ns returned unsupported record "
+                          "type, dropping");
+                break;
+            }
+        }
+    }
+    if (accepted == 0) resp.header.rcode = RCODE_REFUSED;
+    up->done();
+}
+
+}  // namespace bamd

@@ -1,0 +1,111 @@
+/*
+ * binder-amd: cross-DC recursion forwarder (lib/recursion.js equivalent).
+ *
+ * Best-effort forwarding of cache misses to other datacenters' binders:
+ *   - resolver registry refreshed every 5 minutes
+ *     (recursion.js:40, 202-249), init retried every 15 s, 'ready' fires
+ *     regardless (174-197);
+ *   - non-PTR queries route to the DC named by the label immediately
+ *     left of dnsDomain (335-345); PTR fans out to every known resolver
+ *     (346-354);
+ *   - own-NIC addresses filtered with a 30 s cache (356-376);
+ *   - upstream queries have RD cleared and a 3 s timeout (257-261);
+ *   - only A/AAAA/TXT/PTR/CNAME/SRV upstream answers are accepted
+ *     (299-323); empty results => REFUSED (292-296).
+ *
+ * Registry sources are pluggable where the reference hardcodes UFDS/LDAP
+ * (its listResolvers(region) call, recursion.js:210-219):
+ *   "static": {"source":"static","dcs":{"dc1":["10.0.0.5",...]}}
+ *   "zk":     {"source":"zk","path":"/resolvers"} — znode JSON payload
+ *             {"dc1":["ip",...],...}, read via the shared mirror client.
+ * An LDAP driver for UFDS parity is planned (gap tracked in docs/).
+ */
+#pragma once
+
+#include <functional>
+#include <map>
+#include <memory>
+#include <string>
+#include <vector>
+
+#include "../common/json.hpp"
+#include "../common/log.hpp"
+#include "../common/loop.hpp"
+#include "../engine/store.hpp"
+#include "server.hpp"
+
+namespace bamd {
+
+struct RecursionOptions {
+    std::string regionName;
+    std::string datacenterName;
+    std::string dnsDomain;
+    Json config;  // the whole `recursion` config block
+};
+
+class Recursion : public RecursionIface {
+  public:
+    Recursion(EventLoop* loop, Logger log, RecursionOptions opts,
+              const Store* store);
+    ~Recursion() override;
+
+    void init();
+    bool ready() const { return ready_; }
+    void onReady(std::function<void()> cb) { readyCb_ = std::move(cb); }
+
+    /* RecursionIface */
+    void resolve(const dns::Message& query, dns::Message& resp,
+                 std::function<void()> done) override;
+
+    /* Direct registry injection (tests / static source). */
+    void setDcs(std::map<std::string, std::vector<std::string>> dcs) {
+        dcs_ = std::move(dcs);
+    }
+    const std::map<std::string, std::vector<std::string>>& dcs() const {
+        return dcs_;
+    }
+
+  private:
+    struct Upstream {
+        uint16_t qid;
+        std::vector<std::string> hosts;  // remaining unsent
+        int inFlight = 0;
+        int errors = 0;
+        int maxConcurrency;
+        dns::Message* resp;
+        std::function<void()> done;
+        std::string qname;
+        uint64_t timeoutTimer = 0;
+        bool finished = false;
+    };
+
+    void refresh();
+    void scheduleRefresh(int64_t ms);
+    void emitReady();
+    std::vector<std::string> ownAddrs();
+    bool openSocket();
+    void onSockReadable();
+    void sendNext(const std::shared_ptr<Upstream>& up,
+                  const std::vector<uint8_t>& wire);
+    void finish(const std::shared_ptr<Upstream>& up,
+                const dns::Message* answer);
+
+    EventLoop* loop_;
+    Logger log_;
+    RecursionOptions opts_;
+    const Store* store_;
+
+    std::map<std::string, std::vector<std::string>> dcs_;
+    bool ready_ = false;
+    std::function<void()> readyCb_;
+    uint64_t refreshTimer_ = 0;
+
+    int fd_ = -1;  // shared upstream UDP socket
+    std::map<uint16_t, std::shared_ptr<Upstream>> pendingByQid_;
+    uint16_t nextQid_ = 1;
+
+    std::vector<std::string> nicCache_;
+    int64_t nicCacheAtMs_ = 0;
+};
+
+}  // namespace bamd

@@ -1,0 +1,605 @@
+#include "server.hpp"
+
+#include <arpa/inet.h>
+#include <sys/epoll.h>
+#include <unistd.h>
+
+#include <cstdio>
+#include <cstring>
+
+#include "../balancer/protocol.hpp"
+
+namespace bamd {
+
+using namespace dns;
+
+void fillClientInfo(ClientInfo& ci, const struct sockaddr_storage& ss,
+                    const char* family) {
+    ci.family = family;
+    if (ss.ss_family == AF_INET) {
+        const auto* sa = (const struct sockaddr_in*)&ss;
+        inet_ntop(AF_INET, &sa->sin_addr, ci.address, sizeof(ci.address));
+        ci.port = ntohs(sa->sin_port);
+    } else if (ss.ss_family == AF_INET6) {
+        const auto* sa = (const struct sockaddr_in6*)&ss;
+        inet_ntop(AF_INET6, &sa->sin6_addr, ci.address, sizeof(ci.address));
+        ci.port = ntohs(sa->sin6_port);
+    }
+}
+
+DnsServer::DnsServer(EventLoop* loop, Logger log, ServerOptions opts,
+                     Engine* engine, Collector* collector)
+    : loop_(loop), log_(std::move(log)), opts_(std::move(opts)),
+      engine_(engine) {
+    /* Metric names/help strings match the reference
+     * (lib/server.js:31-34, 456-469). */
+    reqCounter_ = collector->counter("binder_requests_completed",
+                                     "count of Binder requests completed");
+    latHist_ = collector->histogram(
+        "binder_request_latency_seconds",
+        "total time to process Binder requests");
+    sizeHist_ = collector->histogram("binder_response_size_bytes",
+                                     "size in bytes of Binder responses");
+
+    rxArena_.resize(kBatch * kInBuf);
+    rxHdrs_.resize(kBatch);
+    rxIovs_.resize(kBatch);
+    rxAddrs_.resize(kBatch);
+    txBufs_.resize(kBatch);
+    txHdrs_.resize(kBatch);
+    txIovs_.resize(kBatch);
+    txAddrs_.resize(kBatch);
+}
+
+DnsServer::~DnsServer() { stop(); }
+
+/* Bind a socket of the given type to opts_.host:port; v6 when the host
+ * looks v6 or is empty (dual-stack any). */
+static int bindSocket(const std::string& host, uint16_t port, int type,
+                      uint16_t* boundPort) {
+    bool v6 = host.empty() || host.find(':') != std::string::npos;
+    int fd = socket(v6 ? AF_INET6 : AF_INET,
+                    type | SOCK_NONBLOCK | SOCK_CLOEXEC, 0);
+    if (fd < 0) return -1;
+    int one = 1;
+    setsockopt(fd, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+    if (type == SOCK_DGRAM) {
+        int sz = 4 << 20;
+        setsockopt(fd, SOL_SOCKET, SO_RCVBUF, &sz, sizeof(sz));
+        setsockopt(fd, SOL_SOCKET, SO_SNDBUF, &sz, sizeof(sz));
+    }
+    int rv;
+    if (v6) {
+        int zero = 0;
+        setsockopt(fd, IPPROTO_IPV6, IPV6_V6ONLY,
+                   host.empty() ? &zero : &one, sizeof(int));
+        struct sockaddr_in6 sa {};
+        sa.sin6_family = AF_INET6;
+        sa.sin6_port = htons(port);
+        if (host.empty())
+            sa.sin6_addr = in6addr_any;
+        else if (inet_pton(AF_INET6, host.c_str(), &sa.sin6_addr) != 1) {
+            close(fd);
+            return -1;
+        }
+        rv = bind(fd, (struct sockaddr*)&sa, sizeof(sa));
+        if (rv == 0 && boundPort) {
+            socklen_t sl = sizeof(sa);
+            getsockname(fd, (struct sockaddr*)&sa, &sl);
+            *boundPort = ntohs(sa.sin6_port);
+        }
+    } else {
+        struct sockaddr_in sa {};
+        sa.sin_family = AF_INET;
+        sa.sin_port = htons(port);
+        if (inet_pton(AF_INET, host.c_str(), &sa.sin_addr) != 1) {
+            close(fd);
+            return -1;
+        }
+        rv = bind(fd, (struct sockaddr*)&sa, sizeof(sa));
+        if (rv == 0 && boundPort) {
+            socklen_t sl = sizeof(sa);
+            getsockname(fd, (struct sockaddr*)&sa, &sl);
+            *boundPort = ntohs(sa.sin_port);
+        }
+    }
+    if (rv != 0) {
+        close(fd);
+        return -1;
+    }
+    return fd;
+}
+
+bool DnsServer::openUdp() {
+    udpFd_ = bindSocket(opts_.host, opts_.port, SOCK_DGRAM, &boundPort_);
+    if (udpFd_ < 0) {
+        log_.error({{"port", Json((int)opts_.port)},
+                    {"host", Json(opts_.host)}},
+                   "failed to bind UDP socket");
+        return false;
+    }
+    loop_->addFd(udpFd_, EPOLLIN, [this](uint32_t) { onUdpReadable(); });
+    log_.info({{"host", Json(opts_.host)}, {"port", Json((int)boundPort_)}},
+              "UDP DNS service started");
+    return true;
+}
+
+bool DnsServer::openTcp() {
+    tcpFd_ = bindSocket(opts_.host, boundPort_ ? boundPort_ : opts_.port,
+                        SOCK_STREAM, nullptr);
+    if (tcpFd_ < 0) {
+        log_.error("failed to bind TCP socket");
+        return false;
+    }
+    if (listen(tcpFd_, 512) != 0) {
+        log_.error("failed to listen on TCP socket");
+        return false;
+    }
+    loop_->addFd(tcpFd_, EPOLLIN, [this](uint32_t) { onTcpAccept(); });
+    log_.info({{"host", Json(opts_.host)}, {"port", Json((int)boundPort_)}},
+              "TCP DNS service started");
+    return true;
+}
+
+bool DnsServer::openBalancer() {
+    balFd_ = socket(AF_UNIX, SOCK_STREAM | SOCK_NONBLOCK | SOCK_CLOEXEC, 0);
+    if (balFd_ < 0) return false;
+    struct sockaddr_un sa {};
+    sa.sun_family = AF_UNIX;
+    snprintf(sa.sun_path, sizeof(sa.sun_path), "%s",
+             opts_.balancerSocket.c_str());
+    /* Pre-unlink stale socket (main.js:196-199). */
+    unlink(sa.sun_path);
+    if (bind(balFd_, (struct sockaddr*)&sa, sizeof(sa)) != 0 ||
+        listen(balFd_, 64) != 0) {
+        log_.error({{"path", Json(opts_.balancerSocket)}},
+                   "failed to bind balancer socket");
+        close(balFd_);
+        balFd_ = -1;
+        return false;
+    }
+    loop_->addFd(balFd_, EPOLLIN, [this](uint32_t) { onBalAccept(); });
+    log_.info({{"path", Json(opts_.balancerSocket)}},
+              "Balancer service started");
+    return true;
+}
+
+bool DnsServer::start() {
+    if (!openUdp() || !openTcp()) return false;
+    if (!opts_.balancerSocket.empty() && !openBalancer()) return false;
+    return true;
+}
+
+void DnsServer::stop() {
+    auto closeFd = [this](int& fd) {
+        if (fd >= 0) {
+            loop_->delFd(fd);
+            close(fd);
+            fd = -1;
+        }
+    };
+    closeFd(udpFd_);
+    closeFd(tcpFd_);
+    closeFd(balFd_);
+    if (!opts_.balancerSocket.empty()) unlink(opts_.balancerSocket.c_str());
+    for (auto& [fd, c] : tcpConns_) {
+        loop_->delFd(fd);
+        close(fd);
+    }
+    tcpConns_.clear();
+    for (auto& [fd, c] : balConns_) {
+        loop_->delFd(fd);
+        close(fd);
+    }
+    balConns_.clear();
+}
+
+/* ---------------- query pipeline ---------------- */
+
+bool DnsServer::process(const uint8_t* data, size_t len, bool udp,
+                        const ClientInfo& ci, std::vector<uint8_t>& out,
+                        std::function<void(std::vector<uint8_t>)>
+                            asyncReply) {
+    int64_t start = monotonicMillis();
+    auto parsed = Message::decode(data, len);
+    if (!parsed) return true;  // drop malformed (out empty)
+    if (parsed->header.qr) return true;  // ignore responses
+
+    Message& query = *parsed;
+    size_t limit = 0;
+    bool hasEdns = query.edns() != nullptr;
+    if (udp) {
+        limit = 512;
+        if (hasEdns) {
+            uint16_t adv = query.edns()->rclass;
+            limit = adv < 512 ? 512 : (adv > 4096 ? 4096 : adv);
+        }
+    }
+
+    Message resp;
+    QueryResult qr = engine_->handle(query, resp);
+
+    if (qr.action == QueryResult::Action::Recurse &&
+        recursion_ != nullptr) {
+        /* Hand off: recursion fills resp then we encode+deliver. */
+        auto respHeap = std::make_shared<Message>(std::move(resp));
+        auto queryHeap = std::make_shared<Message>(std::move(query));
+        ClientInfo ciCopy = ci;
+        size_t lim = limit;
+        bool edns = hasEdns;
+        recursion_->resolve(
+            *queryHeap, *respHeap,
+            [this, respHeap, queryHeap, ciCopy, lim, edns, start,
+             asyncReply, qr]() {
+                if (edns)
+                    respHeap->additionals.push_back(Record::OPT(1400));
+                auto wire = respHeap->encode(lim);
+                size_t n = wire.size();
+                asyncReply(std::move(wire));
+                afterQuery(*queryHeap, *respHeap, qr, ciCopy, n,
+                           start);
+            });
+        return false;
+    }
+
+    if (hasEdns) resp.additionals.push_back(Record::OPT(1400));
+    out = resp.encode(limit);
+    afterQuery(query, resp, qr, ci, out.size(), start);
+    return true;
+}
+
+void DnsServer::afterQuery(const Message& query, const Message& resp,
+                           const QueryResult& qr, const ClientInfo& ci,
+                           size_t bytesSent, int64_t startMs) {
+    ++served_;
+    int64_t lat = monotonicMillis() - startMs;
+
+    const char* qtype = query.questions.empty()
+                            ? nullptr
+                            : typeName(query.questions[0].qtype);
+    if (qtype != nullptr) {
+        std::string label = std::string("type=\"") + qtype + "\"";
+        reqCounter_->increment(label);
+        latHist_->observe(label, (double)lat / 1000.0);
+        sizeHist_->observe(label, (double)bytesSent);
+    }
+
+    /* Per-query log line (server.js:537-590); warn when >1s. */
+    LogLevel lv = lat > 1000 ? LogLevel::Warn : LogLevel::Info;
+    if (!log_.enabled(lv)) return;
+
+    const std::string& dd = engine_->config().dnsDomain;
+    std::string dotdd = dd.empty() ? "" : "." + dd;
+    JsonObject o;
+    o["req_id"] = Json((int64_t)query.header.id);
+    o["client"] = Json(std::string(ci.address));
+    o["port"] = Json(std::to_string(ci.port) + "/" + ci.family);
+    {
+        JsonObject q;
+        if (!query.questions.empty()) {
+            q["name"] = Json(query.questions[0].name);
+            q["type"] = Json(typeName(query.questions[0].qtype));
+        }
+        o["query"] = Json(std::move(q));
+    }
+    o["edns"] = Json(query.edns() != nullptr);
+    o["rcode"] = Json(rcodeName(resp.header.rcode));
+    {
+        JsonArray answers;
+        for (const auto& r : resp.answers) {
+            std::string s = typeName(r.type);
+            if (r.type == TYPE_SRV) {
+                std::string t =
+                    dd.empty() ? r.target : stripSuffix(dotdd, r.target);
+                s += " " + t + ":" + std::to_string(r.port);
+            } else if (r.type == TYPE_A || r.type == TYPE_AAAA) {
+                s += " " + r.addrString();
+            } else if (r.type == TYPE_PTR) {
+                s += " " + r.target;
+            }
+            answers.push_back(Json(std::move(s)));
+        }
+        o["answers"] = Json(std::move(answers));
+    }
+    {
+        JsonArray adds;
+        for (const auto& r : resp.additionals) {
+            if (r.type == TYPE_OPT) continue;  // OPT filtered from logs
+            std::string nm =
+                dd.empty() ? r.name : stripSuffix(dotdd, r.name);
+            adds.push_back(Json(nm + " " + typeName(r.type) + " " +
+                                r.addrString()));
+        }
+        o["additional"] = Json(std::move(adds));
+    }
+    o["latency"] = Json(lat);
+    log_.log(lv, "DNS query", o);
+}
+
+/* ---------------- UDP ---------------- */
+
+void DnsServer::onUdpReadable() {
+    while (true) {
+        for (int i = 0; i < kBatch; ++i) {
+            rxIovs_[i].iov_base = rxArena_.data() + (size_t)i * kInBuf;
+            rxIovs_[i].iov_len = kInBuf;
+            memset(&rxHdrs_[i], 0, sizeof(rxHdrs_[i]));
+            rxHdrs_[i].msg_hdr.msg_iov = &rxIovs_[i];
+            rxHdrs_[i].msg_hdr.msg_iovlen = 1;
+            rxHdrs_[i].msg_hdr.msg_name = &rxAddrs_[i];
+            rxHdrs_[i].msg_hdr.msg_namelen = sizeof(rxAddrs_[i]);
+        }
+        int n = recvmmsg(udpFd_, rxHdrs_.data(), kBatch, 0, nullptr);
+        if (n <= 0) return;  // EAGAIN or error: wait for next wakeup
+
+        int nOut = 0;
+        for (int i = 0; i < n; ++i) {
+            ClientInfo ci;
+            fillClientInfo(ci, rxAddrs_[i], "udp");
+            std::vector<uint8_t>& out = txBufs_[nOut];
+            out.clear();
+            struct sockaddr_storage srcCopy = rxAddrs_[i];
+            socklen_t srcLen = rxHdrs_[i].msg_hdr.msg_namelen;
+            bool sync = process(
+                rxArena_.data() + (size_t)i * kInBuf, rxHdrs_[i].msg_len,
+                true, ci, out,
+                [this, srcCopy, srcLen](std::vector<uint8_t> wire) {
+                    /* async (recursion) reply path */
+                    sendto(udpFd_, wire.data(), wire.size(), 0,
+                           (const struct sockaddr*)&srcCopy, srcLen);
+                });
+            if (sync && !out.empty()) {
+                txAddrs_[nOut] = rxAddrs_[i];
+                txIovs_[nOut].iov_base = out.data();
+                txIovs_[nOut].iov_len = out.size();
+                memset(&txHdrs_[nOut], 0, sizeof(txHdrs_[nOut]));
+                txHdrs_[nOut].msg_hdr.msg_iov = &txIovs_[nOut];
+                txHdrs_[nOut].msg_hdr.msg_iovlen = 1;
+                txHdrs_[nOut].msg_hdr.msg_name = &txAddrs_[nOut];
+                txHdrs_[nOut].msg_hdr.msg_namelen = srcLen;
+                ++nOut;
+            }
+        }
+        int sent = 0;
+        while (sent < nOut) {
+            int rv = sendmmsg(udpFd_, txHdrs_.data() + sent, nOut - sent, 0);
+            if (rv <= 0) break;  // EAGAIN: drop the rest (UDP best-effort)
+            sent += rv;
+        }
+        if (n < kBatch) return;  // drained
+    }
+}
+
+/* ---------------- TCP ---------------- */
+
+void DnsServer::onTcpAccept() {
+    while (true) {
+        struct sockaddr_storage ss;
+        socklen_t sl = sizeof(ss);
+        int fd = accept4(tcpFd_, (struct sockaddr*)&ss, &sl,
+                         SOCK_NONBLOCK | SOCK_CLOEXEC);
+        if (fd < 0) return;
+        auto conn = std::make_shared<TcpConn>();
+        conn->fd = fd;
+        fillClientInfo(conn->ci, ss, "tcp");
+        TcpConn* raw = conn.get();
+        tcpConns_[fd] = std::move(conn);
+        loop_->addFd(fd, EPOLLIN,
+                     [this, raw](uint32_t ev) { onTcpConn(raw, ev); });
+    }
+}
+
+void DnsServer::closeTcp(TcpConn* c) {
+    if (c->closed) return;
+    c->closed = true;
+    loop_->delFd(c->fd);
+    close(c->fd);
+    /* shared_ptr keeps the object alive for in-flight async replies */
+    tcpConns_.erase(c->fd);
+}
+
+void DnsServer::tcpFlush(TcpConn* c) {
+    while (!c->out.empty()) {
+        ssize_t nw = write(c->fd, c->out.data(), c->out.size());
+        if (nw > 0) {
+            c->out.erase(0, (size_t)nw);
+            continue;
+        }
+        if (nw < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) {
+            if (!c->writeBlocked) {
+                c->writeBlocked = true;
+                loop_->modFd(c->fd, EPOLLIN | EPOLLOUT);
+            }
+            return;
+        }
+        closeTcp(c);
+        return;
+    }
+    if (c->writeBlocked) {
+        c->writeBlocked = false;
+        loop_->modFd(c->fd, EPOLLIN);
+    }
+}
+
+void DnsServer::onTcpConn(TcpConn* c, uint32_t events) {
+    /* keep-alive: flush paths may close+erase the connection */
+    std::shared_ptr<TcpConn> keep = tcpConns_[c->fd];
+    if (events & (EPOLLHUP | EPOLLERR)) {
+        closeTcp(c);
+        return;
+    }
+    if (events & EPOLLOUT) tcpFlush(c);
+    if (c->closed || !(events & EPOLLIN)) return;
+
+    char buf[8192];
+    while (true) {
+        ssize_t nr = read(c->fd, buf, sizeof(buf));
+        if (nr > 0) {
+            c->in.append(buf, (size_t)nr);
+            if (c->in.size() > (1 << 20)) {  // runaway client
+                closeTcp(c);
+                return;
+            }
+            continue;
+        }
+        if (nr < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) break;
+        closeTcp(c);
+        return;
+    }
+
+    /* DNS-over-TCP framing: u16be length prefix per message. */
+    while (c->in.size() >= 2) {
+        size_t mlen = ((size_t)(uint8_t)c->in[0] << 8) |
+                      (uint8_t)c->in[1];
+        if (c->in.size() < 2 + mlen) break;
+        std::vector<uint8_t> out;
+        std::shared_ptr<TcpConn> self = keep;
+        bool sync = process(
+            (const uint8_t*)c->in.data() + 2, mlen, false, c->ci, out,
+            [this, self](std::vector<uint8_t> wire) {
+                if (self->closed) return;
+                self->out.push_back((char)(wire.size() >> 8));
+                self->out.push_back((char)wire.size());
+                self->out.append((const char*)wire.data(), wire.size());
+                tcpFlush(self.get());
+            });
+        if (sync) {
+            if (!out.empty()) {
+                c->out.push_back((char)(out.size() >> 8));
+                c->out.push_back((char)out.size());
+                c->out.append((const char*)out.data(), out.size());
+            }
+        }
+        c->in.erase(0, 2 + mlen);
+    }
+    tcpFlush(c);
+}
+
+/* ---------------- balancer socket ---------------- */
+
+void DnsServer::onBalAccept() {
+    while (true) {
+        int fd = accept4(balFd_, nullptr, nullptr,
+                         SOCK_NONBLOCK | SOCK_CLOEXEC);
+        if (fd < 0) return;
+        auto conn = std::make_shared<BalConn>();
+        conn->fd = fd;
+        BalConn* raw = conn.get();
+        balConns_[fd] = std::move(conn);
+        loop_->addFd(fd, EPOLLIN,
+                     [this, raw](uint32_t ev) { onBalConn(raw, ev); });
+    }
+}
+
+void DnsServer::closeBal(BalConn* c) {
+    if (c->closed) return;
+    c->closed = true;
+    loop_->delFd(c->fd);
+    close(c->fd);
+    balConns_.erase(c->fd);
+}
+
+void DnsServer::balFlush(BalConn* c) {
+    while (!c->out.empty()) {
+        ssize_t nw = write(c->fd, c->out.data(), c->out.size());
+        if (nw > 0) {
+            c->out.erase(0, (size_t)nw);
+            continue;
+        }
+        if (nw < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) {
+            if (!c->writeBlocked) {
+                c->writeBlocked = true;
+                loop_->modFd(c->fd, EPOLLIN | EPOLLOUT);
+            }
+            return;
+        }
+        closeBal(c);
+        return;
+    }
+    if (c->writeBlocked) {
+        c->writeBlocked = false;
+        loop_->modFd(c->fd, EPOLLIN);
+    }
+}
+
+void DnsServer::onBalConn(BalConn* c, uint32_t events) {
+    std::shared_ptr<BalConn> keep = balConns_[c->fd];
+    if (events & (EPOLLHUP | EPOLLERR)) {
+        closeBal(c);
+        return;
+    }
+    if (events & EPOLLOUT) balFlush(c);
+    if (c->closed || !(events & EPOLLIN)) return;
+
+    char buf[16384];
+    while (true) {
+        ssize_t nr = read(c->fd, buf, sizeof(buf));
+        if (nr > 0) {
+            c->in.append(buf, (size_t)nr);
+            continue;
+        }
+        if (nr < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) break;
+        closeBal(c);
+        return;
+    }
+
+    while (c->in.size() >= bsock::kHeaderLen) {
+        const uint8_t* h = (const uint8_t*)c->in.data();
+        if (h[0] != bsock::kMagic) {
+            closeBal(c);
+            return;
+        }
+        uint8_t type = h[1];
+        uint32_t plen = bsock::getU32(h + 2);
+        if (plen > bsock::kMaxPayload) {
+            closeBal(c);
+            return;
+        }
+        if (c->in.size() < bsock::kHeaderLen + plen) break;
+        const uint8_t* payload = h + bsock::kHeaderLen;
+
+        if (type == bsock::FRAME_PING) {
+            bsock::appendFrame(c->out, bsock::FRAME_PONG, "");
+        } else if (type == bsock::FRAME_QUERY) {
+            bsock::QueryFrame qf;
+            if (bsock::parseQuery(payload, plen, qf)) {
+                ClientInfo ci;
+                ci.family = qf.proto == 1 ? "tcp" : "udp";
+                ci.port = qf.srcPort;
+                if (qf.family == 4)
+                    inet_ntop(AF_INET, qf.srcAddr, ci.address,
+                              sizeof(ci.address));
+                else
+                    inet_ntop(AF_INET6, qf.srcAddr, ci.address,
+                              sizeof(ci.address));
+                std::vector<uint8_t> out;
+                uint32_t reqId = qf.reqId;
+                bool udp = qf.proto == 0;
+                std::shared_ptr<BalConn> self = keep;
+                bool sync = process(
+                    qf.dns, qf.dnsLen, udp, ci, out,
+                    [this, self, reqId](std::vector<uint8_t> wire) {
+                        if (self->closed) return;
+                        std::string payload2;
+                        bsock::putU32(payload2, reqId);
+                        payload2.append((const char*)wire.data(),
+                                        wire.size());
+                        bsock::appendFrame(self->out, bsock::FRAME_REPLY,
+                                           payload2);
+                        balFlush(self.get());
+                    });
+                if (sync && !out.empty()) {
+                    std::string payload2;
+                    bsock::putU32(payload2, qf.reqId);
+                    payload2.append((const char*)out.data(), out.size());
+                    bsock::appendFrame(c->out, bsock::FRAME_REPLY,
+                                       payload2);
+                }
+            }
+        }
+        c->in.erase(0, bsock::kHeaderLen + plen);
+    }
+    balFlush(c);
+}
+
+}  // namespace bamd

@@ -1,0 +1,496 @@
+#include "client.hpp"
+
+#include <arpa/inet.h>
+#include <netinet/in.h>
+#include <netinet/tcp.h>
+#include <sys/epoll.h>
+#include <sys/socket.h>
+#include <unistd.h>
+
+#include <cstring>
+
+namespace bamd::zk {
+
+ZkClient::ZkClient(EventLoop* loop, Logger log, ZkConfig cfg)
+    : loop_(loop),
+      log_(log.child({{"component", Json("ZkClient")}})),
+      cfg_(std::move(cfg)) {}
+
+ZkClient::~ZkClient() { close(); }
+
+void ZkClient::start() {
+    closing_ = false;
+    if (state_ == State::Closed) connectStart();
+}
+
+void ZkClient::close() {
+    closing_ = true;
+    if (reconnectTimer_) {
+        loop_->cancelTimer(reconnectTimer_);
+        reconnectTimer_ = 0;
+    }
+    if (state_ == State::Connected) {
+        /* best-effort CloseSession */
+        Writer w;
+        w.i32(nextXid());
+        w.i32(OP_CLOSE);
+        sendPacket(w.buf);
+        flush();
+    }
+    teardown();
+    state_ = State::Closed;
+}
+
+void ZkClient::teardown() {
+    if (pingTimer_) {
+        loop_->cancelTimer(pingTimer_);
+        pingTimer_ = 0;
+    }
+    if (fd_ >= 0) {
+        loop_->delFd(fd_);
+        ::close(fd_);
+        fd_ = -1;
+    }
+    inBuf_.clear();
+    outBuf_.clear();
+    writeBlocked_ = false;
+    failAllPending();
+}
+
+void ZkClient::failAllPending() {
+    auto pend = std::move(pending_);
+    pending_.clear();
+    static const Stat kStat;
+    for (auto& [xid, p] : pend) {
+        if (p.dataCb) p.dataCb(ZCONNECTIONLOSS, "", kStat);
+        if (p.childrenCb) p.childrenCb(ZCONNECTIONLOSS, {});
+        if (p.statCb) p.statCb(ZCONNECTIONLOSS, kStat);
+        if (p.stringCb) p.stringCb(ZCONNECTIONLOSS, "");
+        if (p.voidCb) p.voidCb(ZCONNECTIONLOSS);
+    }
+}
+
+void ZkClient::scheduleReconnect() {
+    if (closing_ || reconnectTimer_) return;
+    state_ = State::Closed;
+    reconnects_++;
+    reconnectTimer_ = loop_->addTimer(cfg_.reconnectDelayMs, [this]() {
+        reconnectTimer_ = 0;
+        if (!closing_) connectStart();
+    });
+}
+
+void ZkClient::connectStart() {
+    teardown();
+    state_ = State::Connecting;
+    fd_ = socket(AF_INET, SOCK_STREAM | SOCK_NONBLOCK | SOCK_CLOEXEC, 0);
+    if (fd_ < 0) {
+        scheduleReconnect();
+        return;
+    }
+    int one = 1;
+    setsockopt(fd_, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+    struct sockaddr_in sa {};
+    sa.sin_family = AF_INET;
+    sa.sin_port = htons(cfg_.port);
+    if (inet_pton(AF_INET, cfg_.host.c_str(), &sa.sin_addr) != 1) {
+        /* loopback name fallback */
+        if (cfg_.host == "localhost")
+            inet_pton(AF_INET, "127.0.0.1", &sa.sin_addr);
+        else {
+            log_.error({{"host", Json(cfg_.host)}},
+                       "ZK host is not an IPv4 address");
+            teardown();
+            scheduleReconnect();
+            return;
+        }
+    }
+    int rv = ::connect(fd_, (struct sockaddr*)&sa, sizeof(sa));
+    if (rv != 0 && errno != EINPROGRESS) {
+        teardown();
+        scheduleReconnect();
+        return;
+    }
+    loop_->addFd(fd_, EPOLLIN | EPOLLOUT,
+                 [this](uint32_t ev) { onSockEvent(ev); });
+}
+
+void ZkClient::sendHandshake() {
+    state_ = State::Handshaking;
+    Writer w;
+    w.i32(0);                       // protocolVersion
+    w.i64(lastZxid_);               // lastZxidSeen
+    w.i32(cfg_.sessionTimeoutMs);   // timeOut
+    w.i64(sessionId_);              // resume previous session if any
+    if (passwd_.empty()) passwd_.assign(16, '\0');
+    w.buffer(passwd_);
+    w.boolean(false);               // readOnly
+    sendPacket(w.buf);
+    flush();
+}
+
+void ZkClient::onSockEvent(uint32_t events) {
+    if (state_ == State::Connecting) {
+        int err = 0;
+        socklen_t elen = sizeof(err);
+        getsockopt(fd_, SOL_SOCKET, SO_ERROR, &err, &elen);
+        if (err != 0 || (events & (EPOLLERR | EPOLLHUP))) {
+            log_.debug("ZK connect failed; retrying");
+            teardown();
+            scheduleReconnect();
+            return;
+        }
+        loop_->modFd(fd_, EPOLLIN);
+        sendHandshake();
+        return;
+    }
+    if (events & (EPOLLERR | EPOLLHUP)) {
+        log_.warn("ZK connection lost");
+        teardown();
+        scheduleReconnect();
+        return;
+    }
+    if (events & EPOLLOUT) {
+        writeBlocked_ = false;
+        loop_->modFd(fd_, EPOLLIN);
+        flush();
+    }
+    if (events & EPOLLIN) onReadable();
+}
+
+void ZkClient::onReadable() {
+    char buf[65536];
+    while (fd_ >= 0) {
+        ssize_t nr = read(fd_, buf, sizeof(buf));
+        if (nr > 0) {
+            inBuf_.append(buf, (size_t)nr);
+            continue;
+        }
+        if (nr < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) break;
+        log_.warn("ZK connection closed by server");
+        bool hadSession = sessionId_ != 0;
+        teardown();
+        (void)hadSession;
+        scheduleReconnect();
+        return;
+    }
+    lastPacketRecvMs_ = monotonicMillis();
+
+    while (inBuf_.size() >= 4) {
+        uint32_t plen = ((uint32_t)(uint8_t)inBuf_[0] << 24) |
+                        ((uint32_t)(uint8_t)inBuf_[1] << 16) |
+                        ((uint32_t)(uint8_t)inBuf_[2] << 8) |
+                        (uint32_t)(uint8_t)inBuf_[3];
+        if (plen > (64u << 20)) {
+            log_.error("ZK packet too large; dropping connection");
+            teardown();
+            scheduleReconnect();
+            return;
+        }
+        if (inBuf_.size() < 4 + plen) break;
+        /* copy out: callbacks may mutate inBuf_ by issuing requests */
+        std::string pkt = inBuf_.substr(4, plen);
+        inBuf_.erase(0, 4 + plen);
+        if (state_ == State::Handshaking)
+            onConnectResponse((const uint8_t*)pkt.data(), pkt.size());
+        else
+            onPacket((const uint8_t*)pkt.data(), pkt.size());
+        if (fd_ < 0) return;  // teardown happened in callback
+    }
+}
+
+void ZkClient::onConnectResponse(const uint8_t* data, size_t len) {
+    Reader r{data, len};
+    int32_t proto = r.i32();
+    int32_t timeout = r.i32();
+    int64_t sid = r.i64();
+    std::string pw = r.str();
+    (void)proto;
+    if (!r.ok || timeout <= 0 || sid == 0) {
+        /* session expired (server refuses resume): start a fresh one */
+        log_.warn("ZK session expired; creating a new session");
+        sessionId_ = 0;
+        passwd_.clear();
+        lastZxid_ = 0;
+        teardown();
+        scheduleReconnect();
+        return;
+    }
+    bool isNew = sid != sessionId_;
+    sessionId_ = sid;
+    passwd_ = pw;
+    negotiatedTimeout_ = timeout;
+    state_ = State::Connected;
+    sessions_++;
+    log_.info({{"sessionId", Json((int64_t)sid)},
+               {"timeout", Json((int64_t)timeout)},
+               {"new", Json(isNew)}},
+              "ZK session established");
+    armPingTimer();
+    /* 'session' fires on every usable (re)connection — the mirror above
+     * resyncs, which also covers resumed sessions with missed events. */
+    if (sessionCb_) sessionCb_();
+}
+
+void ZkClient::armPingTimer() {
+    if (pingTimer_) loop_->cancelTimer(pingTimer_);
+    int t = negotiatedTimeout_ > 0 ? negotiatedTimeout_
+                                   : cfg_.sessionTimeoutMs;
+    pingTimer_ = loop_->addTimer(t / 3, [this]() {
+        pingTimer_ = 0;
+        if (state_ != State::Connected) return;
+        int64_t silent = monotonicMillis() - lastPacketRecvMs_;
+        int tt = negotiatedTimeout_ > 0 ? negotiatedTimeout_
+                                        : cfg_.sessionTimeoutMs;
+        if (silent > tt) {
+            log_.warn("ZK server unresponsive; reconnecting");
+            teardown();
+            scheduleReconnect();
+            return;
+        }
+        sendPing();
+        armPingTimer();
+    });
+}
+
+void ZkClient::sendPing() {
+    Writer w;
+    w.i32(XID_PING);
+    w.i32(OP_PING);
+    sendPacket(w.buf);
+    flush();
+}
+
+void ZkClient::sendPacket(const std::string& body) {
+    uint32_t n = (uint32_t)body.size();
+    char b[4] = {(char)(n >> 24), (char)(n >> 16), (char)(n >> 8), (char)n};
+    outBuf_.append(b, 4);
+    outBuf_ += body;
+}
+
+void ZkClient::flush() {
+    while (!outBuf_.empty() && fd_ >= 0) {
+        ssize_t nw = write(fd_, outBuf_.data(), outBuf_.size());
+        if (nw > 0) {
+            outBuf_.erase(0, (size_t)nw);
+            continue;
+        }
+        if (nw < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) {
+            if (!writeBlocked_) {
+                writeBlocked_ = true;
+                loop_->modFd(fd_, EPOLLIN | EPOLLOUT);
+            }
+            return;
+        }
+        teardown();
+        scheduleReconnect();
+        return;
+    }
+}
+
+void ZkClient::onPacket(const uint8_t* data, size_t len) {
+    Reader r{data, len};
+    int32_t xid = r.i32();
+    int64_t zxid = r.i64();
+    int32_t err = r.i32();
+    if (!r.ok) return;
+    if (zxid > 0) lastZxid_ = zxid;
+
+    if (xid == XID_NOTIFICATION) {
+        int32_t type = r.i32();
+        int32_t st = r.i32();
+        std::string path = r.str();
+        (void)st;
+        if (r.ok && watchCb_) watchCb_(type, path);
+        return;
+    }
+    if (xid == XID_PING) return;
+
+    auto it = pending_.find(xid);
+    if (it == pending_.end()) {
+        log_.warn({{"xid", Json((int64_t)xid)}},
+                  "ZK reply for unknown xid");
+        return;
+    }
+    Pending p = std::move(it->second);
+    pending_.erase(it);
+
+    static const Stat kStat;
+    switch (p.op) {
+    case OP_GETDATA: {
+        if (err != ZOK) {
+            p.dataCb(err, "", kStat);
+            break;
+        }
+        std::string d = r.str();
+        Stat s = r.stat();
+        p.dataCb(r.ok ? ZOK : ZMARSHALLINGERROR, d, s);
+        break;
+    }
+    case OP_GETCHILDREN: {
+        if (err != ZOK) {
+            p.childrenCb(err, {});
+            break;
+        }
+        auto kids = r.strVec();
+        p.childrenCb(r.ok ? ZOK : ZMARSHALLINGERROR, kids);
+        break;
+    }
+    case OP_EXISTS: {
+        if (err != ZOK) {
+            p.statCb(err, kStat);
+            break;
+        }
+        Stat s = r.stat();
+        p.statCb(r.ok ? ZOK : ZMARSHALLINGERROR, s);
+        break;
+    }
+    case OP_CREATE: {
+        if (err != ZOK) {
+            p.stringCb(err, "");
+            break;
+        }
+        p.stringCb(ZOK, r.str());
+        break;
+    }
+    case OP_SETDATA:
+    case OP_DELETE: {
+        p.voidCb(err);
+        break;
+    }
+    default:
+        break;
+    }
+}
+
+/* ---------------- requests ---------------- */
+
+void ZkClient::getData(const std::string& path, bool watch, DataCb cb) {
+    if (state_ != State::Connected) {
+        static const Stat kStat;
+        loop_->defer([cb]() { cb(ZCONNECTIONLOSS, "", kStat); });
+        return;
+    }
+    int32_t xid = nextXid();
+    Writer w;
+    w.i32(xid);
+    w.i32(OP_GETDATA);
+    w.str(path);
+    w.boolean(watch);
+    Pending p;
+    p.xid = xid;
+    p.op = OP_GETDATA;
+    p.dataCb = std::move(cb);
+    pending_[xid] = std::move(p);
+    sendPacket(w.buf);
+    flush();
+}
+
+void ZkClient::getChildren(const std::string& path, bool watch,
+                           ChildrenCb cb) {
+    if (state_ != State::Connected) {
+        loop_->defer([cb]() { cb(ZCONNECTIONLOSS, {}); });
+        return;
+    }
+    int32_t xid = nextXid();
+    Writer w;
+    w.i32(xid);
+    w.i32(OP_GETCHILDREN);
+    w.str(path);
+    w.boolean(watch);
+    Pending p;
+    p.xid = xid;
+    p.op = OP_GETCHILDREN;
+    p.childrenCb = std::move(cb);
+    pending_[xid] = std::move(p);
+    sendPacket(w.buf);
+    flush();
+}
+
+void ZkClient::exists(const std::string& path, bool watch, StatCb cb) {
+    if (state_ != State::Connected) {
+        static const Stat kStat;
+        loop_->defer([cb]() { cb(ZCONNECTIONLOSS, kStat); });
+        return;
+    }
+    int32_t xid = nextXid();
+    Writer w;
+    w.i32(xid);
+    w.i32(OP_EXISTS);
+    w.str(path);
+    w.boolean(watch);
+    Pending p;
+    p.xid = xid;
+    p.op = OP_EXISTS;
+    p.statCb = std::move(cb);
+    pending_[xid] = std::move(p);
+    sendPacket(w.buf);
+    flush();
+}
+
+void ZkClient::create(const std::string& path, const std::string& data,
+                      int32_t flags, StringCb cb) {
+    if (state_ != State::Connected) {
+        loop_->defer([cb]() { cb(ZCONNECTIONLOSS, ""); });
+        return;
+    }
+    int32_t xid = nextXid();
+    Writer w;
+    w.i32(xid);
+    w.i32(OP_CREATE);
+    w.str(path);
+    w.buffer(data);
+    writeOpenAcl(w);
+    w.i32(flags);
+    Pending p;
+    p.xid = xid;
+    p.op = OP_CREATE;
+    p.stringCb = std::move(cb);
+    pending_[xid] = std::move(p);
+    sendPacket(w.buf);
+    flush();
+}
+
+void ZkClient::setData(const std::string& path, const std::string& data,
+                       int32_t version, VoidCb cb) {
+    if (state_ != State::Connected) {
+        loop_->defer([cb]() { cb(ZCONNECTIONLOSS); });
+        return;
+    }
+    int32_t xid = nextXid();
+    Writer w;
+    w.i32(xid);
+    w.i32(OP_SETDATA);
+    w.str(path);
+    w.buffer(data);
+    w.i32(version);
+    Pending p;
+    p.xid = xid;
+    p.op = OP_SETDATA;
+    p.voidCb = std::move(cb);
+    pending_[xid] = std::move(p);
+    sendPacket(w.buf);
+    flush();
+}
+
+void ZkClient::del(const std::string& path, int32_t version, VoidCb cb) {
+    if (state_ != State::Connected) {
+        loop_->defer([cb]() { cb(ZCONNECTIONLOSS); });
+        return;
+    }
+    int32_t xid = nextXid();
+    Writer w;
+    w.i32(xid);
+    w.i32(OP_DELETE);
+    w.str(path);
+    w.i32(version);
+    Pending p;
+    p.xid = xid;
+    p.op = OP_DELETE;
+    p.voidCb = std::move(cb);
+    pending_[xid] = std::move(p);
+    sendPacket(w.buf);
+    flush();
+}
+
+}  // namespace bamd::zk

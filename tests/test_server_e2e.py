@@ -1,0 +1,125 @@
+"""End-to-end wire tests against a real binderd process (file store).
+
+Mirrors the reference's dig-driven black-box suite (test/host.test.js,
+test/service.test.js shape) over UDP and TCP, plus EDNS/truncation and
+metrics behaviors the reference leaves untested.
+"""
+import json
+
+import pytest
+
+from binder_amd.harness import BinderProcess
+
+TREE = {
+    "foo.com": None,
+    "bar.foo.com": None,
+    "web.bar.foo.com": {"type": "host",
+                        "host": {"address": "192.168.0.1"}},
+    "svc.foo.com": {
+        "type": "service",
+        "service": {"srvce": "_http", "proto": "_tcp", "port": 80,
+                    "ttl": 60},
+    },
+    "lb0.svc.foo.com": {"type": "load_balancer",
+                        "load_balancer": {"address": "10.0.1.0"}},
+    "lb1.svc.foo.com": {"type": "load_balancer",
+                        "load_balancer": {"address": "10.0.1.1"}},
+    "big.foo.com": {
+        "type": "service",
+        "service": {"srvce": "_x", "proto": "_tcp", "port": 1},
+    },
+    **{f"m{i}.big.foo.com":
+       {"type": "rr_host", "rr_host": {"address": f"10.8.0.{i}"}}
+       for i in range(40)},
+}
+
+
+@pytest.fixture(scope="module")
+def server(tmp_path_factory):
+    tmp = tmp_path_factory.mktemp("e2e")
+    store = tmp / "tree.json"
+    store.write_text(json.dumps(TREE))
+    srv = BinderProcess(store=f"file:{store}", workdir=tmp,
+                        log_path=str(tmp / "binderd.log"))
+    srv.start()
+    yield srv
+    srv.stop()
+
+
+def test_udp_a(server):
+    r = server.dig("web.bar.foo.com")
+    assert r.status == "NOERROR"
+    assert r.answers[0]["address"] == "192.168.0.1"
+    assert r.answers[0]["ttl"] == 30
+    assert r["aa"] is True
+    assert r["ra"] is False
+
+
+def test_tcp_a(server):
+    r = server.dig("web.bar.foo.com", tcp=True)
+    assert r.status == "NOERROR"
+    assert r.answers[0]["address"] == "192.168.0.1"
+
+
+def test_udp_ptr(server):
+    r = server.dig("1.0.168.192.in-addr.arpa", "PTR")
+    assert r.status == "NOERROR"
+    assert r.answers[0]["target"] == "web.bar.foo.com"
+
+
+def test_udp_srv(server):
+    r = server.dig("_http._tcp.svc.foo.com", "SRV")
+    assert r.status == "NOERROR"
+    assert {a["target"] for a in r.answers} == \
+        {"lb0.svc.foo.com", "lb1.svc.foo.com"}
+    assert {x["address"] for x in r["additionals"]} == \
+        {"10.0.1.0", "10.0.1.1"}
+
+
+def test_refused_unknown(server):
+    assert server.dig("zzz.foo.com").status == "REFUSED"
+    assert server.dig("other.example").status == "REFUSED"
+
+
+def test_notimp_aaaa(server):
+    assert server.dig("web.bar.foo.com", "AAAA").status == "NOTIMP"
+
+
+def test_big_response_truncates_on_udp_and_serves_on_tcp(server):
+    r = server.dig("big.foo.com")
+    assert r["tc"] is True
+    assert r.answers == []
+    r = server.dig("big.foo.com", tcp=True)
+    assert r["tc"] is False
+    assert len(r.answers) == 40
+
+    # EDNS with a big buffer avoids truncation on UDP
+    r = server.dig("big.foo.com", edns=4096)
+    assert r["tc"] is False
+    assert len(r.answers) == 40
+    assert any(x["type"] == "OPT" for x in r["additionals"])
+
+
+def test_metrics_exposed(server):
+    server.dig("web.bar.foo.com")
+    text = server.metrics()
+    assert "binder_requests_completed" in text
+    assert 'type="A"' in text
+    assert "binder_request_latency_seconds_bucket" in text
+    assert "binder_response_size_bytes_count" in text
+
+
+def test_query_id_echoed(server):
+    r = server.dig("web.bar.foo.com", qid=4242)
+    assert r["id"] == 4242
+
+
+def test_malformed_packet_ignored(server):
+    import socket
+    with socket.socket(socket.AF_INET, socket.SOCK_DGRAM) as s:
+        s.settimeout(0.3)
+        s.sendto(b"\x01\x02garbage", (server.host, server.port))
+        with pytest.raises(socket.timeout):
+            s.recvfrom(512)
+    # server still alive
+    assert server.dig("web.bar.foo.com").status == "NOERROR"
