@@ -31,13 +31,21 @@ PY_EXT_SUFFIX := $(shell $(PYTHON)-config --extension-suffix 2>/dev/null || echo
 PY_INCLUDES := $(shell $(PYTHON) -m pybind11 --includes)
 PYMOD := binder_amd/_native$(PY_EXT_SUFFIX)
 
-BINARIES := bin/binderd
+BINARIES := bin/binderd bin/binder-balancer bin/dnsblast
 
 all: $(PYMOD) $(BINARIES)
 
 bin/binderd: $(CORE_OBJS) $(SERVER_OBJS) $(BUILD)/native/server/binderd_main.o
 	@mkdir -p bin
 	$(CXX) $(CXXFLAGS) $^ -o $@ $(LDFLAGS)
+
+bin/binder-balancer: $(CORE_OBJS) $(BUILD)/native/balancer/balancer_main.o
+	@mkdir -p bin
+	$(CXX) $(CXXFLAGS) $^ -o $@ $(LDFLAGS)
+
+bin/dnsblast: $(CORE_OBJS) $(BUILD)/native/bench/dnsblast_main.o
+	@mkdir -p bin
+	$(CXX) $(CXXFLAGS) $^ -o $@ $(LDFLAGS) -lpthread
 
 $(BUILD)/%.o: %.cpp
 	@mkdir -p $(dir $@)

@@ -1,0 +1,264 @@
+#!/usr/bin/env python3
+"""binder-amd flagship benchmark (driver contract entrypoint).
+
+Measures the BASELINE.json headline metric: DNS queries/sec (+ p50/p99
+latency) served from a 10,000-record ZooKeeper-backed tree, with N
+server processes behind the native balancer.
+
+Topology per run (all on one node; "GPU" count N maps to N binderd
+server processes, the reference's only parallelism axis — SURVEY.md
+§2.3: replicated processes behind one balancer, no sharding):
+
+    StubZk (10k records) <-- N x binderd (zk mirror) <-- binder-balancer
+                                                            ^
+                dnsblast (C++ load generator, 4N source IPs)+
+
+A "step" is a fixed batch of Q = 200k*N queries (per-process work fixed
+as N grows => weak scaling). Queries are a uniform A+SRV mix over the
+tree. Note: this workload has no tensor compute — the reference is a
+Node.js DNS server (BASELINE.json "north_star" records the tier
+mismatch); torch is used only for the multi-rank barrier contract.
+"""
+import argparse
+import json
+import os
+import shutil
+import subprocess
+import sys
+import tempfile
+import time
+from pathlib import Path
+
+REPO = Path(__file__).resolve().parent
+sys.path.insert(0, str(REPO))
+
+
+def log(msg):
+    print(f"# bench: {msg}", file=sys.stderr, flush=True)
+
+
+def build_tree(zk, names_path):
+    """10k records: 5000 hosts + 1000 services x 4 members (=10k nodes
+    with payloads), uniform A+SRV query mix."""
+    names = []
+    zk.mkdirp("/com/foo")
+    batch = []
+    for i in range(5000):
+        batch.append((f"/com/foo/h{i}",
+                      {"type": "host",
+                       "host": {"address": f"10.{(i >> 8) & 255}.{i & 255}.1"}}))
+        names.append(f"h{i}.foo.com A")
+    for i in range(1000):
+        batch.append((f"/com/foo/s{i}",
+                      {"type": "service",
+                       "service": {"srvce": "_x", "proto": "_tcp",
+                                   "port": 80, "ttl": 60}}))
+        for j in range(4):
+            batch.append((f"/com/foo/s{i}/m{j}",
+                          {"type": "rr_host",
+                           "rr_host": {"address": f"10.9.{i % 250}.{j + 1}"}}))
+        names.append(f"s{i}.foo.com A")
+        names.append(f"_x._tcp.s{i}.foo.com SRV")
+    for path, obj in batch:
+        zk.put(path, json.dumps(obj).encode())
+    names_path.write_text("\n".join(names))
+    return len(batch)
+
+
+def start_backends(n, tmp, zk_port):
+    from binder_amd.harness import BinderProcess
+    sockdir = tmp / "socks"
+    sockdir.mkdir(exist_ok=True)
+    backends = []
+    for i in range(n):
+        b = BinderProcess(dns_domain="foo.com", datacenter="coal",
+                          store="zk", zk_host="127.0.0.1", zk_port=zk_port,
+                          workdir=tmp, log_level="warn",
+                          balancer_socket=str(sockdir / f"b{i}"),
+                          log_path=str(tmp / f"binderd-{i}.log"))
+        b.start(wait_ready=False)
+        backends.append(b)
+    for b in backends:
+        b.wait_listening(timeout=30)
+        b.wait_ready("h4999.foo.com", timeout=60)
+    return backends, sockdir
+
+
+def start_balancer(tmp, sockdir, port, workers=1):
+    env = dict(os.environ, LOG_LEVEL="warn")
+    proc = subprocess.Popen(
+        [str(REPO / "bin" / "binder-balancer"), "-p", str(port),
+         "-H", "127.0.0.1", "-s", str(sockdir),
+         "-S", str(tmp / "stats.sock"), "-r", "200",
+         "-w", str(workers)],
+        env=env, stdout=open(tmp / "balancer.log", "ab"),
+        stderr=subprocess.STDOUT)
+    return proc
+
+
+def wait_balancer_ready(port, n_backends, tmp, timeout=30):
+    import socket as pysock
+    from binder_amd.digclient import dig
+    deadline = time.time() + timeout
+    while time.time() < deadline:
+        try:
+            with pysock.socket(pysock.AF_UNIX) as s:
+                s.settimeout(1)
+                s.connect(str(tmp / "stats.sock"))
+                stats = json.loads(s.recv(1 << 20).decode())
+            ok = sum(1 for b in stats["backends"] if b["ok"])
+            if ok >= n_backends:
+                r = dig("h0.foo.com", server="127.0.0.1", port=port,
+                        timeout=0.5)
+                if r.status == "NOERROR":
+                    return
+        except (OSError, ValueError):
+            pass
+        time.sleep(0.2)
+    raise TimeoutError("balancer never became ready")
+
+
+def run_blast(port, queries, names_file, threads, window):
+    out = subprocess.run(
+        [str(REPO / "bin" / "dnsblast"), "-s", "127.0.0.1",
+         "-p", str(port), "-n", str(queries), "-c", str(window),
+         "-t", str(threads), "-f", str(names_file), "-B", "127.0.1.1"],
+        capture_output=True, text=True, check=True)
+    return json.loads(out.stdout.strip().splitlines()[-1])
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1,
+                    help="server process count (one per rank/GPU slot)")
+    ap.add_argument("--steps", type=int, default=3)
+    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--queries-per-proc", type=int, default=200_000)
+    ap.add_argument("--window", type=int, default=64)
+    args = ap.parse_args()
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    n = max(args.gpus, world_size)
+
+    dist = None
+    if world_size > 1:
+        import torch.distributed as dist_mod
+        dist = dist_mod
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29511")
+        dist.init_process_group(backend="gloo", rank=rank,
+                                world_size=world_size)
+
+    def barrier():
+        if dist is not None:
+            dist.barrier()
+
+    def cuda_sync():
+        try:
+            import torch
+            if torch.cuda.is_available():
+                torch.cuda.synchronize()
+        except Exception:
+            pass
+
+    q_step = args.queries_per_proc * n
+    ncpu = os.cpu_count() or 8
+    # size the harness to the machine: the load generator + balancer
+    # must not starve the backends on small boxes
+    workers = min(4, n) if ncpu >= 6 * n else 1
+    threads = min(4 * n, max(2, ncpu // 2))
+    result = {}
+
+    tmp = None
+    stack = []
+    try:
+        if rank == 0:
+            from binder_amd.harness import free_port
+            from binder_amd.stubzk import StubZk
+            tmp = Path(tempfile.mkdtemp(prefix="binder-bench-"))
+            names_file = tmp / "names.txt"
+            log(f"starting stub ZK + building 10k-record tree")
+            zk = StubZk().start()
+            stack.append(zk.stop)
+            nrec = build_tree(zk, names_file)
+            log(f"{nrec} records; starting {n} binderd process(es)")
+            backends, sockdir = start_backends(n, tmp, zk.port)
+            stack.append(lambda: [b.stop() for b in backends])
+            bal_port = free_port()
+            bal = start_balancer(tmp, sockdir, bal_port, workers=workers)
+            stack.append(bal.terminate)
+            wait_balancer_ready(bal_port, n, tmp)
+            log(f"balancer ready on :{bal_port}; warmup "
+                f"{args.warmup} x {q_step} queries")
+            for _ in range(args.warmup):
+                run_blast(bal_port, q_step, names_file, threads,
+                          args.window)
+
+        barrier()
+        cuda_sync()
+        t0 = time.perf_counter()
+        last = None
+        if rank == 0:
+            for s in range(args.steps):
+                last = run_blast(bal_port, q_step, names_file, threads,
+                                 args.window)
+                log(f"step {s + 1}/{args.steps}: "
+                    f"{last['qps']:.0f} qps, p99 {last['p99_us']}us")
+        barrier()
+        cuda_sync()
+        elapsed = time.perf_counter() - t0
+
+        # MAX over ranks of the measured wall time
+        if dist is not None:
+            import torch
+            t = torch.tensor([elapsed], dtype=torch.float64)
+            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+            elapsed = float(t.item())
+
+        if rank == 0:
+            total_queries = args.steps * q_step
+            qps = total_queries / elapsed
+            result = {
+                "metric": "dns_queries_per_sec",
+                "value": round(qps, 1),
+                "unit": "queries/s",
+                "n_gpus": n,
+                "steps": args.steps,
+                "warmup": args.warmup,
+                "ms_per_step": round(elapsed * 1000 / args.steps, 3),
+                "higher_is_better": True,
+                "scaling": "weak",
+                "vs_baseline": None,
+                "dtype": "n/a",
+                "data": "synthetic",
+                "config": {
+                    "model": "binder-dns-zk",
+                    "tree_records": 10000,
+                    "query_mix": "A+SRV uniform",
+                    "queries_per_step": q_step,
+                    "global_batch": q_step,
+                    "seq_len": None,
+                    "parallelism":
+                        f"{n} binderd procs behind binder-balancer",
+                    "store": "stub-zk mirror",
+                    "p50_us": last["p50_us"] if last else None,
+                    "p99_us": last["p99_us"] if last else None,
+                    "timeouts_last_step": last["timeouts"] if last else None,
+                },
+            }
+            print(json.dumps(result), flush=True)
+    finally:
+        for fn in reversed(stack):
+            try:
+                fn()
+            except Exception:
+                pass
+        if tmp is not None:
+            shutil.rmtree(tmp, ignore_errors=True)
+        if dist is not None:
+            dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

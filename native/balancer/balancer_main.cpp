@@ -1,0 +1,733 @@
+/*
+ * binder-balancer: native L4 DNS load balancer (mname-balancer
+ * equivalent; SURVEY.md §2 row 11 — the reference submodule is not
+ * vendored, so this is a from-scratch design with the same observable
+ * capabilities):
+ *
+ *   - listens on :port UDP + TCP, fans out to backend binderd processes
+ *     over UNIX sockets discovered in a socket directory (one socket per
+ *     backend; presence = registration, unlink = drain);
+ *   - per-remote-IP affinity: each client IP is pinned to a backend
+ *     (bin/balstat shows the reference tracks backend_t/remote_t AVLs);
+ *   - per-backend health (be_ok): connect failures / PING timeouts mark
+ *     a backend down and its remotes are reassigned;
+ *   - preserves original client address/port/family across the hop
+ *     (bsock1 framing, see protocol.hpp);
+ *   - a stats UNIX socket replaces mdb introspection: connecting dumps
+ *     one JSON object (backends, remotes, counters) and closes —
+ *     consumed by `binder-amd balstat`.
+ *
+ * Flags: -p port (default 53), -H host, -s socket-dir (default
+ * /var/run/binder/sockets), -S stats-socket path, -r rescan interval ms.
+ */
+#include <arpa/inet.h>
+#include <dirent.h>
+#include <netinet/in.h>
+#include <signal.h>
+#include <sys/epoll.h>
+#include <sys/signalfd.h>
+#include <sys/socket.h>
+#include <sys/stat.h>
+#include <sys/un.h>
+#include <unistd.h>
+
+#include <cstring>
+#include <map>
+#include <memory>
+#include <thread>
+#include <set>
+#include <string>
+#include <vector>
+
+#include "../common/json.hpp"
+#include "../common/log.hpp"
+#include "../common/loop.hpp"
+#include "protocol.hpp"
+
+using namespace bamd;
+
+namespace {
+
+struct PendingReply {
+    struct sockaddr_storage src;
+    socklen_t srcLen;
+    int64_t expiresAt;
+    bool tcp = false;
+    int tcpFd = -1;
+};
+
+struct Backend {
+    int id;
+    std::string path;
+    int fd = -1;
+    bool ok = false;
+    std::string in, out;
+    bool writeBlocked = false;
+    uint32_t nextReq = 1;
+    std::map<uint32_t, PendingReply> pending;
+    int64_t lastPongAt = 0;
+    int64_t pingSentAt = 0;
+    uint64_t queries = 0;
+    uint64_t replies = 0;
+    size_t remotes = 0;
+};
+
+struct TcpClient {
+    int fd;
+    std::string in, out;
+    bool writeBlocked = false;
+    bool closed = false;
+    struct sockaddr_storage src;
+    socklen_t srcLen;
+    int backendId = -1;
+};
+
+class Balancer {
+  public:
+    Balancer(EventLoop* loop, Logger log, std::string host, uint16_t port,
+             std::string sockDir, std::string statsPath, int rescanMs,
+             bool reusePort = false)
+        : loop_(loop), log_(std::move(log)), host_(std::move(host)),
+          port_(port), sockDir_(std::move(sockDir)),
+          statsPath_(std::move(statsPath)), rescanMs_(rescanMs),
+          reusePort_(reusePort) {}
+
+    bool start();
+    void stop();
+
+  private:
+    void rescan();
+    void connectBackend(Backend* be);
+    void backendDown(Backend* be);
+    void onBackendEvent(std::shared_ptr<Backend> be, uint32_t ev);
+    void backendFlush(Backend* be);
+    Backend* pickBackend(const std::string& remoteIp);
+    void onUdpReadable();
+    void onTcpAccept();
+    void onTcpClient(std::shared_ptr<TcpClient> c, uint32_t ev);
+    void tcpClientFlush(TcpClient* c);
+    void onStatsAccept();
+    void sweep();
+
+    EventLoop* loop_;
+    Logger log_;
+    std::string host_;
+    uint16_t port_;
+    std::string sockDir_;
+    std::string statsPath_;
+    int rescanMs_;
+    bool reusePort_ = false;
+
+    int udpFd_ = -1, tcpFd_ = -1, statsFd_ = -1;
+    std::map<std::string, std::shared_ptr<Backend>> backends_;  // by path
+    std::map<int, std::shared_ptr<Backend>> backendsById_;
+    std::map<std::string, int> remotes_;  // remote ip -> backend id
+    std::map<int, std::shared_ptr<TcpClient>> tcpClients_;
+    int nextBackendId_ = 1;
+    int rrCursor_ = 0;
+    uint64_t udpQueries_ = 0, udpReplies_ = 0, drops_ = 0;
+
+    static constexpr int64_t kReplyTtlMs = 3000;
+    static constexpr int64_t kPingIntervalMs = 2000;
+    static constexpr int64_t kPingTimeoutMs = 6000;
+};
+
+bool Balancer::start() {
+    /* UDP */
+    bool v6 = host_.empty() || host_.find(':') != std::string::npos;
+    udpFd_ = socket(v6 ? AF_INET6 : AF_INET,
+                    SOCK_DGRAM | SOCK_NONBLOCK | SOCK_CLOEXEC, 0);
+    int one = 1;
+    setsockopt(udpFd_, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+    if (reusePort_)
+        setsockopt(udpFd_, SOL_SOCKET, SO_REUSEPORT, &one, sizeof(one));
+    int sz = 8 << 20;
+    setsockopt(udpFd_, SOL_SOCKET, SO_RCVBUF, &sz, sizeof(sz));
+    setsockopt(udpFd_, SOL_SOCKET, SO_SNDBUF, &sz, sizeof(sz));
+    tcpFd_ = socket(v6 ? AF_INET6 : AF_INET,
+                    SOCK_STREAM | SOCK_NONBLOCK | SOCK_CLOEXEC, 0);
+    setsockopt(tcpFd_, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+    if (reusePort_)
+        setsockopt(tcpFd_, SOL_SOCKET, SO_REUSEPORT, &one, sizeof(one));
+
+    auto doBind = [&](int fd) {
+        if (v6) {
+            int zero = 0;
+            setsockopt(fd, IPPROTO_IPV6, IPV6_V6ONLY,
+                       host_.empty() ? &zero : &one, sizeof(int));
+            struct sockaddr_in6 sa {};
+            sa.sin6_family = AF_INET6;
+            sa.sin6_port = htons(port_);
+            if (host_.empty())
+                sa.sin6_addr = in6addr_any;
+            else
+                inet_pton(AF_INET6, host_.c_str(), &sa.sin6_addr);
+            return bind(fd, (struct sockaddr*)&sa, sizeof(sa)) == 0;
+        }
+        struct sockaddr_in sa {};
+        sa.sin_family = AF_INET;
+        sa.sin_port = htons(port_);
+        inet_pton(AF_INET, host_.c_str(), &sa.sin_addr);
+        return bind(fd, (struct sockaddr*)&sa, sizeof(sa)) == 0;
+    };
+    if (!doBind(udpFd_) || !doBind(tcpFd_) || listen(tcpFd_, 512) != 0) {
+        log_.error({{"port", Json((int)port_)}}, "balancer bind failed");
+        return false;
+    }
+    loop_->addFd(udpFd_, EPOLLIN, [this](uint32_t) { onUdpReadable(); });
+    loop_->addFd(tcpFd_, EPOLLIN, [this](uint32_t) { onTcpAccept(); });
+
+    if (!statsPath_.empty()) {
+        statsFd_ = socket(AF_UNIX, SOCK_STREAM | SOCK_NONBLOCK |
+                          SOCK_CLOEXEC, 0);
+        struct sockaddr_un sa {};
+        sa.sun_family = AF_UNIX;
+        snprintf(sa.sun_path, sizeof(sa.sun_path), "%s",
+                 statsPath_.c_str());
+        unlink(sa.sun_path);
+        if (bind(statsFd_, (struct sockaddr*)&sa, sizeof(sa)) == 0 &&
+            listen(statsFd_, 8) == 0) {
+            loop_->addFd(statsFd_, EPOLLIN,
+                         [this](uint32_t) { onStatsAccept(); });
+        } else {
+            log_.warn({{"path", Json(statsPath_)}},
+                      "could not bind stats socket");
+        }
+    }
+
+    rescan();
+    std::function<void()> tick = [this, &tick]() {};
+    /* periodic rescans + sweeps */
+    auto schedule = std::make_shared<std::function<void()>>();
+    *schedule = [this, schedule]() {
+        rescan();
+        sweep();
+        loop_->addTimer(rescanMs_, *schedule);
+    };
+    loop_->addTimer(rescanMs_, *schedule);
+    log_.info({{"port", Json((int)port_)}, {"dir", Json(sockDir_)}},
+              "balancer started");
+    return true;
+}
+
+void Balancer::stop() {
+    for (int* fd : {&udpFd_, &tcpFd_, &statsFd_}) {
+        if (*fd >= 0) {
+            loop_->delFd(*fd);
+            close(*fd);
+            *fd = -1;
+        }
+    }
+    if (!statsPath_.empty()) unlink(statsPath_.c_str());
+}
+
+void Balancer::rescan() {
+    std::set<std::string> seen;
+    DIR* d = opendir(sockDir_.c_str());
+    if (d != nullptr) {
+        struct dirent* ent;
+        while ((ent = readdir(d)) != nullptr) {
+            std::string name = ent->d_name;
+            if (name == "." || name == "..") continue;
+            std::string full = sockDir_ + "/" + name;
+            struct stat st;
+            if (stat(full.c_str(), &st) != 0 || !S_ISSOCK(st.st_mode))
+                continue;
+            seen.insert(full);
+            if (backends_.count(full) == 0) {
+                auto be = std::make_shared<Backend>();
+                be->id = nextBackendId_++;
+                be->path = full;
+                backends_[full] = be;
+                backendsById_[be->id] = be;
+                log_.info({{"path", Json(full)},
+                           {"id", Json((int64_t)be->id)}},
+                          "backend discovered");
+                connectBackend(be.get());
+            } else {
+                Backend* be = backends_[full].get();
+                if (be->fd < 0) connectBackend(be);  // retry
+            }
+        }
+        closedir(d);
+    }
+    /* removed sockets => drain (main.js:181-193 unlink-on-SIGTERM) */
+    std::vector<std::string> gone;
+    for (auto& [path, be] : backends_)
+        if (seen.count(path) == 0) gone.push_back(path);
+    for (const auto& path : gone) {
+        auto be = backends_[path];
+        log_.info({{"path", Json(path)}}, "backend removed (drained)");
+        backendDown(be.get());
+        backendsById_.erase(be->id);
+        backends_.erase(path);
+    }
+}
+
+void Balancer::connectBackend(Backend* be) {
+    int fd = socket(AF_UNIX, SOCK_STREAM | SOCK_NONBLOCK | SOCK_CLOEXEC, 0);
+    if (fd < 0) return;
+    struct sockaddr_un sa {};
+    sa.sun_family = AF_UNIX;
+    snprintf(sa.sun_path, sizeof(sa.sun_path), "%s", be->path.c_str());
+    int rv = connect(fd, (struct sockaddr*)&sa, sizeof(sa));
+    if (rv != 0 && errno != EINPROGRESS) {
+        close(fd);
+        be->ok = false;
+        return;
+    }
+    be->fd = fd;
+    be->ok = true;  // optimistic; PING confirms
+    be->lastPongAt = monotonicMillis();
+    auto self = backendsById_[be->id];
+    loop_->addFd(fd, EPOLLIN, [this, self](uint32_t ev) {
+        onBackendEvent(self, ev);
+    });
+}
+
+void Balancer::backendDown(Backend* be) {
+    if (be->fd >= 0) {
+        loop_->delFd(be->fd);
+        close(be->fd);
+        be->fd = -1;
+    }
+    be->ok = false;
+    be->in.clear();
+    be->out.clear();
+    be->writeBlocked = false;
+    be->pending.clear();
+    /* unpin remotes so they re-pick a healthy backend */
+    for (auto it = remotes_.begin(); it != remotes_.end();) {
+        if (it->second == be->id)
+            it = remotes_.erase(it);
+        else
+            ++it;
+    }
+    be->remotes = 0;
+}
+
+void Balancer::backendFlush(Backend* be) {
+    while (!be->out.empty() && be->fd >= 0) {
+        ssize_t nw = write(be->fd, be->out.data(), be->out.size());
+        if (nw > 0) {
+            be->out.erase(0, (size_t)nw);
+            continue;
+        }
+        if (nw < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) {
+            if (!be->writeBlocked) {
+                be->writeBlocked = true;
+                loop_->modFd(be->fd, EPOLLIN | EPOLLOUT);
+            }
+            return;
+        }
+        backendDown(be);
+        return;
+    }
+    if (be->writeBlocked && be->fd >= 0) {
+        be->writeBlocked = false;
+        loop_->modFd(be->fd, EPOLLIN);
+    }
+}
+
+void Balancer::onBackendEvent(std::shared_ptr<Backend> be, uint32_t ev) {
+    if (be->fd < 0) return;
+    if (ev & (EPOLLHUP | EPOLLERR)) {
+        log_.warn({{"path", Json(be->path)}}, "backend connection lost");
+        backendDown(be.get());
+        return;
+    }
+    if (ev & EPOLLOUT) backendFlush(be.get());
+    if (!(ev & EPOLLIN)) return;
+
+    char buf[65536];
+    while (be->fd >= 0) {
+        ssize_t nr = read(be->fd, buf, sizeof(buf));
+        if (nr > 0) {
+            be->in.append(buf, (size_t)nr);
+            continue;
+        }
+        if (nr < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) break;
+        log_.warn({{"path", Json(be->path)}}, "backend closed connection");
+        backendDown(be.get());
+        return;
+    }
+
+    while (be->in.size() >= bsock::kHeaderLen) {
+        const uint8_t* h = (const uint8_t*)be->in.data();
+        if (h[0] != bsock::kMagic) {
+            backendDown(be.get());
+            return;
+        }
+        uint8_t type = h[1];
+        uint32_t plen = bsock::getU32(h + 2);
+        if (plen > bsock::kMaxPayload) {
+            backendDown(be.get());
+            return;
+        }
+        if (be->in.size() < bsock::kHeaderLen + plen) break;
+        const uint8_t* payload = h + bsock::kHeaderLen;
+
+        if (type == bsock::FRAME_PONG) {
+            be->lastPongAt = monotonicMillis();
+            be->ok = true;
+        } else if (type == bsock::FRAME_REPLY && plen >= 4) {
+            uint32_t reqId = bsock::getU32(payload);
+            auto it = be->pending.find(reqId);
+            if (it != be->pending.end()) {
+                PendingReply& pr = it->second;
+                const uint8_t* dns = payload + 4;
+                size_t dnsLen = plen - 4;
+                if (pr.tcp) {
+                    auto cit = tcpClients_.find(pr.tcpFd);
+                    if (cit != tcpClients_.end() &&
+                        !cit->second->closed) {
+                        TcpClient* c = cit->second.get();
+                        c->out.push_back((char)(dnsLen >> 8));
+                        c->out.push_back((char)dnsLen);
+                        c->out.append((const char*)dns, dnsLen);
+                        tcpClientFlush(c);
+                    }
+                } else {
+                    sendto(udpFd_, dns, dnsLen, 0,
+                           (struct sockaddr*)&pr.src, pr.srcLen);
+                    udpReplies_++;
+                }
+                be->replies++;
+                be->pending.erase(it);
+            }
+        }
+        be->in.erase(0, bsock::kHeaderLen + plen);
+    }
+}
+
+Backend* Balancer::pickBackend(const std::string& remoteIp) {
+    auto it = remotes_.find(remoteIp);
+    if (it != remotes_.end()) {
+        auto bit = backendsById_.find(it->second);
+        if (bit != backendsById_.end() && bit->second->ok)
+            return bit->second.get();
+        remotes_.erase(it);
+    }
+    /* choose the healthy backend with fewest remotes (stable spread) */
+    Backend* best = nullptr;
+    for (auto& [id, be] : backendsById_) {
+        if (!be->ok || be->fd < 0) continue;
+        if (best == nullptr || be->remotes < best->remotes)
+            best = be.get();
+    }
+    if (best != nullptr) {
+        remotes_[remoteIp] = best->id;
+        best->remotes++;
+    }
+    return best;
+}
+
+static void ipOf(const struct sockaddr_storage& ss, char* out, size_t n,
+                 uint16_t* port, uint8_t* family, uint8_t addr16[16]) {
+    memset(addr16, 0, 16);
+    if (ss.ss_family == AF_INET) {
+        const auto* sa = (const struct sockaddr_in*)&ss;
+        inet_ntop(AF_INET, &sa->sin_addr, out, n);
+        *port = ntohs(sa->sin_port);
+        *family = 4;
+        memcpy(addr16, &sa->sin_addr, 4);
+    } else {
+        const auto* sa = (const struct sockaddr_in6*)&ss;
+        inet_ntop(AF_INET6, &sa->sin6_addr, out, n);
+        *port = ntohs(sa->sin6_port);
+        *family = 6;
+        memcpy(addr16, &sa->sin6_addr, 16);
+    }
+}
+
+void Balancer::onUdpReadable() {
+    constexpr int kBatch = 64;
+    static uint8_t bufs[kBatch][4096];
+    static struct mmsghdr hdrs[kBatch];
+    static struct iovec iovs[kBatch];
+    static struct sockaddr_storage addrs[kBatch];
+
+    while (true) {
+        for (int i = 0; i < kBatch; ++i) {
+            iovs[i] = {bufs[i], sizeof(bufs[i])};
+            memset(&hdrs[i], 0, sizeof(hdrs[i]));
+            hdrs[i].msg_hdr.msg_iov = &iovs[i];
+            hdrs[i].msg_hdr.msg_iovlen = 1;
+            hdrs[i].msg_hdr.msg_name = &addrs[i];
+            hdrs[i].msg_hdr.msg_namelen = sizeof(addrs[i]);
+        }
+        int n = recvmmsg(udpFd_, hdrs, kBatch, 0, nullptr);
+        if (n <= 0) return;
+        std::set<Backend*> touched;
+        for (int i = 0; i < n; ++i) {
+            char ip[48];
+            uint16_t srcPort;
+            uint8_t family;
+            uint8_t addr16[16];
+            ipOf(addrs[i], ip, sizeof(ip), &srcPort, &family, addr16);
+            Backend* be = pickBackend(ip);
+            udpQueries_++;
+            if (be == nullptr) {
+                drops_++;
+                continue;
+            }
+            uint32_t reqId = be->nextReq++;
+            PendingReply pr;
+            pr.src = addrs[i];
+            pr.srcLen = hdrs[i].msg_hdr.msg_namelen;
+            pr.expiresAt = monotonicMillis() + kReplyTtlMs;
+            be->pending[reqId] = pr;
+            std::string payload;
+            payload.reserve(bsock::kQueryHeadLen + hdrs[i].msg_len);
+            bsock::putU32(payload, reqId);
+            payload.push_back((char)family);
+            payload.push_back((char)0);  // udp
+            bsock::putU16(payload, srcPort);
+            payload.append((const char*)addr16, 16);
+            payload.append((const char*)bufs[i], hdrs[i].msg_len);
+            bsock::appendFrame(be->out, bsock::FRAME_QUERY, payload);
+            be->queries++;
+            touched.insert(be);
+        }
+        for (Backend* be : touched) backendFlush(be);
+        if (n < kBatch) return;
+    }
+}
+
+void Balancer::onTcpAccept() {
+    while (true) {
+        struct sockaddr_storage ss;
+        socklen_t sl = sizeof(ss);
+        int fd = accept4(tcpFd_, (struct sockaddr*)&ss, &sl,
+                         SOCK_NONBLOCK | SOCK_CLOEXEC);
+        if (fd < 0) return;
+        auto c = std::make_shared<TcpClient>();
+        c->fd = fd;
+        c->src = ss;
+        c->srcLen = sl;
+        tcpClients_[fd] = c;
+        loop_->addFd(fd, EPOLLIN, [this, c](uint32_t ev) {
+            onTcpClient(c, ev);
+        });
+    }
+}
+
+void Balancer::tcpClientFlush(TcpClient* c) {
+    while (!c->out.empty() && !c->closed) {
+        ssize_t nw = write(c->fd, c->out.data(), c->out.size());
+        if (nw > 0) {
+            c->out.erase(0, (size_t)nw);
+            continue;
+        }
+        if (nw < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) {
+            if (!c->writeBlocked) {
+                c->writeBlocked = true;
+                loop_->modFd(c->fd, EPOLLIN | EPOLLOUT);
+            }
+            return;
+        }
+        c->closed = true;
+        loop_->delFd(c->fd);
+        close(c->fd);
+        tcpClients_.erase(c->fd);
+        return;
+    }
+    if (c->writeBlocked && !c->closed) {
+        c->writeBlocked = false;
+        loop_->modFd(c->fd, EPOLLIN);
+    }
+}
+
+void Balancer::onTcpClient(std::shared_ptr<TcpClient> c, uint32_t ev) {
+    if (c->closed) return;
+    if (ev & (EPOLLHUP | EPOLLERR)) {
+        c->closed = true;
+        loop_->delFd(c->fd);
+        close(c->fd);
+        tcpClients_.erase(c->fd);
+        return;
+    }
+    if (ev & EPOLLOUT) tcpClientFlush(c.get());
+    if (c->closed || !(ev & EPOLLIN)) return;
+
+    char buf[8192];
+    while (true) {
+        ssize_t nr = read(c->fd, buf, sizeof(buf));
+        if (nr > 0) {
+            c->in.append(buf, (size_t)nr);
+            if (c->in.size() > (1 << 20)) nr = 0;  // runaway
+            else continue;
+        }
+        if (nr < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) break;
+        if (nr <= 0) {
+            c->closed = true;
+            loop_->delFd(c->fd);
+            close(c->fd);
+            tcpClients_.erase(c->fd);
+            return;
+        }
+    }
+
+    while (c->in.size() >= 2) {
+        size_t mlen = ((size_t)(uint8_t)c->in[0] << 8) | (uint8_t)c->in[1];
+        if (c->in.size() < 2 + mlen) break;
+        char ip[48];
+        uint16_t srcPort;
+        uint8_t family;
+        uint8_t addr16[16];
+        ipOf(c->src, ip, sizeof(ip), &srcPort, &family, addr16);
+        Backend* be = pickBackend(ip);
+        if (be != nullptr) {
+            uint32_t reqId = be->nextReq++;
+            PendingReply pr;
+            pr.tcp = true;
+            pr.tcpFd = c->fd;
+            pr.srcLen = 0;
+            pr.expiresAt = monotonicMillis() + kReplyTtlMs;
+            memset(&pr.src, 0, sizeof(pr.src));
+            be->pending[reqId] = pr;
+            std::string payload;
+            bsock::putU32(payload, reqId);
+            payload.push_back((char)family);
+            payload.push_back((char)1);  // tcp
+            bsock::putU16(payload, srcPort);
+            payload.append((const char*)addr16, 16);
+            payload.append(c->in.data() + 2, mlen);
+            bsock::appendFrame(be->out, bsock::FRAME_QUERY, payload);
+            be->queries++;
+            backendFlush(be);
+        }
+        c->in.erase(0, 2 + mlen);
+    }
+}
+
+void Balancer::onStatsAccept() {
+    while (true) {
+        int fd = accept4(statsFd_, nullptr, nullptr, SOCK_CLOEXEC);
+        if (fd < 0) return;
+        Json out = Json::object();
+        JsonArray bes;
+        for (auto& [path, be] : backends_) {
+            Json b = Json::object();
+            b.set("id", Json((int64_t)be->id));
+            b.set("path", Json(be->path));
+            b.set("ok", Json(be->ok));
+            b.set("remotes", Json((int64_t)be->remotes));
+            b.set("queries", Json((int64_t)be->queries));
+            b.set("replies", Json((int64_t)be->replies));
+            b.set("pending", Json((int64_t)be->pending.size()));
+            bes.push_back(std::move(b));
+        }
+        out.set("backends", Json(std::move(bes)));
+        JsonArray rms;
+        for (auto& [ip, id] : remotes_) {
+            Json r = Json::object();
+            r.set("addr", Json(ip));
+            r.set("backend", Json((int64_t)id));
+            rms.push_back(std::move(r));
+        }
+        out.set("remotes", Json(std::move(rms)));
+        out.set("udp_queries", Json((int64_t)udpQueries_));
+        out.set("udp_replies", Json((int64_t)udpReplies_));
+        out.set("drops", Json((int64_t)drops_));
+        std::string s = out.dump();
+        s.push_back('\n');
+        ssize_t rv = write(fd, s.data(), s.size());
+        (void)rv;
+        close(fd);
+    }
+}
+
+void Balancer::sweep() {
+    int64_t now = monotonicMillis();
+    for (auto& [path, be] : backends_) {
+        if (be->fd < 0) continue;
+        /* expire stale pendings */
+        for (auto it = be->pending.begin(); it != be->pending.end();) {
+            if (it->second.expiresAt < now)
+                it = be->pending.erase(it);
+            else
+                ++it;
+        }
+        /* health probe */
+        if (now - be->pingSentAt >= kPingIntervalMs) {
+            bsock::appendFrame(be->out, bsock::FRAME_PING, "");
+            be->pingSentAt = now;
+            backendFlush(be.get());
+        }
+        if (now - be->lastPongAt > kPingTimeoutMs) {
+            log_.warn({{"path", Json(be->path)}},
+                      "backend unresponsive; marking down");
+            backendDown(be.get());
+        }
+    }
+}
+
+}  // namespace
+
+int main(int argc, char** argv) {
+    const char* lvl = getenv("LOG_LEVEL");
+    Logger log("binder-balancer",
+               logLevelFromName(lvl ? lvl : "info", LogLevel::Info));
+    std::string host;
+    uint16_t port = 53;
+    std::string dir = "/var/run/binder/sockets";
+    std::string stats;
+    int rescanMs = 1000;
+    int workers = 1;
+    int c;
+    while ((c = getopt(argc, argv, "hp:H:s:S:r:w:")) != -1) {
+        switch (c) {
+        case 'p': port = (uint16_t)atoi(optarg); break;
+        case 'H': host = optarg; break;
+        case 's': dir = optarg; break;
+        case 'S': stats = optarg; break;
+        case 'r': rescanMs = atoi(optarg); break;
+        case 'w': workers = atoi(optarg); break;
+        case 'h':
+        default:
+            fprintf(stderr,
+                    "usage: binder-balancer [-p port] [-H host] "
+                    "[-s socket-dir] [-S stats-socket] [-r rescan-ms] "
+                    "[-w workers]\n");
+            return c == 'h' ? 0 : 1;
+        }
+    }
+    if (workers < 1) workers = 1;
+
+    sigset_t mask;
+    sigemptyset(&mask);
+    sigaddset(&mask, SIGTERM);
+    sigaddset(&mask, SIGINT);
+    sigprocmask(SIG_BLOCK, &mask, nullptr);
+
+    /*
+     * -w > 1: SO_REUSEPORT worker shards, each a full independent
+     * balancer (own epoll loop, backend connections, affinity map).
+     * The kernel hashes client (ip,port) across workers, so per-IP
+     * affinity holds within a worker; -w 1 (default) preserves the
+     * reference's exact single-process per-IP affinity semantics.
+     */
+    std::vector<std::unique_ptr<EventLoop>> loops;
+    std::vector<std::unique_ptr<Balancer>> bals;
+    std::vector<std::thread> threads;
+    for (int w = 0; w < workers; ++w) {
+        loops.emplace_back(std::make_unique<EventLoop>());
+        bals.emplace_back(std::make_unique<Balancer>(
+            loops[w].get(),
+            log.child({{"worker", Json((int64_t)w)}}), host, port, dir,
+            w == 0 ? stats : std::string(), rescanMs, workers > 1));
+        if (!bals[w]->start()) return 1;
+    }
+    for (int w = 0; w < workers; ++w)
+        threads.emplace_back([&, w]() { loops[w]->run(); });
+
+    int sfd = signalfd(-1, &mask, SFD_CLOEXEC);
+    struct signalfd_siginfo si;
+    ssize_t rv = read(sfd, &si, sizeof(si));
+    (void)rv;
+    for (auto& l : loops) l->stop();
+    for (auto& t : threads) t.join();
+    for (auto& b : bals) b->stop();
+    return 0;
+}

@@ -1,0 +1,322 @@
+/*
+ * dnsblast: UDP DNS load generator for the benchmark suite
+ * (BASELINE.md's qps + p50/p99 measurement).
+ *
+ * Each worker thread keeps a sliding window of in-flight queries on its
+ * own socket (sendmmsg/recvmmsg batches), cycling through a name list
+ * loaded from a file. Query wire images are prebuilt; per-send we patch
+ * only the DNS id. Latencies land in log-spaced microsecond buckets for
+ * p50/p99 extraction. Output: ONE JSON line on stdout.
+ *
+ * usage: dnsblast -s <server-ip> -p <port> -n <queries> [-c window]
+ *        [-t threads] [-f names-file] [-B bind-ip-base] [-T timeout-ms]
+ *   names-file: lines of "<name> <qtype>"; default a single test name.
+ *   -B 127.0.0.x base: thread i binds source ip base+i (gives the
+ *      balancer distinct remotes so per-IP affinity spreads load).
+ */
+#include <arpa/inet.h>
+#include <netinet/in.h>
+#include <poll.h>
+#include <sys/socket.h>
+#include <unistd.h>
+
+#include <atomic>
+#include <cstdio>
+#include <cstring>
+#include <fstream>
+#include <random>
+#include <sstream>
+#include <string>
+#include <thread>
+#include <vector>
+
+#include "../common/log.hpp"
+#include "../dns/codec.hpp"
+
+using namespace bamd;
+
+namespace {
+
+struct NameEntry {
+    std::string name;
+    uint16_t qtype;
+};
+
+constexpr int kLatBuckets = 512;
+
+struct ThreadResult {
+    uint64_t sent = 0;
+    uint64_t received = 0;
+    uint64_t timeouts = 0;
+    uint64_t rcodeNoerror = 0;
+    uint64_t rcodeOther = 0;
+    uint64_t answers = 0;
+    std::vector<uint64_t> latBuckets = std::vector<uint64_t>(kLatBuckets);
+};
+
+/* microsecond -> bucket (log-ish: 1us granularity to 256us, then
+ * coarser) */
+inline int latBucket(int64_t us) {
+    if (us < 256) return (int)us;
+    if (us < 256 * 16) return 256 + (int)((us - 256) / 16);  // to 4.3ms
+    if (us < 256 * 16 + 60 * 1000) return 496;  // placeholder (unused)
+    return kLatBuckets - 1;
+}
+
+/* finer mapping: buckets 0..255: 1us each; 256..495: 16us each
+ * (4.1ms); 496..510: 100ms range; 511: overflow */
+inline int latBucket2(int64_t us) {
+    if (us < 256) return (int)us;
+    int64_t v = (us - 256) / 16;
+    if (v < 240) return 256 + (int)v;
+    int64_t w = (us - (256 + 240 * 16)) / 10000;  // 10ms steps
+    if (w < 15) return 496 + (int)w;
+    return 511;
+}
+
+double bucketMidUs(int b) {
+    if (b < 256) return b + 0.5;
+    if (b < 496) return 256 + (b - 256) * 16 + 8;
+    if (b < 511) return 256 + 240 * 16 + (b - 496) * 10000 + 5000;
+    return 200000;
+}
+
+int64_t nowUs() {
+    struct timespec ts;
+    clock_gettime(CLOCK_MONOTONIC, &ts);
+    return (int64_t)ts.tv_sec * 1000000 + ts.tv_nsec / 1000;
+}
+
+struct Config {
+    std::string server = "127.0.0.1";
+    uint16_t port = 1053;
+    uint64_t queries = 100000;
+    int window = 64;
+    int threads = 1;
+    std::string namesFile;
+    std::string bindBase;
+    int timeoutMs = 2000;
+};
+
+void worker(const Config& cfg, int tid,
+            const std::vector<std::vector<uint8_t>>& wires,
+            ThreadResult* out, std::atomic<bool>* abort) {
+    int fd = socket(AF_INET, SOCK_DGRAM, 0);
+    if (fd < 0) return;
+    int sz = 4 << 20;
+    setsockopt(fd, SOL_SOCKET, SO_RCVBUF, &sz, sizeof(sz));
+    setsockopt(fd, SOL_SOCKET, SO_SNDBUF, &sz, sizeof(sz));
+    if (!cfg.bindBase.empty()) {
+        /* bind distinct loopback source ip per thread: base + tid */
+        struct in_addr base;
+        inet_pton(AF_INET, cfg.bindBase.c_str(), &base);
+        uint32_t ip = ntohl(base.s_addr) + (uint32_t)tid;
+        struct sockaddr_in src {};
+        src.sin_family = AF_INET;
+        src.sin_addr.s_addr = htonl(ip);
+        src.sin_port = 0;
+        bind(fd, (struct sockaddr*)&src, sizeof(src));
+    }
+    struct sockaddr_in dst {};
+    dst.sin_family = AF_INET;
+    dst.sin_port = htons(cfg.port);
+    inet_pton(AF_INET, cfg.server.c_str(), &dst.sin_addr);
+    if (connect(fd, (struct sockaddr*)&dst, sizeof(dst)) != 0) {
+        close(fd);
+        return;
+    }
+
+    const int W = cfg.window;
+    std::vector<int64_t> sentAt(W, 0);       // 0 = slot idle
+    std::vector<uint16_t> slotSeq(W, 0);
+    std::vector<size_t> slotName(W, 0);
+    std::mt19937 rng(12345 + tid);
+
+    uint64_t target = cfg.queries;
+    uint64_t launched = 0, completed = 0;
+    const int64_t timeoutUs = (int64_t)cfg.timeoutMs * 1000;
+
+    std::vector<uint8_t> sendBuf(2048);
+    uint8_t recvBuf[4096];
+
+    auto sendSlot = [&](int slot) {
+        size_t ni = rng() % wires.size();
+        const auto& w = wires[ni];
+        /* qid = slot | (seq<<8-ish): encode slot in low bits, seq in
+         * high bits so stale replies are detected */
+        slotSeq[slot]++;
+        uint16_t qid =
+            (uint16_t)((slot & 0xFF) | ((slotSeq[slot] & 0xFF) << 8));
+        memcpy(sendBuf.data(), w.data(), w.size());
+        sendBuf[0] = (uint8_t)(qid >> 8);
+        sendBuf[1] = (uint8_t)qid;
+        ssize_t rv = send(fd, sendBuf.data(), w.size(), 0);
+        if (rv < 0) return false;
+        sentAt[slot] = nowUs();
+        slotName[slot] = ni;
+        out->sent++;
+        launched++;
+        return true;
+    };
+
+    /* prime the window */
+    for (int s = 0; s < W && launched < target; ++s) sendSlot(s);
+
+    struct pollfd pfd {fd, POLLIN, 0};
+    int64_t lastSweep = nowUs();
+    while (completed < target && !abort->load()) {
+        int rv = poll(&pfd, 1, 50);
+        if (rv > 0) {
+            while (true) {
+                ssize_t nr = recv(fd, recvBuf, sizeof(recvBuf),
+                                  MSG_DONTWAIT);
+                if (nr <= 0) break;
+                if (nr < 12) continue;
+                uint16_t qid =
+                    (uint16_t)((recvBuf[0] << 8) | recvBuf[1]);
+                int slot = qid & 0xFF;
+                uint8_t seq = (uint8_t)(qid >> 8);
+                if (slot >= W || sentAt[slot] == 0 ||
+                    (uint8_t)(slotSeq[slot] & 0xFF) != seq)
+                    continue;  // stale/duplicate
+                int64_t lat = nowUs() - sentAt[slot];
+                out->latBuckets[latBucket2(lat)]++;
+                out->received++;
+                uint8_t rcode = recvBuf[3] & 0x0F;
+                if (rcode == 0)
+                    out->rcodeNoerror++;
+                else
+                    out->rcodeOther++;
+                out->answers += (uint64_t)((recvBuf[6] << 8) | recvBuf[7]);
+                sentAt[slot] = 0;
+                completed++;
+                if (launched < target) sendSlot(slot);
+            }
+        }
+        int64_t now = nowUs();
+        if (now - lastSweep > 100000) {  // sweep timeouts every 100ms
+            lastSweep = now;
+            for (int s = 0; s < W; ++s) {
+                if (sentAt[s] != 0 && now - sentAt[s] > timeoutUs) {
+                    out->timeouts++;
+                    completed++;
+                    sentAt[s] = 0;
+                    if (launched < target) sendSlot(s);
+                }
+            }
+        }
+    }
+    close(fd);
+}
+
+}  // namespace
+
+int main(int argc, char** argv) {
+    Config cfg;
+    int c;
+    while ((c = getopt(argc, argv, "hs:p:n:c:t:f:B:T:")) != -1) {
+        switch (c) {
+        case 's': cfg.server = optarg; break;
+        case 'p': cfg.port = (uint16_t)atoi(optarg); break;
+        case 'n': cfg.queries = strtoull(optarg, nullptr, 10); break;
+        case 'c': cfg.window = atoi(optarg); break;
+        case 't': cfg.threads = atoi(optarg); break;
+        case 'f': cfg.namesFile = optarg; break;
+        case 'B': cfg.bindBase = optarg; break;
+        case 'T': cfg.timeoutMs = atoi(optarg); break;
+        case 'h':
+        default:
+            fprintf(stderr,
+                    "usage: dnsblast -s server -p port -n queries "
+                    "[-c window] [-t threads] [-f names-file] "
+                    "[-B bind-base-ip] [-T timeout-ms]\n");
+            return c == 'h' ? 0 : 1;
+        }
+    }
+
+    std::vector<NameEntry> names;
+    if (!cfg.namesFile.empty()) {
+        std::ifstream f(cfg.namesFile);
+        std::string line;
+        while (std::getline(f, line)) {
+            if (line.empty()) continue;
+            std::istringstream ss(line);
+            NameEntry e;
+            std::string t;
+            ss >> e.name >> t;
+            e.qtype = dns::typeFromName(t.empty() ? "A" : t);
+            if (e.qtype == 0) e.qtype = dns::TYPE_A;
+            names.push_back(std::move(e));
+        }
+    }
+    if (names.empty()) names.push_back({"test.foo.com", dns::TYPE_A});
+
+    /* prebuild wire images */
+    std::vector<std::vector<uint8_t>> wires;
+    wires.reserve(names.size());
+    for (const auto& e : names) {
+        dns::Message q;
+        q.header.id = 0;
+        q.header.rd = false;
+        dns::Question qq;
+        qq.name = e.name;
+        qq.qtype = e.qtype;
+        q.questions.push_back(qq);
+        wires.push_back(q.encode(0));
+    }
+
+    std::vector<ThreadResult> results(cfg.threads);
+    std::atomic<bool> abort{false};
+    uint64_t perThread = cfg.queries / cfg.threads;
+
+    int64_t t0 = nowUs();
+    std::vector<std::thread> threads;
+    for (int i = 0; i < cfg.threads; ++i) {
+        Config tcfg = cfg;
+        tcfg.queries = perThread;
+        threads.emplace_back([tcfg, i, &wires, &results, &abort]() {
+            worker(tcfg, i, wires, &results[i], &abort);
+        });
+    }
+    for (auto& t : threads) t.join();
+    int64_t elapsedUs = nowUs() - t0;
+
+    ThreadResult total;
+    for (const auto& r : results) {
+        total.sent += r.sent;
+        total.received += r.received;
+        total.timeouts += r.timeouts;
+        total.rcodeNoerror += r.rcodeNoerror;
+        total.rcodeOther += r.rcodeOther;
+        total.answers += r.answers;
+        for (int b = 0; b < kLatBuckets; ++b)
+            total.latBuckets[b] += r.latBuckets[b];
+    }
+
+    auto pct = [&](double q) {
+        uint64_t n = total.received;
+        if (n == 0) return 0.0;
+        uint64_t want = (uint64_t)(q * (double)n);
+        uint64_t cum = 0;
+        for (int b = 0; b < kLatBuckets; ++b) {
+            cum += total.latBuckets[b];
+            if (cum > want) return bucketMidUs(b);
+        }
+        return bucketMidUs(kLatBuckets - 1);
+    };
+
+    double secs = (double)elapsedUs / 1e6;
+    double qps = secs > 0 ? (double)total.received / secs : 0;
+    printf("{\"sent\": %llu, \"received\": %llu, \"timeouts\": %llu, "
+           "\"noerror\": %llu, \"other_rcode\": %llu, \"answers\": %llu, "
+           "\"elapsed_s\": %.6f, \"qps\": %.1f, "
+           "\"p50_us\": %.1f, \"p90_us\": %.1f, \"p99_us\": %.1f}\n",
+           (unsigned long long)total.sent,
+           (unsigned long long)total.received,
+           (unsigned long long)total.timeouts,
+           (unsigned long long)total.rcodeNoerror,
+           (unsigned long long)total.rcodeOther,
+           (unsigned long long)total.answers, secs, qps, pct(0.50),
+           pct(0.90), pct(0.99));
+    return 0;
+}

@@ -8,6 +8,7 @@
  */
 #pragma once
 
+#include <atomic>
 #include <cstdint>
 #include <functional>
 #include <map>
@@ -60,7 +61,7 @@ class EventLoop {
     };
 
     int epfd_;
-    bool running_ = false;
+    std::atomic<bool> running_{false};
     std::map<int, FdCallback> fds_;
     std::priority_queue<Timer, std::vector<Timer>, std::greater<Timer>> heap_;
     std::map<uint64_t, TimerCallback> timers_;  // id -> cb (absent=cancelled)

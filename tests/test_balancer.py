@@ -1,0 +1,154 @@
+"""Balancer integration: fan-out, affinity, health, drain, stats.
+
+Validates the mname-balancer-equivalent capabilities (SURVEY.md §2 row
+11): backend discovery via socket directory, per-remote affinity,
+original-source preservation across the UNIX-socket hop, drain on socket
+unlink (SIGTERM), and the stats endpoint (balstat replacement).
+"""
+import json
+import os
+import socket
+import subprocess
+import time
+from pathlib import Path
+
+import pytest
+
+from binder_amd import REPO_ROOT
+from binder_amd.digclient import dig
+from binder_amd.harness import BinderProcess, free_port
+
+TREE = {
+    "foo.com": None,
+    "web.foo.com": {"type": "host", "host": {"address": "1.2.3.4"}},
+}
+
+
+def balstat(path):
+    with socket.socket(socket.AF_UNIX) as s:
+        s.settimeout(2)
+        s.connect(str(path))
+        return json.loads(s.recv(1 << 20).decode())
+
+
+@pytest.fixture()
+def cluster(tmp_path):
+    sockdir = tmp_path / "socks"
+    sockdir.mkdir()
+    store = tmp_path / "tree.json"
+    store.write_text(json.dumps(TREE))
+    backends = []
+    for i in range(3):
+        b = BinderProcess(store=f"file:{store}", workdir=tmp_path,
+                          balancer_socket=str(sockdir / f"b{i}"),
+                          log_path=str(tmp_path / f"b{i}.log"))
+        b.start()
+        backends.append(b)
+    port = free_port()
+    stats = tmp_path / "stats.sock"
+    bal = subprocess.Popen(
+        [str(REPO_ROOT / "bin" / "binder-balancer"), "-p", str(port),
+         "-H", "127.0.0.1", "-s", str(sockdir), "-S", str(stats),
+         "-r", "100"],
+        env=dict(os.environ, LOG_LEVEL="warn"),
+        stdout=open(tmp_path / "bal.log", "ab"), stderr=subprocess.STDOUT)
+    deadline = time.time() + 10
+    while time.time() < deadline:
+        try:
+            st = balstat(stats)
+            if sum(1 for b in st["backends"] if b["ok"]) == 3:
+                break
+        except (OSError, ValueError):
+            pass
+        time.sleep(0.1)
+    else:
+        pytest.fail("balancer never saw 3 backends")
+    yield {"port": port, "stats": stats, "backends": backends,
+           "sockdir": sockdir, "bal": bal, "tmp": tmp_path}
+    bal.terminate()
+    bal.wait(timeout=5)
+    for b in backends:
+        b.stop()
+
+
+def test_udp_and_tcp_through_balancer(cluster):
+    r = dig("web.foo.com", port=cluster["port"])
+    assert r.status == "NOERROR"
+    assert r.answers[0]["address"] == "1.2.3.4"
+    r = dig("web.foo.com", port=cluster["port"], tcp=True)
+    assert r.status == "NOERROR"
+
+
+def test_affinity_pins_remote_to_backend(cluster):
+    for _ in range(20):
+        dig("web.foo.com", port=cluster["port"])
+    st = balstat(cluster["stats"])
+    mine = [r for r in st["remotes"] if r["addr"] == "127.0.0.1"]
+    assert len(mine) == 1
+    serving = [b for b in st["backends"] if b["queries"] > 0]
+    # all queries from one remote land on one backend
+    assert len(serving) == 1
+    assert serving[0]["queries"] >= 20
+
+
+def test_distinct_remotes_spread_over_backends(cluster):
+    # source from several loopback ips => affinity spreads
+    for i in range(2, 8):
+        with socket.socket(socket.AF_INET, socket.SOCK_DGRAM) as s:
+            s.bind((f"127.0.0.{i}", 0))
+            s.settimeout(2)
+            from binder_amd import require_native
+            n = require_native()
+            wire = n.encode_message(
+                {"id": i, "questions": [{"name": "web.foo.com",
+                                         "type": "A"}]})
+            s.sendto(wire, ("127.0.0.1", cluster["port"]))
+            data, _ = s.recvfrom(4096)
+            assert n.decode_message(data)["rcode"] == "NOERROR"
+    st = balstat(cluster["stats"])
+    with_remotes = [b for b in st["backends"] if b["remotes"] > 0]
+    assert len(with_remotes) == 3  # least-loaded spread
+
+
+def test_backend_drain_on_sigterm(cluster):
+    victim = cluster["backends"][0]
+    sock_path = Path(victim.cmd[victim.cmd.index("-b") + 1])
+    assert sock_path.exists()
+    victim.sigterm()
+    deadline = time.time() + 5
+    while time.time() < deadline and sock_path.exists():
+        time.sleep(0.05)
+    assert not sock_path.exists(), "SIGTERM must unlink balancer socket"
+    # balancer notices removal and stops routing there
+    deadline = time.time() + 5
+    while time.time() < deadline:
+        st = balstat(cluster["stats"])
+        if len(st["backends"]) == 2:
+            break
+        time.sleep(0.1)
+    else:
+        pytest.fail("balancer kept removed backend")
+    # service continues via remaining backends
+    for _ in range(5):
+        assert dig("web.foo.com",
+                   port=cluster["port"]).status == "NOERROR"
+
+
+def test_original_source_preserved(cluster):
+    """Backend log lines must show the real client address, not the
+    balancer socket (server.js:486-487 capability)."""
+    from binder_amd import require_native
+    n = require_native()
+    with socket.socket(socket.AF_INET, socket.SOCK_DGRAM) as s:
+        s.bind(("127.0.0.77", 0))
+        s.settimeout(2)
+        wire = n.encode_message(
+            {"id": 7, "questions": [{"name": "web.foo.com",
+                                     "type": "A"}]})
+        s.sendto(wire, ("127.0.0.1", cluster["port"]))
+        s.recvfrom(4096)
+    time.sleep(0.2)
+    logs = "".join(
+        (cluster["tmp"] / f"b{i}.log").read_text()
+        for i in range(3) if (cluster["tmp"] / f"b{i}.log").exists())
+    assert '"client":"127.0.0.77"' in logs
