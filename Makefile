@@ -31,7 +31,8 @@ PY_EXT_SUFFIX := $(shell $(PYTHON)-config --extension-suffix 2>/dev/null || echo
 PY_INCLUDES := $(shell $(PYTHON) -m pybind11 --includes)
 PYMOD := binder_amd/_native$(PY_EXT_SUFFIX)
 
-BINARIES := bin/binderd bin/binder-balancer bin/dnsblast
+BINARIES := bin/binderd bin/binder-balancer bin/dnsblast \
+	bin/binder-adjust bin/binder-supervisor bin/zklogcat
 
 all: $(PYMOD) $(BINARIES)
 
@@ -46,6 +47,18 @@ bin/binder-balancer: $(CORE_OBJS) $(BUILD)/native/balancer/balancer_main.o
 bin/dnsblast: $(CORE_OBJS) $(BUILD)/native/bench/dnsblast_main.o
 	@mkdir -p bin
 	$(CXX) $(CXXFLAGS) $^ -o $@ $(LDFLAGS) -lpthread
+
+bin/binder-adjust: $(CORE_OBJS) $(BUILD)/native/adjust/adjust_main.o
+	@mkdir -p bin
+	$(CXX) $(CXXFLAGS) $^ -o $@ $(LDFLAGS)
+
+bin/binder-supervisor: $(CORE_OBJS) $(BUILD)/native/adjust/supervisor_main.o
+	@mkdir -p bin
+	$(CXX) $(CXXFLAGS) $^ -o $@ $(LDFLAGS)
+
+bin/zklogcat: $(CORE_OBJS) $(BUILD)/native/zklog/zklogcat_main.o
+	@mkdir -p bin
+	$(CXX) $(CXXFLAGS) $^ -o $@ $(LDFLAGS)
 
 $(BUILD)/%.o: %.cpp
 	@mkdir -p $(dir $@)

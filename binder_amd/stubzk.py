@@ -16,10 +16,13 @@ thread-safe and fires watches inline.
 """
 from __future__ import annotations
 
+import os
 import selectors
 import socket
 import struct
 import threading
+import time
+import zlib
 from typing import Dict, Optional, Set, Tuple
 
 # op codes
@@ -65,10 +68,20 @@ def _parent(path: str) -> str:
 
 class StubZk:
     def __init__(self, host: str = "127.0.0.1", port: int = 0,
-                 session_timeout_ms: int = 30000):
+                 session_timeout_ms: int = 30000,
+                 txnlog_dir: Optional[str] = None):
         self.host = host
         self._port = port
         self.session_timeout_ms = session_timeout_ms
+        # Optional on-disk transaction log in the real ZooKeeper
+        # FileTxnLog v2 format (magic ZKLG) — lets zklogcat be tested
+        # against logs this stub writes.
+        self._txnlog = None
+        if txnlog_dir:
+            os.makedirs(txnlog_dir, exist_ok=True)
+            self._txnlog = open(os.path.join(txnlog_dir, "log.1"), "wb")
+            self._txnlog.write(struct.pack(">iiq", 0x5A4B4C47, 2, 0))
+            self._txnlog.flush()
         self._lock = threading.RLock()
         self._nodes: Dict[str, _Node] = {"/": _Node()}
         self._conns: Set[_Conn] = set()
@@ -118,6 +131,9 @@ class StubZk:
                     pass
                 self._listener.close()
                 self._listener = None
+            if self._txnlog:
+                self._txnlog.close()
+                self._txnlog = None
         self._sel.close()
 
     def _wake(self):
@@ -218,6 +234,22 @@ class StubZk:
 
     # ------------- tree internals (lock held) -------------
 
+    def _txn(self, ttype: int, body: bytes, client_id: int = 0):
+        if self._txnlog is None:
+            return
+        hdr = struct.pack(">qiqqi", client_id, 0, self._zxid,
+                          int(time.time() * 1000), ttype)
+        txn = hdr + body
+        crc = zlib.adler32(txn) & 0xFFFFFFFF
+        self._txnlog.write(struct.pack(">qi", crc, len(txn)) + txn +
+                           b"\x42")
+        self._txnlog.flush()
+
+    @staticmethod
+    def _jstr(s) -> bytes:
+        b = s.encode() if isinstance(s, str) else s
+        return struct.pack(">i", len(b)) + b
+
     def _do_create(self, path: str, data: bytes) -> int:
         if path in self._nodes:
             return ZNODEEXISTS
@@ -226,6 +258,10 @@ class StubZk:
         if pn is None:
             return ZNONODE
         self._zxid += 1
+        self._txn(1, self._jstr(path) + self._jstr(data) +
+                  struct.pack(">i", 1) + struct.pack(">i", 31) +
+                  self._jstr("world") + self._jstr("anyone") +
+                  b"\x00" + struct.pack(">i", pn.cversion + 1))
         self._nodes[path] = _Node(data)
         pn.children.add(path[path.rfind("/") + 1:])
         pn.cversion += 1
@@ -240,6 +276,8 @@ class StubZk:
         self._zxid += 1
         n.data = data
         n.version += 1
+        self._txn(5, self._jstr(path) + self._jstr(data) +
+                  struct.pack(">i", n.version))
         self._fire(path, EV_DATA)
         return ZOK
 
@@ -250,6 +288,7 @@ class StubZk:
         if n.children:
             return ZNOTEMPTY
         self._zxid += 1
+        self._txn(2, self._jstr(path))
         del self._nodes[path]
         parent = _parent(path)
         pn = self._nodes.get(parent)
@@ -395,6 +434,9 @@ class StubZk:
             self._next_session += 1
             self._valid_sessions.add(session_id)
             self.stats["sessions"] += 1
+            self._zxid += 1
+            self._txn(-10, struct.pack(">i", self.session_timeout_ms),
+                      client_id=session_id)
         c.session_id = session_id
         c.handshaken = True
         neg = min(timeout or self.session_timeout_ms,
@@ -418,6 +460,8 @@ class StubZk:
         self.stats["ops"] += 1
 
         if op == OP_CLOSE:
+            self._zxid += 1
+            self._txn(-11, b"", client_id=c.session_id)
             self._reply(c, xid, ZOK)
             self._try_flush(c)
             self._close_conn(c)

@@ -199,6 +199,8 @@ int main(int argc, char** argv) {
                 ? rj.get("datacenterName").asString() : dcName;
         ropts.dnsDomain = rj.get("dnsDomain").isString()
                               ? rj.get("dnsDomain").asString() : dnsDomain;
+        if (rj.get("upstreamPort").isNumber())
+            ropts.upstreamPort = (uint16_t)rj.get("upstreamPort").asInt();
         ropts.config = rj;
         recursion = std::make_unique<Recursion>(&loop, log, ropts, store);
         recursion->init();

@@ -253,7 +253,7 @@ void Recursion::sendNext(const std::shared_ptr<Upstream>& up,
         up->hosts.erase(up->hosts.begin());
         struct sockaddr_in sa {};
         sa.sin_family = AF_INET;
-        sa.sin_port = htons(53);
+        sa.sin_port = htons(opts_.upstreamPort);
         if (inet_pton(AF_INET, host.c_str(), &sa.sin_addr) != 1) {
             up->errors++;
             continue;

@@ -40,6 +40,7 @@ struct RecursionOptions {
     std::string regionName;
     std::string datacenterName;
     std::string dnsDomain;
+    uint16_t upstreamPort = 53;  // tests point this at ephemeral ports
     Json config;  // the whole `recursion` config block
 };
 

@@ -1,0 +1,76 @@
+"""zklogcat: decode ZooKeeper FileTxnLog written by the stub server."""
+import json
+import subprocess
+
+from binder_amd import REPO_ROOT
+from binder_amd.stubzk import StubZk
+
+
+def run_zklogcat(args):
+    out = subprocess.run([str(REPO_ROOT / "bin" / "zklogcat")] + args,
+                         capture_output=True, text=True)
+    assert out.returncode == 0, out.stderr
+    return [json.loads(line) for line in out.stdout.splitlines()]
+
+
+def test_decode_create_set_delete(tmp_path):
+    logdir = tmp_path / "txnlog"
+    zk = StubZk(txnlog_dir=str(logdir)).start()
+    try:
+        zk.mkdirp("/com/foo")
+        zk.put("/com/foo/web", b'{"type":"host"}')
+        zk.set("/com/foo/web", b'{"type":"host","host":{}}')
+        zk.delete("/com/foo/web")
+    finally:
+        zk.stop()
+
+    txns = run_zklogcat([str(logdir / "log.1")])
+    types = [t["type"] for t in txns]
+    assert types == ["create", "create", "create", "setData", "delete"]
+    assert txns[2]["path"] == "/com/foo/web"
+    assert txns[3]["version"] == 1
+    assert txns[4]["path"] == "/com/foo/web"
+    assert all(t["zxid"] > 0 for t in txns)
+
+
+def test_hex_data_and_sessions(tmp_path):
+    logdir = tmp_path / "txnlog"
+    zk = StubZk(txnlog_dir=str(logdir)).start()
+    try:
+        zk.put("/x", b"\x01\x02")
+    finally:
+        zk.stop()
+    txns = run_zklogcat(["-d", str(logdir / "log.1")])
+    assert txns[-1]["data"] == "0102"
+
+    out = run_zklogcat(["-S", str(logdir / "log.1")])
+    # -S appends session summaries (none opened via wire here, but the
+    # flag path must not crash); all rows are valid json
+    assert isinstance(out, list)
+
+
+def test_session_txns_from_wire_client(tmp_path):
+    """A real wire session (native ZK client via binderd would do, but a
+    raw socket handshake suffices) produces createSession txns."""
+    import socket
+    import struct
+    logdir = tmp_path / "txnlog"
+    zk = StubZk(txnlog_dir=str(logdir)).start()
+    try:
+        s = socket.socket()
+        s.connect(("127.0.0.1", zk.port))
+        req = struct.pack(">iqiq", 0, 0, 30000, 0) + \
+            struct.pack(">i", 16) + b"\x00" * 16 + b"\x00"
+        s.sendall(struct.pack(">i", len(req)) + req)
+        s.recv(4096)
+        s.close()
+        import time
+        time.sleep(0.3)
+    finally:
+        zk.stop()
+    txns = run_zklogcat(["-S", str(logdir / "log.1")])
+    created = [t for t in txns if t.get("type") == "createSession"]
+    assert len(created) == 1
+    assert created[0]["timeout_ms"] == 30000
+    sessions = [t for t in txns if "open" in t]
+    assert len(sessions) >= 1
