@@ -8,6 +8,7 @@
 #include <cstring>
 
 #include "../balancer/protocol.hpp"
+#include "../common/probes.hpp"
 
 namespace bamd {
 
@@ -196,14 +197,22 @@ void DnsServer::stop() {
 
 /* ---------------- query pipeline ---------------- */
 
+static inline int64_t nowUs() {
+    struct timespec ts;
+    clock_gettime(CLOCK_MONOTONIC, &ts);
+    return (int64_t)ts.tv_sec * 1000000 + ts.tv_nsec / 1000;
+}
+
 bool DnsServer::process(const uint8_t* data, size_t len, bool udp,
                         const ClientInfo& ci, std::vector<uint8_t>& out,
                         std::function<void(std::vector<uint8_t>)>
                             asyncReply) {
-    int64_t start = monotonicMillis();
+    int64_t start = nowUs();
+    BAMD_PROBE2(op_req_start, len, (int)udp);
     auto parsed = Message::decode(data, len);
     if (!parsed) return true;  // drop malformed (out empty)
     if (parsed->header.qr) return true;  // ignore responses
+    int64_t tParse = nowUs();
 
     Message& query = *parsed;
     size_t limit = 0;
@@ -218,6 +227,7 @@ bool DnsServer::process(const uint8_t* data, size_t len, bool udp,
 
     Message resp;
     QueryResult qr = engine_->handle(query, resp);
+    int64_t tResolve = nowUs();
 
     if (qr.action == QueryResult::Action::Recurse &&
         recursion_ != nullptr) {
@@ -236,23 +246,30 @@ bool DnsServer::process(const uint8_t* data, size_t len, bool udp,
                 auto wire = respHeap->encode(lim);
                 size_t n = wire.size();
                 asyncReply(std::move(wire));
-                afterQuery(*queryHeap, *respHeap, qr, ciCopy, n,
-                           start);
+                afterQuery(*queryHeap, *respHeap, qr, ciCopy, n, start,
+                           QueryTimers{});
             });
         return false;
     }
 
     if (hasEdns) resp.additionals.push_back(Record::OPT(1400));
     out = resp.encode(limit);
-    afterQuery(query, resp, qr, ci, out.size(), start);
+    QueryTimers tm;
+    tm.parseUs = tParse - start;
+    tm.resolveUs = tResolve - tParse;
+    tm.encodeUs = nowUs() - tResolve;
+    afterQuery(query, resp, qr, ci, out.size(), start, tm);
     return true;
 }
 
 void DnsServer::afterQuery(const Message& query, const Message& resp,
                            const QueryResult& qr, const ClientInfo& ci,
-                           size_t bytesSent, int64_t startMs) {
+                           size_t bytesSent, int64_t startUs,
+                           const QueryTimers& tm) {
     ++served_;
-    int64_t lat = monotonicMillis() - startMs;
+    int64_t latUs = nowUs() - startUs;
+    int64_t lat = latUs / 1000;
+    BAMD_PROBE2(op_req_done, bytesSent, (int)resp.header.rcode);
 
     const char* qtype = query.questions.empty()
                             ? nullptr
@@ -260,7 +277,7 @@ void DnsServer::afterQuery(const Message& query, const Message& resp,
     if (qtype != nullptr) {
         std::string label = std::string("type=\"") + qtype + "\"";
         reqCounter_->increment(label);
-        latHist_->observe(label, (double)lat / 1000.0);
+        latHist_->observe(label, (double)latUs / 1e6);
         sizeHist_->observe(label, (double)bytesSent);
     }
 
@@ -313,6 +330,15 @@ void DnsServer::afterQuery(const Message& query, const Message& resp,
         o["additional"] = Json(std::move(adds));
     }
     o["latency"] = Json(lat);
+    {
+        /* phase timers, the query._times equivalent
+         * (server.js:476-483) */
+        JsonObject t;
+        t["parse_us"] = Json(tm.parseUs);
+        t["resolve_us"] = Json(tm.resolveUs);
+        t["encode_us"] = Json(tm.encodeUs);
+        o["timers"] = Json(std::move(t));
+    }
     log_.log(lv, "DNS query", o);
 }
 

@@ -38,6 +38,12 @@ struct ServerOptions {
     std::string balancerSocket;  // unix socket path; empty = none
 };
 
+struct QueryTimers {
+    int64_t parseUs = 0;
+    int64_t resolveUs = 0;
+    int64_t encodeUs = 0;
+};
+
 struct ClientInfo {
     char address[48] = "";  // presentation form
     uint16_t port = 0;
@@ -108,7 +114,8 @@ class DnsServer {
 
     void afterQuery(const dns::Message& query, const dns::Message& resp,
                     const QueryResult& qr, const ClientInfo& ci,
-                    size_t bytesSent, int64_t startNanos);
+                    size_t bytesSent, int64_t startUs,
+                    const QueryTimers& tm);
 
     void tcpFlush(TcpConn* c);
     void tcpMaybeClose(TcpConn* c);

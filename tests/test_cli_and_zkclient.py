@@ -1,0 +1,73 @@
+"""CLI subcommands + Python ZK client against the stub server."""
+import json
+
+import pytest
+
+from binder_amd import cli
+from binder_amd.stubzk import StubZk
+from binder_amd.zkclient import ZkConn, ZkError
+
+
+@pytest.fixture()
+def zk():
+    z = StubZk().start()
+    yield z
+    z.stop()
+
+
+def test_zkclient_crud(zk):
+    with ZkConn("127.0.0.1", zk.port) as c:
+        c.mkdirp("/com/foo/bar")
+        assert c.exists("/com/foo/bar")
+        c.set("/com/foo/bar", b'{"x":1}')
+        assert c.get("/com/foo/bar") == b'{"x":1}'
+        c.create("/com/foo/baz", b"null")
+        assert sorted(c.children("/com/foo")) == ["bar", "baz"]
+        c.rmr("/com")
+        assert not c.exists("/com")
+
+
+def test_zkclient_errors(zk):
+    with ZkConn("127.0.0.1", zk.port) as c:
+        with pytest.raises(ZkError) as e:
+            c.get("/nope")
+        assert e.value.code == -101
+        c.create("/a")
+        with pytest.raises(ZkError) as e:
+            c.create("/a")
+        assert e.value.code == -110
+
+
+def test_register_then_resolve(zk, tmp_path):
+    """`binder-amd register` writes the registrar layout; a binderd
+    mirror must then serve the _dns._udp SRV + A for it."""
+    rc = cli.main(["register", "binder.coal.foo.com", "10.77.77.1",
+                   "-i", "binder0", "--zk-host", "127.0.0.1",
+                   "--zk-port", str(zk.port)])
+    assert rc == 0
+    data = json.loads(zk.get("/com/foo/coal/binder").decode())
+    assert data["type"] == "service"
+    assert data["service"]["srvce"] == "_dns"
+
+    from binder_amd.harness import BinderProcess
+    srv = BinderProcess(store="zk", zk_host="127.0.0.1",
+                        zk_port=zk.port, workdir=tmp_path)
+    srv.start()
+    try:
+        r = srv.wait_ready("_dns._udp.binder.coal.foo.com", qtype="SRV")
+        assert r.answers[0]["port"] == 53
+        assert r["additionals"][0]["address"] == "10.77.77.1"
+        r = srv.dig("binder0.binder.coal.foo.com")
+        assert r.answers[0]["address"] == "10.77.77.1"
+    finally:
+        srv.stop()
+
+
+def test_cli_zk_ops(zk, capsys):
+    cli.main(["zk", "mkdirp", "/x/y", "--zk-port", str(zk.port)])
+    cli.main(["zk", "set", "/x/y", '{"k":1}', "--zk-port", str(zk.port)])
+    cli.main(["zk", "get", "/x/y", "--zk-port", str(zk.port)])
+    out = capsys.readouterr().out
+    assert '{"k":1}' in out
+    cli.main(["zk", "ls", "/x", "--zk-port", str(zk.port)])
+    assert "y" in capsys.readouterr().out
