@@ -162,9 +162,22 @@ namespace {
 
 struct Encoder {
     std::vector<uint8_t> buf;
-    // Maps a (sub)name in presentation form to its wire offset, for
-    // compression pointers (only offsets < 0x4000 are representable).
-    std::map<std::string, uint16_t> offsets;
+    // Compression table: (suffix view, wire offset). Views reference
+    // the record name strings, which outlive the encode; linear scan
+    // beats a map for the few dozen names in a response, and avoids
+    // all substring allocations on the hot path.
+    static constexpr size_t kMaxOffsets = 64;
+    std::pair<std::string_view, uint16_t> offsets[kMaxOffsets];
+    size_t nOffsets = 0;
+
+    const uint16_t* findOffset(std::string_view sv) const {
+        for (size_t i = 0; i < nOffsets; ++i)
+            if (offsets[i].first == sv) return &offsets[i].second;
+        return nullptr;
+    }
+    void rememberOffset(std::string_view sv, uint16_t off) {
+        if (nOffsets < kMaxOffsets) offsets[nOffsets++] = {sv, off};
+    }
 
     void u8(uint8_t v) { buf.push_back(v); }
     void u16(uint16_t v) {
@@ -183,28 +196,29 @@ struct Encoder {
     }
 
     /* Encode a name with compression. `name` presentation form, no
-     * trailing dot. */
-    void encodeName(const std::string& name, bool compress = true) {
-        size_t pos = 0;
-        std::string rest = name;
+     * trailing dot. NOTE: the view must stay valid for the whole
+     * encode (record name strings do). */
+    void encodeName(std::string_view rest, bool compress = true) {
         while (!rest.empty()) {
             if (compress) {
-                auto it = offsets.find(rest);
-                if (it != offsets.end()) {
-                    u16((uint16_t)(0xC000 | it->second));
+                const uint16_t* off = findOffset(rest);
+                if (off != nullptr) {
+                    u16((uint16_t)(0xC000 | *off));
                     return;
                 }
             }
             if (buf.size() < 0x4000)
-                offsets[rest] = (uint16_t)buf.size();
+                rememberOffset(rest, (uint16_t)buf.size());
             size_t dot = rest.find('.');
-            std::string label =
-                dot == std::string::npos ? rest : rest.substr(0, dot);
-            rest = dot == std::string::npos ? "" : rest.substr(dot + 1);
-            if (label.size() > 63) label.resize(63);
-            u8((uint8_t)label.size());
-            raw(label.data(), label.size());
-            (void)pos;
+            std::string_view label =
+                dot == std::string_view::npos ? rest
+                                              : rest.substr(0, dot);
+            rest = dot == std::string_view::npos
+                       ? std::string_view()
+                       : rest.substr(dot + 1);
+            size_t n = label.size() > 63 ? 63 : label.size();
+            u8((uint8_t)n);
+            raw(label.data(), n);
         }
         u8(0);
     }
