@@ -130,6 +130,18 @@ class Balancer {
     static constexpr int64_t kReplyTtlMs = 3000;
     static constexpr int64_t kPingIntervalMs = 2000;
     static constexpr int64_t kPingTimeoutMs = 6000;
+
+    /* per-instance batch I/O arenas — workers are threads, so these
+     * must NOT be static (a shared-static race here collapsed
+     * multi-worker throughput) */
+    static constexpr int kBatch = 64;
+    uint8_t rxBufs_[kBatch][4096];
+    struct mmsghdr rxHdrs_[kBatch];
+    struct iovec rxIovs_[kBatch];
+    struct sockaddr_storage rxAddrs_[kBatch];
+    struct mmsghdr replyHdrs_[kBatch];
+    struct iovec replyIovs_[kBatch];
+    struct sockaddr_storage replyAddrs_[kBatch];
 };
 
 bool Balancer::start() {
@@ -355,10 +367,10 @@ void Balancer::onBackendEvent(std::shared_ptr<Backend> be, uint32_t ev) {
     /* Walk complete frames without per-frame erase; UDP replies are
      * batched into sendmmsg (a syscall per reply was the balancer's
      * top cost at high QPS). */
-    constexpr int kReplyBatch = 64;
-    static struct mmsghdr rh[kReplyBatch];
-    static struct iovec riov[kReplyBatch];
-    static struct sockaddr_storage raddr[kReplyBatch];
+    constexpr int kReplyBatch = kBatch;
+    struct mmsghdr* rh = replyHdrs_;
+    struct iovec* riov = replyIovs_;
+    struct sockaddr_storage* raddr = replyAddrs_;
     int nReply = 0;
     auto flushReplies = [&]() {
         int sent = 0;
@@ -469,11 +481,10 @@ static void ipOf(const struct sockaddr_storage& ss, char* out, size_t n,
 }
 
 void Balancer::onUdpReadable() {
-    constexpr int kBatch = 64;
-    static uint8_t bufs[kBatch][4096];
-    static struct mmsghdr hdrs[kBatch];
-    static struct iovec iovs[kBatch];
-    static struct sockaddr_storage addrs[kBatch];
+    uint8_t (*bufs)[4096] = rxBufs_;
+    struct mmsghdr* hdrs = rxHdrs_;
+    struct iovec* iovs = rxIovs_;
+    struct sockaddr_storage* addrs = rxAddrs_;
 
     while (true) {
         for (int i = 0; i < kBatch; ++i) {
