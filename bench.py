@@ -166,8 +166,8 @@ def main():
     ncpu = os.cpu_count() or 8
     # size the harness to the machine: the load generator + balancer
     # must not starve the backends on small boxes
-    workers = min(4, n) if ncpu >= 6 * n else 1
-    threads = min(4 * n, max(2, ncpu // 2))
+    workers = min(8, n) if ncpu >= 6 * n else 1
+    threads = min(6 * n, max(2, ncpu // 2))
     result = {}
 
     tmp = None

@@ -24,6 +24,7 @@
 #include <cstdio>
 #include <cstring>
 #include <fstream>
+#include <array>
 #include <random>
 #include <sstream>
 #include <string>
@@ -136,31 +137,63 @@ void worker(const Config& cfg, int tid,
     uint64_t launched = 0, completed = 0;
     const int64_t timeoutUs = (int64_t)cfg.timeoutMs * 1000;
 
-    std::vector<uint8_t> sendBuf(2048);
-    uint8_t recvBuf[4096];
+    /* batched RX (recvmmsg) and TX (sendmmsg): syscall count, not
+     * packet handling, bounds the generator at high QPS */
+    constexpr int kRxBatch = 64;
+    std::vector<std::array<uint8_t, 2048>> rxBufs(kRxBatch);
+    std::vector<struct mmsghdr> rxHdrs(kRxBatch);
+    std::vector<struct iovec> rxIovs(kRxBatch);
+    std::vector<std::array<uint8_t, 2048>> txBufs(kRxBatch);
+    std::vector<struct mmsghdr> txHdrs(kRxBatch);
+    std::vector<struct iovec> txIovs(kRxBatch);
+    std::vector<int> freeSlots;
+    freeSlots.reserve(W);
 
-    auto sendSlot = [&](int slot) {
-        size_t ni = rng() % wires.size();
-        const auto& w = wires[ni];
-        /* qid = slot | (seq<<8-ish): encode slot in low bits, seq in
-         * high bits so stale replies are detected */
-        slotSeq[slot]++;
-        uint16_t qid =
-            (uint16_t)((slot & 0xFF) | ((slotSeq[slot] & 0xFF) << 8));
-        memcpy(sendBuf.data(), w.data(), w.size());
-        sendBuf[0] = (uint8_t)(qid >> 8);
-        sendBuf[1] = (uint8_t)qid;
-        ssize_t rv = send(fd, sendBuf.data(), w.size(), 0);
-        if (rv < 0) return false;
-        sentAt[slot] = nowUs();
-        slotName[slot] = ni;
-        out->sent++;
-        launched++;
-        return true;
+    auto batchSend = [&](std::vector<int>& slots) {
+        int nTx = 0;
+        for (int slot : slots) {
+            if (launched >= target) break;
+            size_t ni = rng() % wires.size();
+            const auto& w = wires[ni];
+            slotSeq[slot]++;
+            uint16_t qid = (uint16_t)((slot & 0xFF) |
+                                      ((slotSeq[slot] & 0xFF) << 8));
+            memcpy(txBufs[nTx].data(), w.data(), w.size());
+            txBufs[nTx][0] = (uint8_t)(qid >> 8);
+            txBufs[nTx][1] = (uint8_t)qid;
+            txIovs[nTx] = {txBufs[nTx].data(), w.size()};
+            memset(&txHdrs[nTx], 0, sizeof(txHdrs[nTx]));
+            txHdrs[nTx].msg_hdr.msg_iov = &txIovs[nTx];
+            txHdrs[nTx].msg_hdr.msg_iovlen = 1;
+            sentAt[slot] = nowUs();
+            slotName[slot] = ni;
+            out->sent++;
+            launched++;
+            nTx++;
+            if (nTx == kRxBatch) {
+                int done = 0;
+                while (done < nTx) {
+                    int rv = sendmmsg(fd, txHdrs.data() + done,
+                                      nTx - done, 0);
+                    if (rv <= 0) break;
+                    done += rv;
+                }
+                nTx = 0;
+            }
+        }
+        int done = 0;
+        while (done < nTx) {
+            int rv = sendmmsg(fd, txHdrs.data() + done, nTx - done, 0);
+            if (rv <= 0) break;
+            done += rv;
+        }
+        slots.clear();
     };
 
     /* prime the window */
-    for (int s = 0; s < W && launched < target; ++s) sendSlot(s);
+    for (int s = 0; s < W && (uint64_t)s < target; ++s)
+        freeSlots.push_back(s);
+    batchSend(freeSlots);
 
     struct pollfd pfd {fd, POLLIN, 0};
     int64_t lastSweep = nowUs();
@@ -168,29 +201,42 @@ void worker(const Config& cfg, int tid,
         int rv = poll(&pfd, 1, 50);
         if (rv > 0) {
             while (true) {
-                ssize_t nr = recv(fd, recvBuf, sizeof(recvBuf),
-                                  MSG_DONTWAIT);
+                for (int i = 0; i < kRxBatch; ++i) {
+                    rxIovs[i] = {rxBufs[i].data(), rxBufs[i].size()};
+                    memset(&rxHdrs[i], 0, sizeof(rxHdrs[i]));
+                    rxHdrs[i].msg_hdr.msg_iov = &rxIovs[i];
+                    rxHdrs[i].msg_hdr.msg_iovlen = 1;
+                }
+                int nr = recvmmsg(fd, rxHdrs.data(), kRxBatch,
+                                  MSG_DONTWAIT, nullptr);
                 if (nr <= 0) break;
-                if (nr < 12) continue;
-                uint16_t qid =
-                    (uint16_t)((recvBuf[0] << 8) | recvBuf[1]);
-                int slot = qid & 0xFF;
-                uint8_t seq = (uint8_t)(qid >> 8);
-                if (slot >= W || sentAt[slot] == 0 ||
-                    (uint8_t)(slotSeq[slot] & 0xFF) != seq)
-                    continue;  // stale/duplicate
-                int64_t lat = nowUs() - sentAt[slot];
-                out->latBuckets[latBucket2(lat)]++;
-                out->received++;
-                uint8_t rcode = recvBuf[3] & 0x0F;
-                if (rcode == 0)
-                    out->rcodeNoerror++;
-                else
-                    out->rcodeOther++;
-                out->answers += (uint64_t)((recvBuf[6] << 8) | recvBuf[7]);
-                sentAt[slot] = 0;
-                completed++;
-                if (launched < target) sendSlot(slot);
+                int64_t now = nowUs();
+                for (int i = 0; i < nr; ++i) {
+                    const uint8_t* rb = rxBufs[i].data();
+                    if (rxHdrs[i].msg_len < 12) continue;
+                    uint16_t qid = (uint16_t)((rb[0] << 8) | rb[1]);
+                    int slot = qid & 0xFF;
+                    uint8_t seq = (uint8_t)(qid >> 8);
+                    if (slot >= W || sentAt[slot] == 0 ||
+                        (uint8_t)(slotSeq[slot] & 0xFF) != seq)
+                        continue;  // stale/duplicate
+                    int64_t lat = now - sentAt[slot];
+                    out->latBuckets[latBucket2(lat)]++;
+                    out->received++;
+                    uint8_t rcode = rb[3] & 0x0F;
+                    if (rcode == 0)
+                        out->rcodeNoerror++;
+                    else
+                        out->rcodeOther++;
+                    out->answers +=
+                        (uint64_t)((rb[6] << 8) | rb[7]);
+                    sentAt[slot] = 0;
+                    completed++;
+                    freeSlots.push_back(slot);
+                }
+                if (launched < target) batchSend(freeSlots);
+                else freeSlots.clear();
+                if (nr < kRxBatch) break;
             }
         }
         int64_t now = nowUs();
@@ -201,9 +247,11 @@ void worker(const Config& cfg, int tid,
                     out->timeouts++;
                     completed++;
                     sentAt[s] = 0;
-                    if (launched < target) sendSlot(s);
+                    if (launched < target) freeSlots.push_back(s);
                 }
             }
+            if (launched < target) batchSend(freeSlots);
+            else freeSlots.clear();
         }
     }
     close(fd);
