@@ -33,7 +33,9 @@
 
 #include <cstring>
 #include <map>
+#include <unordered_map>
 #include <memory>
+#include <unordered_map>
 #include <thread>
 #include <set>
 #include <string>
@@ -64,7 +66,7 @@ struct Backend {
     std::string in, out;
     bool writeBlocked = false;
     uint32_t nextReq = 1;
-    std::map<uint32_t, PendingReply> pending;
+    std::unordered_map<uint32_t, PendingReply> pending;
     int64_t lastPongAt = 0;
     int64_t pingSentAt = 0;
     uint64_t queries = 0;
@@ -102,6 +104,8 @@ class Balancer {
     void onBackendEvent(std::shared_ptr<Backend> be, uint32_t ev);
     void backendFlush(Backend* be);
     Backend* pickBackend(const std::string& remoteIp);
+    Backend* pickBackendFast(const struct sockaddr_storage& ss);
+    Backend* chooseLeastLoaded();
     void onUdpReadable();
     void onTcpAccept();
     void onTcpClient(std::shared_ptr<TcpClient> c, uint32_t ev);
@@ -121,7 +125,11 @@ class Balancer {
     int udpFd_ = -1, tcpFd_ = -1, statsFd_ = -1;
     std::map<std::string, std::shared_ptr<Backend>> backends_;  // by path
     std::map<int, std::shared_ptr<Backend>> backendsById_;
-    std::map<std::string, int> remotes_;  // remote ip -> backend id
+    std::map<std::string, int> remotes_;  // remote ip -> backend id (stats)
+    /* hot-path affinity: FNV of raw addr bytes -> backend id (avoids
+     * inet_ntop + string alloc per packet; a hash collision only means
+     * two IPs share a pin, which is harmless) */
+    std::unordered_map<uint64_t, int> remotesFast_;
     std::map<int, std::shared_ptr<TcpClient>> tcpClients_;
     int nextBackendId_ = 1;
     int rrCursor_ = 0;
@@ -315,6 +323,12 @@ void Balancer::backendDown(Backend* be) {
         else
             ++it;
     }
+    for (auto it = remotesFast_.begin(); it != remotesFast_.end();) {
+        if (it->second == be->id)
+            it = remotesFast_.erase(it);
+        else
+            ++it;
+    }
     be->remotes = 0;
 }
 
@@ -440,6 +454,17 @@ void Balancer::onBackendEvent(std::shared_ptr<Backend> be, uint32_t ev) {
     be->in.erase(0, consumed);
 }
 
+Backend* Balancer::chooseLeastLoaded() {
+    /* healthy backend with fewest remotes (stable spread) */
+    Backend* best = nullptr;
+    for (auto& [id, be] : backendsById_) {
+        if (!be->ok || be->fd < 0) continue;
+        if (best == nullptr || be->remotes < best->remotes)
+            best = be.get();
+    }
+    return best;
+}
+
 Backend* Balancer::pickBackend(const std::string& remoteIp) {
     auto it = remotes_.find(remoteIp);
     if (it != remotes_.end()) {
@@ -448,18 +473,51 @@ Backend* Balancer::pickBackend(const std::string& remoteIp) {
             return bit->second.get();
         remotes_.erase(it);
     }
-    /* choose the healthy backend with fewest remotes (stable spread) */
-    Backend* best = nullptr;
-    for (auto& [id, be] : backendsById_) {
-        if (!be->ok || be->fd < 0) continue;
-        if (best == nullptr || be->remotes < best->remotes)
-            best = be.get();
-    }
+    Backend* best = chooseLeastLoaded();
     if (best != nullptr) {
         remotes_[remoteIp] = best->id;
         best->remotes++;
     }
     return best;
+}
+
+static uint64_t addrKey(const struct sockaddr_storage& ss) {
+    const uint8_t* p;
+    size_t n;
+    if (ss.ss_family == AF_INET) {
+        p = (const uint8_t*)&((const struct sockaddr_in*)&ss)->sin_addr;
+        n = 4;
+    } else {
+        p = (const uint8_t*)&((const struct sockaddr_in6*)&ss)->sin6_addr;
+        n = 16;
+    }
+    uint64_t h = 0xcbf29ce484222325ull ^ (uint64_t)ss.ss_family;
+    for (size_t i = 0; i < n; ++i) h = (h ^ p[i]) * 0x100000001b3ull;
+    return h;
+}
+
+Backend* Balancer::pickBackendFast(const struct sockaddr_storage& ss) {
+    uint64_t key = addrKey(ss);
+    auto it = remotesFast_.find(key);
+    if (it != remotesFast_.end()) {
+        auto bit = backendsById_.find(it->second);
+        if (bit != backendsById_.end() && bit->second->ok)
+            return bit->second.get();
+        remotesFast_.erase(it);
+    }
+    /* cold path: render the IP once for the stats map */
+    char ip[48] = "";
+    if (ss.ss_family == AF_INET)
+        inet_ntop(AF_INET,
+                  &((const struct sockaddr_in*)&ss)->sin_addr, ip,
+                  sizeof(ip));
+    else
+        inet_ntop(AF_INET6,
+                  &((const struct sockaddr_in6*)&ss)->sin6_addr, ip,
+                  sizeof(ip));
+    Backend* be = pickBackend(ip);
+    if (be != nullptr) remotesFast_[key] = be->id;
+    return be;
 }
 
 static void ipOf(const struct sockaddr_storage& ss, char* out, size_t n,
@@ -497,14 +555,10 @@ void Balancer::onUdpReadable() {
         }
         int n = recvmmsg(udpFd_, hdrs, kBatch, 0, nullptr);
         if (n <= 0) return;
+        int64_t expiry = monotonicMillis() + kReplyTtlMs;
         std::set<Backend*> touched;
         for (int i = 0; i < n; ++i) {
-            char ip[48];
-            uint16_t srcPort;
-            uint8_t family;
-            uint8_t addr16[16];
-            ipOf(addrs[i], ip, sizeof(ip), &srcPort, &family, addr16);
-            Backend* be = pickBackend(ip);
+            Backend* be = pickBackendFast(addrs[i]);
             udpQueries_++;
             if (be == nullptr) {
                 drops_++;
@@ -514,17 +568,31 @@ void Balancer::onUdpReadable() {
             PendingReply pr;
             pr.src = addrs[i];
             pr.srcLen = hdrs[i].msg_hdr.msg_namelen;
-            pr.expiresAt = monotonicMillis() + kReplyTtlMs;
+            pr.expiresAt = expiry;
             be->pending[reqId] = pr;
-            std::string payload;
-            payload.reserve(bsock::kQueryHeadLen + hdrs[i].msg_len);
-            bsock::putU32(payload, reqId);
-            payload.push_back((char)family);
-            payload.push_back((char)0);  // udp
-            bsock::putU16(payload, srcPort);
-            payload.append((const char*)addr16, 16);
-            payload.append((const char*)bufs[i], hdrs[i].msg_len);
-            bsock::appendFrame(be->out, bsock::FRAME_QUERY, payload);
+            /* frame written straight into the backend buffer */
+            std::string& o = be->out;
+            size_t dnsLen = hdrs[i].msg_len;
+            uint32_t plen = (uint32_t)(bsock::kQueryHeadLen + dnsLen);
+            o.push_back((char)bsock::kMagic);
+            o.push_back((char)bsock::FRAME_QUERY);
+            bsock::putU32(o, plen);
+            bsock::putU32(o, reqId);
+            if (addrs[i].ss_family == AF_INET) {
+                const auto* sa = (const struct sockaddr_in*)&addrs[i];
+                o.push_back((char)4);
+                o.push_back((char)0);  // udp
+                bsock::putU16(o, ntohs(sa->sin_port));
+                o.append((const char*)&sa->sin_addr, 4);
+                o.append(12, '\0');
+            } else {
+                const auto* sa = (const struct sockaddr_in6*)&addrs[i];
+                o.push_back((char)6);
+                o.push_back((char)0);
+                bsock::putU16(o, ntohs(sa->sin6_port));
+                o.append((const char*)&sa->sin6_addr, 16);
+            }
+            o.append((const char*)bufs[i], dnsLen);
             be->queries++;
             touched.insert(be);
         }

@@ -161,7 +161,8 @@ const Record* Message::edns() const {
 namespace {
 
 struct Encoder {
-    std::vector<uint8_t> buf;
+    std::vector<uint8_t>& buf;
+    explicit Encoder(std::vector<uint8_t>& b) : buf(b) { buf.clear(); }
     // Compression table: (suffix view, wire offset). Views reference
     // the record name strings, which outlive the encode; linear scan
     // beats a map for the few dozen names in a response, and avoids
@@ -293,8 +294,9 @@ uint16_t flagsWord(const Header& h) {
     return f;
 }
 
-std::vector<uint8_t> encodeImpl(const Message& m, bool truncated) {
-    Encoder e;
+void encodeImpl(const Message& m, bool truncated,
+                std::vector<uint8_t>& out) {
+    Encoder e(out);
     e.buf.reserve(512);
     Header h = m.header;
     if (truncated) h.tc = true;
@@ -327,14 +329,19 @@ std::vector<uint8_t> encodeImpl(const Message& m, bool truncated) {
         for (const auto& r : m.additionals)
             if (r.type == TYPE_OPT) e.encodeRecord(r);
     }
-    return std::move(e.buf);
 }
 
 }  // namespace
 
+void Message::encodeInto(std::vector<uint8_t>& out, size_t maxSize) const {
+    encodeImpl(*this, false, out);
+    if (maxSize > 0 && out.size() > maxSize)
+        encodeImpl(*this, true, out);
+}
+
 std::vector<uint8_t> Message::encode(size_t maxSize) const {
-    std::vector<uint8_t> out = encodeImpl(*this, false);
-    if (maxSize > 0 && out.size() > maxSize) out = encodeImpl(*this, true);
+    std::vector<uint8_t> out;
+    encodeInto(out, maxSize);
     return out;
 }
 

@@ -125,6 +125,9 @@ struct Message {
      * over TCP).
      */
     std::vector<uint8_t> encode(size_t maxSize = 0) const;
+    /* Same, reusing the caller's buffer (cleared first) — the hot-path
+     * variant: no allocation once the buffer has warmed up. */
+    void encodeInto(std::vector<uint8_t>& out, size_t maxSize = 0) const;
 
     /* Decode; nullopt on malformed wire data. */
     static std::optional<Message> decode(const uint8_t* data, size_t len);

@@ -253,7 +253,7 @@ bool DnsServer::process(const uint8_t* data, size_t len, bool udp,
     }
 
     if (hasEdns) resp.additionals.push_back(Record::OPT(1400));
-    out = resp.encode(limit);
+    resp.encodeInto(out, limit);
     QueryTimers tm;
     tm.parseUs = tParse - start;
     tm.resolveUs = tResolve - tParse;
