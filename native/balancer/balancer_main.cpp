@@ -36,6 +36,7 @@
 #include <unordered_map>
 #include <memory>
 #include <unordered_map>
+#include <mutex>
 #include <thread>
 #include <set>
 #include <string>
@@ -49,6 +50,26 @@
 using namespace bamd;
 
 namespace {
+
+/* Remote-assignment counts shared across SO_REUSEPORT workers, keyed by
+ * backend socket path: without this each worker balances only its own
+ * remotes and the aggregate backend load skews badly. Assignment is the
+ * cold path, so a mutex is fine. */
+struct GlobalPins {
+    std::mutex m;
+    std::map<std::string, int> counts;
+    void add(const std::string& path, int d) {
+        std::lock_guard<std::mutex> g(m);
+        counts[path] += d;
+        if (counts[path] < 0) counts[path] = 0;
+    }
+    int get(const std::string& path) {
+        std::lock_guard<std::mutex> g(m);
+        auto it = counts.find(path);
+        return it == counts.end() ? 0 : it->second;
+    }
+};
+GlobalPins g_pins;
 
 struct PendingReply {
     struct sockaddr_storage src;
@@ -329,6 +350,7 @@ void Balancer::backendDown(Backend* be) {
         else
             ++it;
     }
+    g_pins.add(be->path, -(int)be->remotes);
     be->remotes = 0;
 }
 
@@ -455,12 +477,16 @@ void Balancer::onBackendEvent(std::shared_ptr<Backend> be, uint32_t ev) {
 }
 
 Backend* Balancer::chooseLeastLoaded() {
-    /* healthy backend with fewest remotes (stable spread) */
+    /* healthy backend with fewest remotes ACROSS ALL WORKERS */
     Backend* best = nullptr;
+    int bestCount = 0;
     for (auto& [id, be] : backendsById_) {
         if (!be->ok || be->fd < 0) continue;
-        if (best == nullptr || be->remotes < best->remotes)
+        int c = g_pins.get(be->path);
+        if (best == nullptr || c < bestCount) {
             best = be.get();
+            bestCount = c;
+        }
     }
     return best;
 }
@@ -477,6 +503,7 @@ Backend* Balancer::pickBackend(const std::string& remoteIp) {
     if (best != nullptr) {
         remotes_[remoteIp] = best->id;
         best->remotes++;
+        g_pins.add(best->path, 1);
     }
     return best;
 }
