@@ -1,0 +1,103 @@
+#!/usr/bin/env python3
+"""One-off scaling probe: where does the N-process chain saturate?
+
+Starts N binderd (file store for fast startup) + balancer with W
+workers, runs dnsblast at several window/thread settings, and dumps
+per-backend query distribution from the balancer stats socket.
+
+usage: scale_probe.py [N] [workers]
+"""
+import json
+import os
+import socket
+import subprocess
+import sys
+import tempfile
+import time
+from pathlib import Path
+
+REPO = Path(__file__).resolve().parent.parent
+sys.path.insert(0, str(REPO))
+
+from binder_amd.harness import BinderProcess, free_port  # noqa: E402
+
+
+def main():
+    n = int(sys.argv[1]) if len(sys.argv) > 1 else 8
+    workers = int(sys.argv[2]) if len(sys.argv) > 2 else min(8, n)
+    tmp = Path(tempfile.mkdtemp(prefix="scale-probe-"))
+    tree = {"foo.com": None}
+    names = []
+    for i in range(5000):
+        tree[f"h{i}.foo.com"] = {
+            "type": "host",
+            "host": {"address": f"10.{(i >> 8) & 255}.{i & 255}.1"}}
+        names.append(f"h{i}.foo.com A")
+    (tmp / "tree.json").write_text(json.dumps(tree))
+    (tmp / "names.txt").write_text("\n".join(names))
+
+    sockdir = tmp / "socks"
+    sockdir.mkdir()
+    backends = []
+    for i in range(n):
+        b = BinderProcess(store=f"file:{tmp/'tree.json'}", workdir=tmp,
+                          log_level="warn",
+                          balancer_socket=str(sockdir / f"b{i}"))
+        b.start()
+        backends.append(b)
+
+    bal_port = free_port()
+    bal = subprocess.Popen(
+        [str(REPO / "bin" / "binder-balancer"), "-p", str(bal_port),
+         "-H", "127.0.0.1", "-s", str(sockdir),
+         "-S", str(tmp / "stats.sock"), "-r", "200",
+         "-w", str(workers)],
+        env=dict(os.environ, LOG_LEVEL="warn"),
+        stdout=subprocess.DEVNULL, stderr=subprocess.STDOUT)
+    time.sleep(1.0)
+
+    def stats():
+        agg = {}
+        with socket.socket(socket.AF_UNIX) as s:
+            s.connect(str(tmp / "stats.sock"))
+            return json.loads(s.recv(1 << 20).decode())
+
+    def blast(q, window, threads):
+        out = subprocess.run(
+            [str(REPO / "bin" / "dnsblast"), "-s", "127.0.0.1",
+             "-p", str(bal_port), "-n", str(q), "-c", str(window),
+             "-t", str(threads), "-f", str(tmp / "names.txt"),
+             "-B", "127.0.1.1"],
+            capture_output=True, text=True, check=True)
+        return json.loads(out.stdout.strip())
+
+    try:
+        # direct-to-one-backend ceiling
+        d = subprocess.run(
+            [str(REPO / "bin" / "dnsblast"), "-s", "127.0.0.1",
+             "-p", str(backends[0].port), "-n", "300000", "-c", "64",
+             "-t", "4", "-f", str(tmp / "names.txt")],
+            capture_output=True, text=True, check=True)
+        print("direct-1-backend:", d.stdout.strip())
+
+        for window, threads in [(64, 4 * n), (128, 4 * n),
+                                (64, 8 * n), (128, 8 * n)]:
+            r = blast(150000 * n, window, threads)
+            print(f"w={window} t={threads}: qps={r['qps']:.0f} "
+                  f"p50={r['p50_us']} p99={r['p99_us']} "
+                  f"timeouts={r['timeouts']}")
+        st = stats()
+        qtot = sum(b["queries"] for b in st["backends"])
+        print("backend distribution (queries):",
+              sorted(round(b["queries"] / max(qtot, 1), 3)
+                     for b in st["backends"]))
+        print("worker0 udp_queries:", st["udp_queries"],
+              "drops:", st["drops"])
+    finally:
+        bal.terminate()
+        for b in backends:
+            b.stop()
+
+
+if __name__ == "__main__":
+    main()
