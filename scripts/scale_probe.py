@@ -33,6 +33,18 @@ def main():
             "type": "host",
             "host": {"address": f"10.{(i >> 8) & 255}.{i & 255}.1"}}
         names.append(f"h{i}.foo.com A")
+    # same A+SRV mix as bench.py
+    for i in range(1000):
+        tree[f"s{i}.foo.com"] = {
+            "type": "service",
+            "service": {"srvce": "_x", "proto": "_tcp", "port": 80,
+                        "ttl": 60}}
+        for j in range(4):
+            tree[f"m{j}.s{i}.foo.com"] = {
+                "type": "rr_host",
+                "rr_host": {"address": f"10.9.{i % 250}.{j + 1}"}}
+        names.append(f"s{i}.foo.com A")
+        names.append(f"_x._tcp.s{i}.foo.com SRV")
     (tmp / "tree.json").write_text(json.dumps(tree))
     (tmp / "names.txt").write_text("\n".join(names))
 
@@ -47,17 +59,24 @@ def main():
         backends.append(b)
 
     bal_port = free_port()
-    bal = subprocess.Popen(
-        [str(REPO / "bin" / "binder-balancer"), "-p", str(bal_port),
-         "-H", "127.0.0.1", "-s", str(sockdir),
-         "-S", str(tmp / "stats.sock"), "-r", "200",
-         "-w", str(workers)],
-        env=dict(os.environ, LOG_LEVEL="warn"),
-        stdout=subprocess.DEVNULL, stderr=subprocess.STDOUT)
-    time.sleep(1.0)
+    bal = None
+
+    def restart_balancer(w):
+        nonlocal bal, bal_port
+        if bal is not None:
+            bal.terminate()
+            bal.wait(timeout=5)
+        bal_port = free_port()
+        bal = subprocess.Popen(
+            [str(REPO / "bin" / "binder-balancer"), "-p", str(bal_port),
+             "-H", "127.0.0.1", "-s", str(sockdir),
+             "-S", str(tmp / "stats.sock"), "-r", "200",
+             "-w", str(w)],
+            env=dict(os.environ, LOG_LEVEL="warn"),
+            stdout=subprocess.DEVNULL, stderr=subprocess.STDOUT)
+        time.sleep(1.2)
 
     def stats():
-        agg = {}
         with socket.socket(socket.AF_UNIX) as s:
             s.connect(str(tmp / "stats.sock"))
             return json.loads(s.recv(1 << 20).decode())
@@ -80,19 +99,19 @@ def main():
             capture_output=True, text=True, check=True)
         print("direct-1-backend:", d.stdout.strip())
 
-        for window, threads in [(64, 4 * n), (128, 4 * n),
-                                (64, 8 * n), (128, 8 * n)]:
+        for bal_workers, window, threads in [
+                (n, 64, 4 * n), (2 * n, 64, 4 * n),
+                (2 * n, 64, 6 * n), (2 * n, 128, 4 * n),
+                (16, 64, 32)]:
+            restart_balancer(bal_workers)
             r = blast(150000 * n, window, threads)
-            print(f"w={window} t={threads}: qps={r['qps']:.0f} "
-                  f"p50={r['p50_us']} p99={r['p99_us']} "
-                  f"timeouts={r['timeouts']}")
-        st = stats()
-        qtot = sum(b["queries"] for b in st["backends"])
-        print("backend distribution (queries):",
-              sorted(round(b["queries"] / max(qtot, 1), 3)
-                     for b in st["backends"]))
-        print("worker0 udp_queries:", st["udp_queries"],
-              "drops:", st["drops"])
+            st = stats()
+            qtot = sum(b["queries"] for b in st["backends"])
+            dist = sorted(round(b["queries"] / max(qtot, 1), 3)
+                          for b in st["backends"])
+            print(f"bw={bal_workers} w={window} t={threads}: "
+                  f"qps={r['qps']:.0f} p50={r['p50_us']} "
+                  f"p99={r['p99_us']} to={r['timeouts']} dist={dist}")
     finally:
         bal.terminate()
         for b in backends:
