@@ -165,9 +165,16 @@ def main():
     q_step = args.queries_per_proc * n
     ncpu = os.cpu_count() or 8
     # size the harness to the machine: the load generator + balancer
-    # must not starve the backends on small boxes
-    workers = min(8, n) if ncpu >= 6 * n else 1
-    threads = min(6 * n, max(2, ncpu // 2))
+    # must not starve the backends on small boxes (probe-derived
+    # operating point: scripts/scale_probe.py + profiles/SCALING.md)
+    if ncpu >= 8 * n + 8:
+        workers = min(16, max(4, 2 * n))
+        threads = 4 * n
+        window = 128
+    else:
+        workers = 1
+        threads = min(4 * n, max(2, ncpu // 2))
+        window = args.window
     result = {}
 
     tmp = None
@@ -192,8 +199,7 @@ def main():
             log(f"balancer ready on :{bal_port}; warmup "
                 f"{args.warmup} x {q_step} queries")
             for _ in range(args.warmup):
-                run_blast(bal_port, q_step, names_file, threads,
-                          args.window)
+                run_blast(bal_port, q_step, names_file, threads, window)
 
         barrier()
         cuda_sync()
@@ -202,7 +208,7 @@ def main():
         if rank == 0:
             for s in range(args.steps):
                 last = run_blast(bal_port, q_step, names_file, threads,
-                                 args.window)
+                                 window)
                 log(f"step {s + 1}/{args.steps}: "
                     f"{last['qps']:.0f} qps, p99 {last['p99_us']}us")
         barrier()
