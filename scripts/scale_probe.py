@@ -86,7 +86,7 @@ def main():
             [str(REPO / "bin" / "dnsblast"), "-s", "127.0.0.1",
              "-p", str(bal_port), "-n", str(q), "-c", str(window),
              "-t", str(threads), "-f", str(tmp / "names.txt"),
-             "-B", "127.0.1.1"],
+             "-B", "127.0.1.1", "-T", "5000"],
             capture_output=True, text=True, check=True)
         return json.loads(out.stdout.strip())
 
@@ -99,10 +99,14 @@ def main():
             capture_output=True, text=True, check=True)
         print("direct-1-backend:", d.stdout.strip())
 
-        for bal_workers, window, threads in [
-                (n, 64, 4 * n), (2 * n, 64, 4 * n),
-                (2 * n, 64, 6 * n), (2 * n, 128, 4 * n),
-                (16, 64, 32)]:
+        if len(sys.argv) > 3:
+            configs = [tuple(int(x) for x in a.split(":"))
+                       for a in sys.argv[3:]]
+        else:
+            configs = [(n, 64, 4 * n), (2 * n, 64, 4 * n),
+                       (2 * n, 64, 6 * n), (2 * n, 128, 4 * n),
+                       (16, 64, 32)]
+        for bal_workers, window, threads in configs:
             restart_balancer(bal_workers)
             r = blast(150000 * n, window, threads)
             st = stats()
