@@ -122,7 +122,8 @@ def run_blast(port, queries, names_file, threads, window):
     out = subprocess.run(
         [str(REPO / "bin" / "dnsblast"), "-s", "127.0.0.1",
          "-p", str(port), "-n", str(queries), "-c", str(window),
-         "-t", str(threads), "-f", str(names_file), "-B", "127.0.1.1"],
+         "-t", str(threads), "-f", str(names_file), "-B", "127.0.1.1",
+         "-T", "10000"],
         capture_output=True, text=True, check=True)
     return json.loads(out.stdout.strip().splitlines()[-1])
 
@@ -168,9 +169,9 @@ def main():
     # must not starve the backends on small boxes (probe-derived
     # operating point: scripts/scale_probe.py + profiles/SCALING.md)
     if ncpu >= 8 * n + 8:
-        workers = min(16, max(4, 2 * n))
-        threads = 4 * n
-        window = 128
+        workers = min(16, 4 * n)
+        threads = max(8, 4 * n)
+        window = 256
     else:
         workers = 1
         threads = min(4 * n, max(2, ncpu // 2))
