@@ -169,9 +169,9 @@ def main():
     # must not starve the backends on small boxes (probe-derived
     # operating point: scripts/scale_probe.py + profiles/SCALING.md)
     if ncpu >= 8 * n + 8:
-        workers = min(16, 4 * n)
+        workers = min(16, max(4, 2 * n))
         threads = max(8, 4 * n)
-        window = 256
+        window = 192
     else:
         workers = 1
         threads = min(4 * n, max(2, ncpu // 2))

@@ -308,6 +308,9 @@ void Balancer::rescan() {
 void Balancer::connectBackend(Backend* be) {
     int fd = socket(AF_UNIX, SOCK_STREAM | SOCK_NONBLOCK | SOCK_CLOEXEC, 0);
     if (fd < 0) return;
+    int sz = 4 << 20;
+    setsockopt(fd, SOL_SOCKET, SO_SNDBUF, &sz, sizeof(sz));
+    setsockopt(fd, SOL_SOCKET, SO_RCVBUF, &sz, sizeof(sz));
     struct sockaddr_un sa {};
     sa.sun_family = AF_UNIX;
     snprintf(sa.sun_path, sizeof(sa.sun_path), "%s", be->path.c_str());

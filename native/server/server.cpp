@@ -508,6 +508,9 @@ void DnsServer::onBalAccept() {
         int fd = accept4(balFd_, nullptr, nullptr,
                          SOCK_NONBLOCK | SOCK_CLOEXEC);
         if (fd < 0) return;
+        int sz = 4 << 20;
+        setsockopt(fd, SOL_SOCKET, SO_SNDBUF, &sz, sizeof(sz));
+        setsockopt(fd, SOL_SOCKET, SO_RCVBUF, &sz, sizeof(sz));
         auto conn = std::make_shared<BalConn>();
         conn->fd = fd;
         BalConn* raw = conn.get();
