@@ -4,7 +4,7 @@
 CXX ?= g++
 PYTHON ?= python3
 CXXFLAGS ?= -O2 -g -std=c++20 -fPIC -fno-omit-frame-pointer \
-	-Wall -Wextra -Werror -Wno-unused-parameter
+	-Wall -Wextra -Werror -Wno-unused-parameter -MMD -MP
 LDFLAGS ?=
 
 BUILD := build
@@ -21,6 +21,7 @@ SERVER_SRCS := \
 	native/server/metrics.cpp \
 	native/server/server.cpp \
 	native/server/recursion.cpp \
+	native/server/ldap.cpp \
 	native/zk/client.cpp \
 	native/zk/mirror.cpp
 
@@ -38,7 +39,7 @@ all: $(PYMOD) $(BINARIES)
 
 bin/binderd: $(CORE_OBJS) $(SERVER_OBJS) $(BUILD)/native/server/binderd_main.o
 	@mkdir -p bin
-	$(CXX) $(CXXFLAGS) $^ -o $@ $(LDFLAGS)
+	$(CXX) $(CXXFLAGS) $^ -o $@ $(LDFLAGS) -lssl -lcrypto -lpthread
 
 bin/binder-balancer: $(CORE_OBJS) $(BUILD)/native/balancer/balancer_main.o
 	@mkdir -p bin
@@ -75,5 +76,7 @@ $(PYMOD): $(CORE_OBJS) $(BUILD)/native/pybind/module.o
 
 clean:
 	rm -rf $(BUILD) $(PYMOD) $(BINARIES)
+
+-include $(shell find $(BUILD) -name '*.d' 2>/dev/null)
 
 .PHONY: all clean

@@ -2,6 +2,7 @@
 
 #include <fcntl.h>
 #include <sys/epoll.h>
+#include <sys/eventfd.h>
 #include <unistd.h>
 
 #include <cstdio>
@@ -23,9 +24,40 @@ EventLoop::EventLoop() {
     if (epfd_ < 0)
         throw std::runtime_error(std::string("epoll_create1: ") +
                                  strerror(errno));
+    wakeFd_ = eventfd(0, EFD_NONBLOCK | EFD_CLOEXEC);
+    if (wakeFd_ >= 0) {
+        struct epoll_event ev {};
+        ev.events = EPOLLIN;
+        ev.data.fd = wakeFd_;
+        epoll_ctl(epfd_, EPOLL_CTL_ADD, wakeFd_, &ev);
+        fds_[wakeFd_] = [this](uint32_t) {
+            uint64_t v;
+            while (read(wakeFd_, &v, sizeof(v)) == sizeof(v)) {
+            }
+            std::vector<TimerCallback> tasks;
+            {
+                std::lock_guard<std::mutex> g(postMutex_);
+                tasks.swap(posted_);
+            }
+            for (auto& t : tasks) t();
+        };
+    }
 }
 
-EventLoop::~EventLoop() { close(epfd_); }
+EventLoop::~EventLoop() {
+    if (wakeFd_ >= 0) close(wakeFd_);
+    close(epfd_);
+}
+
+void EventLoop::postFromThread(TimerCallback cb) {
+    {
+        std::lock_guard<std::mutex> g(postMutex_);
+        posted_.push_back(std::move(cb));
+    }
+    uint64_t one = 1;
+    ssize_t rv = write(wakeFd_, &one, sizeof(one));
+    (void)rv;
+}
 
 void EventLoop::addFd(int fd, uint32_t events, FdCallback cb) {
     struct epoll_event ev {};

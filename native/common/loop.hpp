@@ -12,6 +12,7 @@
 #include <cstdint>
 #include <functional>
 #include <map>
+#include <mutex>
 #include <queue>
 #include <vector>
 
@@ -36,8 +37,12 @@ class EventLoop {
     uint64_t addTimer(int64_t delayMs, TimerCallback cb);
     void cancelTimer(uint64_t id);
 
-    /* Run cb on the next loop iteration. */
+    /* Run cb on the next loop iteration (loop thread only). */
     void defer(TimerCallback cb);
+
+    /* Thread-safe: queue cb for execution on the loop thread and wake
+     * it. Used by helper threads (e.g. the blocking LDAP refresh). */
+    void postFromThread(TimerCallback cb);
 
     void run();
     void stop() { running_ = false; }
@@ -67,6 +72,10 @@ class EventLoop {
     std::map<uint64_t, TimerCallback> timers_;  // id -> cb (absent=cancelled)
     uint64_t nextTimerId_ = 1;
     std::vector<TimerCallback> deferred_;
+
+    int wakeFd_ = -1;
+    std::mutex postMutex_;
+    std::vector<TimerCallback> posted_;
 };
 
 /* fcntl O_NONBLOCK helper; returns false on error. */

@@ -1,5 +1,7 @@
 #include "recursion.hpp"
 
+#include "ldap.hpp"
+
 #include <arpa/inet.h>
 #include <ifaddrs.h>
 #include <netinet/in.h>
@@ -24,6 +26,7 @@ Recursion::Recursion(EventLoop* loop, Logger log, RecursionOptions opts,
       opts_(std::move(opts)), store_(store) {}
 
 Recursion::~Recursion() {
+    if (ldapThread_.joinable()) ldapThread_.join();
     if (fd_ >= 0) {
         loop_->delFd(fd_);
         close(fd_);
@@ -62,8 +65,19 @@ void Recursion::scheduleRefresh(int64_t ms) {
 void Recursion::refresh() {
     const Json& cfg = opts_.config;
     std::string source = cfg.get("source").asString();
-    if (source.empty())
-        source = cfg.get("dcs").isObject() ? "static" : "zk";
+    if (source.empty()) {
+        if (cfg.get("dcs").isObject())
+            source = "static";
+        else if (cfg.get("ufds").isObject())
+            source = "ufds";  /* the reference's config shape */
+        else
+            source = "zk";
+    }
+
+    if (source == "ufds") {
+        refreshViaUfds();
+        return;
+    }
 
     if (source == "static") {
         std::map<std::string, std::vector<std::string>> dcs;
@@ -132,6 +146,122 @@ void Recursion::refresh() {
               "(best effort)");
     emitReady();
     scheduleRefresh(kRetryInitMs);
+}
+
+/*
+ * UFDS source: bootstrap the LDAP address from our own mirror, then
+ * listResolvers(region) — recursion.js:104-127 + 202-249. The blocking
+ * LDAP conversation runs on a helper thread; only the result lands on
+ * the loop thread.
+ */
+void Recursion::refreshViaUfds() {
+    if (ldapBusy_) {
+        scheduleRefresh(kRetryInitMs);
+        return;
+    }
+    const Json& ucfg = opts_.config.get("ufds");
+    std::string url = ucfg.get("url").asString();
+    bool tls = url.rfind("ldaps://", 0) == 0;
+    std::string domain = url;
+    for (const char* scheme : {"ldaps://", "ldap://"}) {
+        if (domain.rfind(scheme, 0) == 0) {
+            domain = domain.substr(strlen(scheme));
+            break;
+        }
+    }
+    uint16_t port = tls ? 636 : 389;
+    size_t colon = domain.rfind(':');
+    if (colon != std::string::npos &&
+        domain.find(':') == colon /* not v6 */) {
+        port = (uint16_t)atoi(domain.c_str() + colon + 1);
+        domain = domain.substr(0, colon);
+    }
+
+    /* resolveUfds: must be a service node with a first child */
+    std::string addr = domain;
+    bool isIp = true;
+    for (char c : domain)
+        isIp = isIp && ((c >= '0' && c <= '9') || c == '.');
+    if (!isIp) {
+        if (!store_->ready()) {
+            log_.warn("Recursion: ZK is not yet available");
+            emitReady();
+            scheduleRefresh(kRetryInitMs);
+            return;
+        }
+        const StoreNode* node = store_->lookup(domain);
+        const StoreNode* kid = nullptr;
+        if (node != nullptr && node->rec().type == RecType::Service) {
+            auto kids = node->children();
+            if (!kids.empty()) kid = kids[0];
+        }
+        if (kid == nullptr || !kid->rec().valid ||
+            kid->rec().address.empty()) {
+            log_.warn("Recursion: not yet able to resolve ufds");
+            emitReady();
+            scheduleRefresh(kRetryInitMs);
+            return;
+        }
+        addr = kid->rec().address;
+    }
+
+    ldap::Options lopts;
+    lopts.host = addr;
+    lopts.port = port;
+    lopts.tls = tls;
+    lopts.bindDn = ucfg.get("bindDN").asString();
+    lopts.bindPassword = ucfg.get("bindPassword").asString();
+    std::string region = opts_.regionName;
+
+    ldapBusy_ = true;
+    if (ldapThread_.joinable()) ldapThread_.join();
+    EventLoop* loop = loop_;
+    Recursion* self = this;
+    ldapThread_ = std::thread([self, loop, lopts, region]() {
+        ldap::Client client(lopts);
+        bool ok = client.connect();
+        std::vector<ldap::Entry> entries;
+        std::string errMsg;
+        if (ok) {
+            ok = client.search("region=" + region + ", o=smartdc",
+                               "objectclass", "resolver", entries);
+        }
+        if (!ok) errMsg = client.error();
+        client.close();
+        loop->postFromThread([self, ok, errMsg,
+                              entries = std::move(entries)]() {
+            self->ldapBusy_ = false;
+            if (!ok) {
+                self->log_.warn(
+                    {{"err", Json(errMsg)}},
+                    "Recursion: Binder is configured for recursive dns "
+                    "but is unable to reach UFDS. Will try again in 15 "
+                    "seconds (best effort).");
+                self->emitReady();
+                self->scheduleRefresh(kRetryInitMs);
+                return;
+            }
+            std::map<std::string, std::vector<std::string>> dcs;
+            for (const auto& e : entries) {
+                auto dcIt = e.find("datacenter");
+                auto ipIt = e.find("ip");
+                if (dcIt == e.end() || ipIt == e.end() ||
+                    dcIt->second.empty() || ipIt->second.empty())
+                    continue;
+                const std::string& dc = dcIt->second[0];
+                const std::string& ip = ipIt->second[0];
+                auto& v = dcs[dc];
+                bool dup = false;
+                for (const auto& x : v) dup = dup || x == ip;
+                if (!dup) v.push_back(ip);
+            }
+            self->dcs_ = std::move(dcs);
+            self->log_.info({{"dcs", Json((int64_t)self->dcs_.size())}},
+                            "setting recursion resolvers (ufds)");
+            self->emitReady();
+            self->scheduleRefresh(kRefreshIntervalMs);
+        });
+    });
 }
 
 std::vector<std::string> Recursion::ownAddrs() {

@@ -16,9 +16,12 @@
  * Registry sources are pluggable where the reference hardcodes UFDS/LDAP
  * (its listResolvers(region) call, recursion.js:210-219):
  *   "static": {"source":"static","dcs":{"dc1":["10.0.0.5",...]}}
- *   "zk":     {"source":"zk","path":"/resolvers"} — znode JSON payload
- *             {"dc1":["ip",...],...}, read via the shared mirror client.
- * An LDAP driver for UFDS parity is planned (gap tracked in docs/).
+ *   "zk":     {"source":"zk","registryDomain":"resolvers.<domain>"} —
+ *             host-like children keyed "<dc>-<n>" under that domain.
+ *   "ufds":   the reference's own path: resolve the UFDS address from
+ *             the mirror (recursion.js:104-127), then LDAP-search
+ *             resolvers for the region (ldap.hpp) every 5 minutes on a
+ *             helper thread.
  */
 #pragma once
 
@@ -26,6 +29,7 @@
 #include <map>
 #include <memory>
 #include <string>
+#include <thread>
 #include <vector>
 
 #include "../common/json.hpp"
@@ -107,6 +111,12 @@ class Recursion : public RecursionIface {
 
     std::vector<std::string> nicCache_;
     int64_t nicCacheAtMs_ = 0;
+
+    /* UFDS/LDAP refresh runs on a helper thread (blocking client);
+     * results are posted back via EventLoop::postFromThread. */
+    std::thread ldapThread_;
+    bool ldapBusy_ = false;
+    void refreshViaUfds();
 };
 
 }  // namespace bamd

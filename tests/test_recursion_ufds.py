@@ -1,0 +1,122 @@
+"""UFDS/LDAP resolver discovery: the reference's full recursion
+bootstrap path (recursion.js:104-249) against stub ZK + stub LDAP.
+
+Flow under test: binderd mirrors the tree (which contains the UFDS
+service registration) -> recursion resolves the UFDS address from its
+own mirror -> LDAP bind + search region resolvers -> misses forward to
+the discovered DC resolver.
+"""
+import json
+import time
+
+import pytest
+
+from binder_amd.harness import BinderProcess
+from binder_amd.stubldap import StubLdap
+from binder_amd.stubzk import StubZk
+
+
+@pytest.fixture()
+def stack(tmp_path):
+    ldap = StubLdap().start()
+
+    # upstream binder for "dc2" on a second loopback address
+    up_tree = tmp_path / "up.json"
+    up_tree.write_text(json.dumps({
+        "dc2.foo.com": None,
+        "api.dc2.foo.com": {"type": "host",
+                            "host": {"address": "10.22.5.5"}},
+    }))
+    upstream = BinderProcess(dns_domain="dc2.foo.com", datacenter="dc2",
+                             host="127.0.0.2", store=f"file:{up_tree}",
+                             workdir=tmp_path,
+                             log_path=str(tmp_path / "up.log"))
+    upstream.start()
+
+    ldap.resolvers = [{"datacenter": "dc2", "ip": "127.0.0.2"}]
+
+    zk = StubZk().start()
+    zk.mkdirp("/com/foo")
+    # UFDS registered as a service in ZK (the resolveUfds path)
+    zk.put("/com/foo/ufds", json.dumps({
+        "type": "service",
+        "service": {"srvce": "_ldap", "proto": "_tcp", "port": ldap.port},
+    }).encode())
+    zk.put("/com/foo/ufds/ufds0", json.dumps({
+        "type": "host", "host": {"address": "127.0.0.1"}}).encode())
+
+    local = BinderProcess(
+        dns_domain="foo.com", datacenter="dc1", store="zk",
+        zk_host="127.0.0.1", zk_port=zk.port, workdir=tmp_path,
+        log_path=str(tmp_path / "local.log"),
+        config={"recursion": {
+            "source": "ufds",
+            "regionName": "region-1",
+            "dnsDomain": "foo.com",
+            "upstreamPort": upstream.port,
+            "ufds": {
+                "url": f"ldap://ufds.foo.com:{ldap.port}",
+                "bindDN": "cn=root",
+                "bindPassword": "secret",
+            },
+        }})
+    local.start()
+    yield {"local": local, "upstream": upstream, "zk": zk, "ldap": ldap}
+    local.stop()
+    upstream.stop()
+    zk.stop()
+    ldap.stop()
+
+
+def test_ufds_discovery_and_forwarding(stack):
+    local, ldap = stack["local"], stack["ldap"]
+    # wait for mirror + first LDAP refresh (retries every 15s on
+    # failure; the first attempt may race mirror readiness)
+    deadline = time.time() + 40
+    while time.time() < deadline:
+        try:
+            r = local.dig("api.dc2.foo.com", rd=True, timeout=4)
+            if r.status == "NOERROR":
+                break
+        except OSError:
+            pass
+        time.sleep(0.5)
+    else:
+        log = open(local.log_path).read()[-3000:]
+        pytest.fail(f"never forwarded via ufds resolvers; log: {log}")
+    assert r.answers[0]["address"] == "10.22.5.5"
+    # the LDAP server saw a bind and the region search
+    assert any("region=region-1" in s for s in ldap.searches)
+    assert "cn=root" in stack["ldap"].binds
+
+
+def test_ufds_bad_credentials_best_effort(tmp_path):
+    """Bind failure leaves recursion empty (best effort) but the server
+    still serves and refuses misses."""
+    ldap = StubLdap()
+    ldap.require_password = "correct"
+    ldap.start()
+    zk = StubZk().start()
+    zk.mkdirp("/com/foo")
+    try:
+        srv = BinderProcess(
+            dns_domain="foo.com", datacenter="dc1", store="zk",
+            zk_host="127.0.0.1", zk_port=zk.port, workdir=tmp_path,
+            config={"recursion": {
+                "source": "ufds",
+                "regionName": "r1",
+                "dnsDomain": "foo.com",
+                "ufds": {"url": f"ldap://127.0.0.1:{ldap.port}",
+                         "bindDN": "cn=root",
+                         "bindPassword": "wrong"},
+            }})
+        srv.start()
+        try:
+            time.sleep(1.0)
+            r = srv.dig("nope.dc9.foo.com", rd=True, timeout=4)
+            assert r.status == "REFUSED"
+        finally:
+            srv.stop()
+    finally:
+        zk.stop()
+        ldap.stop()
