@@ -74,9 +74,18 @@ $(BUILD)/native/pybind/module.o: native/pybind/module.cpp
 $(PYMOD): $(CORE_OBJS) $(BUILD)/native/pybind/module.o
 	$(CXX) -shared $(CXXFLAGS) $^ -o $@ $(LDFLAGS)
 
+# `make test` parity with the reference's nodeunit target
+# (Makefile:169-171 there)
+test: all
+	$(PYTHON) -m pytest tests -q -m "not gpu"
+
+check: all
+	$(PYTHON) -m py_compile binder_amd/*.py bench.py __graft_entry__.py
+	@echo "check OK"
+
 clean:
 	rm -rf $(BUILD) $(PYMOD) $(BINARIES)
 
 -include $(shell find $(BUILD) -name '*.d' 2>/dev/null)
 
-.PHONY: all clean
+.PHONY: all clean test check

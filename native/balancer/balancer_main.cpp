@@ -103,6 +103,7 @@ struct TcpClient {
     struct sockaddr_storage src;
     socklen_t srcLen;
     int backendId = -1;
+    int64_t lastActivityMs = 0;
 };
 
 class Balancer {
@@ -642,6 +643,7 @@ void Balancer::onTcpAccept() {
         c->fd = fd;
         c->src = ss;
         c->srcLen = sl;
+        c->lastActivityMs = monotonicMillis();
         tcpClients_[fd] = c;
         loop_->addFd(fd, EPOLLIN, [this, c](uint32_t ev) {
             onTcpClient(c, ev);
@@ -686,6 +688,7 @@ void Balancer::onTcpClient(std::shared_ptr<TcpClient> c, uint32_t ev) {
     }
     if (ev & EPOLLOUT) tcpClientFlush(c.get());
     if (c->closed || !(ev & EPOLLIN)) return;
+    c->lastActivityMs = monotonicMillis();
 
     char buf[8192];
     while (true) {
@@ -777,6 +780,17 @@ void Balancer::onStatsAccept() {
 
 void Balancer::sweep() {
     int64_t now = monotonicMillis();
+    /* reap idle TCP clients (60s) */
+    std::vector<std::shared_ptr<TcpClient>> idle;
+    for (auto& [fd, c] : tcpClients_)
+        if (now - c->lastActivityMs > 60000) idle.push_back(c);
+    for (auto& c : idle) {
+        if (c->closed) continue;
+        c->closed = true;
+        loop_->delFd(c->fd);
+        close(c->fd);
+        tcpClients_.erase(c->fd);
+    }
     for (auto& [path, be] : backends_) {
         if (be->fd < 0) continue;
         /* expire stale pendings */

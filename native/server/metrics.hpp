@@ -24,6 +24,9 @@ class Counter {
     void increment(const std::string& labels, uint64_t by = 1) {
         vals_[labels] += by;
     }
+    void set(const std::string& labels, uint64_t v) {  // gauges
+        vals_[labels] = v;
+    }
     const std::map<std::string, uint64_t>& values() const { return vals_; }
 
   private:

@@ -168,7 +168,25 @@ bool DnsServer::openBalancer() {
 bool DnsServer::start() {
     if (!openUdp() || !openTcp()) return false;
     if (!opts_.balancerSocket.empty() && !openBalancer()) return false;
+    /* idle TCP sweep: DNS-over-TCP clients that neither query nor
+     * close get reaped after 60s (resource protection; the balancer
+     * socket is exempt — it is a long-lived peer) */
+    auto sweep = std::make_shared<std::function<void()>>();
+    *sweep = [this, sweep]() {
+        sweepIdleTcp();
+        loop_->addTimer(10000, *sweep);
+    };
+    loop_->addTimer(10000, *sweep);
     return true;
+}
+
+void DnsServer::sweepIdleTcp() {
+    int64_t cutoff = monotonicMillis() - 60000;
+    std::vector<TcpConn*> idle;
+    for (auto& [fd, c] : tcpConns_)
+        if (c->lastActivityMs < cutoff && c->pendingAsync == 0)
+            idle.push_back(c.get());
+    for (TcpConn* c : idle) closeTcp(c);
 }
 
 void DnsServer::stop() {
@@ -407,6 +425,7 @@ void DnsServer::onTcpAccept() {
         if (fd < 0) return;
         auto conn = std::make_shared<TcpConn>();
         conn->fd = fd;
+        conn->lastActivityMs = monotonicMillis();
         fillClientInfo(conn->ci, ss, "tcp");
         TcpConn* raw = conn.get();
         tcpConns_[fd] = std::move(conn);
@@ -457,6 +476,7 @@ void DnsServer::onTcpConn(TcpConn* c, uint32_t events) {
     if (events & EPOLLOUT) tcpFlush(c);
     if (c->closed || !(events & EPOLLIN)) return;
 
+    c->lastActivityMs = monotonicMillis();
     char buf[8192];
     while (true) {
         ssize_t nr = read(c->fd, buf, sizeof(buf));

@@ -83,6 +83,7 @@ class DnsServer {
         ClientInfo ci;
         uint32_t pendingAsync = 0;
         bool closed = false;
+        int64_t lastActivityMs = 0;
     };
     struct BalConn {
         int fd;
@@ -119,6 +120,7 @@ class DnsServer {
 
     void tcpFlush(TcpConn* c);
     void tcpMaybeClose(TcpConn* c);
+    void sweepIdleTcp();
     void balFlush(BalConn* c);
     void closeTcp(TcpConn* c);
     void closeBal(BalConn* c);

@@ -92,10 +92,6 @@ void ZkMirror::rebuild() {
         bind(n);
         for (auto& [label, kid] : n->kids_) stack.push_back(kid);
     }
-    if (nodesGauge_) {
-        /* re-render gauge: single value series */
-        // (gauge Counter semantics: we just set via fresh label)
-    }
     log_.info({{"nodes", Json((int64_t)byPath_.size())}},
               "ZK session established; mirror resync started");
 }
@@ -149,6 +145,8 @@ void ZkMirror::onChildren(const std::string& path,
                           const std::vector<std::string>& kids) {
     Node* n = nodeAt(path);
     if (n == nullptr) return;
+    if (nodesGauge_ != nullptr)
+        nodesGauge_->set("", byPath_.size());
 
     std::unordered_map<std::string, Node*> newKids;
     for (const std::string& kid : kids) {
