@@ -37,18 +37,22 @@ def log(msg):
     print(f"# bench: {msg}", file=sys.stderr, flush=True)
 
 
-def build_tree(zk, names_path):
-    """10k records: 5000 hosts + 1000 services x 4 members (=10k nodes
-    with payloads), uniform A+SRV query mix."""
+def build_tree(zk, names_path, records=10000):
+    """R records: R/2 hosts + R/10 services x 4 members (=R nodes with
+    payloads), uniform A+SRV query mix. R=10000 is the BASELINE
+    headline config; --tree-records 100000 with --churn-qps exercises
+    BASELINE config 5."""
     names = []
     zk.mkdirp("/com/foo")
     batch = []
-    for i in range(5000):
+    n_hosts = records // 2
+    n_svcs = records // 10
+    for i in range(n_hosts):
         batch.append((f"/com/foo/h{i}",
                       {"type": "host",
                        "host": {"address": f"10.{(i >> 8) & 255}.{i & 255}.1"}}))
         names.append(f"h{i}.foo.com A")
-    for i in range(1000):
+    for i in range(n_svcs):
         batch.append((f"/com/foo/s{i}",
                       {"type": "service",
                        "service": {"srvce": "_x", "proto": "_tcp",
@@ -65,7 +69,7 @@ def build_tree(zk, names_path):
     return len(batch)
 
 
-def start_backends(n, tmp, zk_port):
+def start_backends(n, tmp, zk_port, last_name="h4999.foo.com"):
     from binder_amd.harness import BinderProcess
     sockdir = tmp / "socks"
     sockdir.mkdir(exist_ok=True)
@@ -80,7 +84,7 @@ def start_backends(n, tmp, zk_port):
         backends.append(b)
     for b in backends:
         b.wait_listening(timeout=30)
-        b.wait_ready("h4999.foo.com", timeout=60)
+        b.wait_ready(last_name, timeout=120)
     return backends, sockdir
 
 
@@ -136,6 +140,10 @@ def main():
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--queries-per-proc", type=int, default=200_000)
     ap.add_argument("--window", type=int, default=64)
+    ap.add_argument("--tree-records", type=int, default=10_000)
+    ap.add_argument("--churn-qps", type=int, default=0,
+                    help="ZK mutations/sec during timed steps "
+                         "(BASELINE config 5)")
     args = ap.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -186,12 +194,15 @@ def main():
             from binder_amd.stubzk import StubZk
             tmp = Path(tempfile.mkdtemp(prefix="binder-bench-"))
             names_file = tmp / "names.txt"
-            log(f"starting stub ZK + building 10k-record tree")
+            log(f"starting stub ZK + building "
+                f"{args.tree_records}-record tree")
             zk = StubZk().start()
             stack.append(zk.stop)
-            nrec = build_tree(zk, names_file)
+            nrec = build_tree(zk, names_file, args.tree_records)
             log(f"{nrec} records; starting {n} binderd process(es)")
-            backends, sockdir = start_backends(n, tmp, zk.port)
+            backends, sockdir = start_backends(
+                n, tmp, zk.port,
+                last_name=f"h{args.tree_records // 2 - 1}.foo.com")
             stack.append(lambda: [b.stop() for b in backends])
             bal_port = free_port()
             bal = start_balancer(tmp, sockdir, bal_port, workers=workers)
@@ -201,6 +212,31 @@ def main():
                 f"{args.warmup} x {q_step} queries")
             for _ in range(args.warmup):
                 run_blast(bal_port, q_step, names_file, threads, window)
+
+        churn_stop = None
+        if rank == 0 and args.churn_qps > 0:
+            import random
+            import threading
+            churn_stop = threading.Event()
+            n_hosts = args.tree_records // 2
+
+            def churner():
+                rng = random.Random(7)
+                interval = 1.0 / args.churn_qps
+                i = 0
+                while not churn_stop.is_set():
+                    h = rng.randrange(n_hosts)
+                    zk.put(f"/com/foo/h{h}", json.dumps(
+                        {"type": "host",
+                         "host": {"address":
+                                  f"10.{(h >> 8) & 255}.{h & 255}."
+                                  f"{1 + (i % 200)}"}}).encode())
+                    i += 1
+                    time.sleep(interval)
+
+            churn_thread = threading.Thread(target=churner, daemon=True)
+            churn_thread.start()
+            stack.append(lambda: (churn_stop.set(), churn_thread.join()))
 
         barrier()
         cuda_sync()
@@ -241,7 +277,8 @@ def main():
                 "data": "synthetic",
                 "config": {
                     "model": "binder-dns-zk",
-                    "tree_records": 10000,
+                    "tree_records": args.tree_records,
+                    "churn_qps": args.churn_qps,
                     "query_mix": "A+SRV uniform",
                     "queries_per_step": q_step,
                     "global_batch": q_step,
