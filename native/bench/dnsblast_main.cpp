@@ -55,17 +55,8 @@ struct ThreadResult {
     std::vector<uint64_t> latBuckets = std::vector<uint64_t>(kLatBuckets);
 };
 
-/* microsecond -> bucket (log-ish: 1us granularity to 256us, then
- * coarser) */
-inline int latBucket(int64_t us) {
-    if (us < 256) return (int)us;
-    if (us < 256 * 16) return 256 + (int)((us - 256) / 16);  // to 4.3ms
-    if (us < 256 * 16 + 60 * 1000) return 496;  // placeholder (unused)
-    return kLatBuckets - 1;
-}
-
-/* finer mapping: buckets 0..255: 1us each; 256..495: 16us each
- * (4.1ms); 496..510: 100ms range; 511: overflow */
+/* latency mapping: buckets 0..255: 1us each; 256..495: 16us each
+ * (to ~4.1ms); 496..510: 10ms steps; 511: overflow */
 inline int latBucket2(int64_t us) {
     if (us < 256) return (int)us;
     int64_t v = (us - 256) / 16;
