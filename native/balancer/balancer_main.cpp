@@ -154,7 +154,6 @@ class Balancer {
     std::unordered_map<uint64_t, int> remotesFast_;
     std::map<int, std::shared_ptr<TcpClient>> tcpClients_;
     int nextBackendId_ = 1;
-    int rrCursor_ = 0;
     uint64_t udpQueries_ = 0, udpReplies_ = 0, drops_ = 0;
 
     static constexpr int64_t kReplyTtlMs = 3000;

@@ -17,9 +17,21 @@
 namespace py = pybind11;
 using namespace bamd;
 
+/* DNS names from the wire are arbitrary bytes; decode lossily for
+ * Python (invalid UTF-8 becomes U+FFFD) instead of throwing. */
+static py::str pystr(const std::string& s) {
+    PyObject* o = PyUnicode_DecodeUTF8(s.data(), (Py_ssize_t)s.size(),
+                                       "replace");
+    if (o == nullptr) {
+        PyErr_Clear();
+        return py::str("");
+    }
+    return py::reinterpret_steal<py::str>(o);
+}
+
 static py::dict recordToDict(const dns::Record& r) {
     py::dict d;
-    d["name"] = r.name;
+    d["name"] = pystr(r.name);
     d["type"] = dns::typeName(r.type);
     d["class"] = r.rclass;
     d["ttl"] = r.ttl;
@@ -29,14 +41,14 @@ static py::dict recordToDict(const dns::Record& r) {
         d["address"] = r.addrString();
         break;
     case dns::TYPE_SRV:
-        d["target"] = r.target;
+        d["target"] = pystr(r.target);
         d["port"] = r.port;
         d["priority"] = r.priority;
         d["weight"] = r.weight;
         break;
     case dns::TYPE_SOA:
-        d["mname"] = r.soa.mname;
-        d["rname"] = r.soa.rname;
+        d["mname"] = pystr(r.soa.mname);
+        d["rname"] = pystr(r.soa.rname);
         d["serial"] = r.soa.serial;
         d["minimum"] = r.soa.minimum;
         break;
@@ -44,7 +56,7 @@ static py::dict recordToDict(const dns::Record& r) {
         d["udp_size"] = r.rclass;
         break;
     default:
-        d["target"] = r.target;
+        d["target"] = pystr(r.target);
         break;
     }
     return d;
@@ -100,7 +112,7 @@ static py::dict messageToDict(const dns::Message& m) {
     py::list qs;
     for (const auto& q : m.questions) {
         py::dict qd;
-        qd["name"] = q.name;
+        qd["name"] = pystr(q.name);
         qd["type"] = dns::typeName(q.qtype);
         qd["class"] = q.qclass;
         qs.append(qd);

@@ -119,7 +119,6 @@ class DnsServer {
                     const QueryTimers& tm);
 
     void tcpFlush(TcpConn* c);
-    void tcpMaybeClose(TcpConn* c);
     void sweepIdleTcp();
     void balFlush(BalConn* c);
     void closeTcp(TcpConn* c);
