@@ -36,13 +36,15 @@ STATE_CONNECTED = 3
 
 
 class _Node:
-    __slots__ = ("data", "children", "version", "cversion")
+    __slots__ = ("data", "children", "version", "cversion",
+                 "ephemeral_owner")
 
-    def __init__(self, data: bytes = b""):
+    def __init__(self, data: bytes = b"", ephemeral_owner: int = 0):
         self.data = data
         self.children: Set[str] = set()
         self.version = 0
         self.cversion = 0
+        self.ephemeral_owner = ephemeral_owner
 
 
 class _Conn:
@@ -153,9 +155,13 @@ class StubZk:
 
     def expire_sessions(self):
         """Invalidate all sessions and cut connections => clients must
-        build fresh sessions."""
+        build fresh sessions. Ephemeral nodes owned by the expired
+        sessions are reaped (real-ZK semantics)."""
         with self._lock:
+            sessions = set(self._valid_sessions)
             self._valid_sessions.clear()
+            for sid in sessions:
+                self._reap_ephemerals(sid)
             for c in list(self._conns):
                 self._close_conn(c)
         self._wake()
@@ -184,9 +190,10 @@ class StubZk:
             n = self._nodes.get(path)
             return set() if n is None else set(n.children)
 
-    def create(self, path: str, data: bytes = b"null") -> int:
+    def create(self, path: str, data: bytes = b"null",
+               ephemeral_owner: int = 0) -> int:
         with self._lock:
-            rc = self._do_create(path, data)
+            rc = self._do_create(path, data, ephemeral_owner)
         self._wake()
         return rc
 
@@ -250,19 +257,23 @@ class StubZk:
         b = s.encode() if isinstance(s, str) else s
         return struct.pack(">i", len(b)) + b
 
-    def _do_create(self, path: str, data: bytes) -> int:
+    def _do_create(self, path: str, data: bytes,
+                   ephemeral_owner: int = 0) -> int:
         if path in self._nodes:
             return ZNODEEXISTS
         parent = _parent(path)
         pn = self._nodes.get(parent)
         if pn is None:
             return ZNONODE
+        if pn.ephemeral_owner:
+            return -108  # ZNOCHILDRENFOREPHEMERALS
         self._zxid += 1
+        eph = b"\x01" if ephemeral_owner else b"\x00"
         self._txn(1, self._jstr(path) + self._jstr(data) +
                   struct.pack(">i", 1) + struct.pack(">i", 31) +
                   self._jstr("world") + self._jstr("anyone") +
-                  b"\x00" + struct.pack(">i", pn.cversion + 1))
-        self._nodes[path] = _Node(data)
+                  eph + struct.pack(">i", pn.cversion + 1))
+        self._nodes[path] = _Node(data, ephemeral_owner)
         pn.children.add(path[path.rfind("/") + 1:])
         pn.cversion += 1
         self._fire(path, EV_CREATED, exists_only=True)
@@ -355,6 +366,27 @@ class StubZk:
             with self._lock:
                 self._conns.add(c)
             self._sel.register(sock, selectors.EVENT_READ, c)
+
+    def _reap_ephemerals(self, session_id: int):
+        """Delete ephemeral znodes owned by a dead session (lock held),
+        firing watches — the production mechanism by which dead
+        registrars vanish from service discovery."""
+        if not session_id:
+            return
+        doomed = [p for p, n in self._nodes.items()
+                  if n.ephemeral_owner == session_id]
+        for p in sorted(doomed, key=len, reverse=True):
+            self._do_delete(p)
+
+    def expire_session(self, session_id: int):
+        """Expire one session: reap its ephemerals + drop its conns."""
+        with self._lock:
+            self._valid_sessions.discard(session_id)
+            self._reap_ephemerals(session_id)
+            for c in list(self._conns):
+                if c.session_id == session_id:
+                    self._close_conn(c)
+        self._wake()
 
     def _close_conn(self, c: _Conn):
         if c.closed:
@@ -515,8 +547,9 @@ class StubZk:
                 r.i32()
                 r.s()
                 r.s()
-            r.i32()  # flags (ephemeral/sequential unsupported in stub)
-            rc = self._do_create(path, data)
+            flags = r.i32()
+            owner = c.session_id if (flags & 1) else 0  # EPHEMERAL
+            rc = self._do_create(path, data, owner)
             self._reply(c, xid, rc, _s(path) if rc == ZOK else b"")
             return
 

@@ -71,6 +71,25 @@ struct GlobalPins {
 };
 GlobalPins g_pins;
 
+/* Per-worker stats snapshots, refreshed each sweep; worker 0 serves
+ * the merged view on the stats socket (an mdb-free balstat needs a
+ * whole-process picture, not one shard's). */
+struct StatsHub {
+    std::mutex m;
+    std::map<int, std::string> byWorker;  // worker id -> JSON fragment
+    void publish(int worker, std::string json) {
+        std::lock_guard<std::mutex> g(m);
+        byWorker[worker] = std::move(json);
+    }
+    std::vector<std::string> all() {
+        std::lock_guard<std::mutex> g(m);
+        std::vector<std::string> out;
+        for (auto& [w, j] : byWorker) out.push_back(j);
+        return out;
+    }
+};
+StatsHub g_stats;
+
 struct PendingReply {
     struct sockaddr_storage src;
     socklen_t srcLen;
@@ -110,11 +129,11 @@ class Balancer {
   public:
     Balancer(EventLoop* loop, Logger log, std::string host, uint16_t port,
              std::string sockDir, std::string statsPath, int rescanMs,
-             bool reusePort = false)
+             bool reusePort = false, int workerId = 0)
         : loop_(loop), log_(std::move(log)), host_(std::move(host)),
           port_(port), sockDir_(std::move(sockDir)),
           statsPath_(std::move(statsPath)), rescanMs_(rescanMs),
-          reusePort_(reusePort) {}
+          reusePort_(reusePort), workerId_(workerId) {}
 
     bool start();
     void stop();
@@ -134,6 +153,7 @@ class Balancer {
     void tcpClientFlush(TcpClient* c);
     void onStatsAccept();
     void sweep();
+    Json snapshot() const;
 
     EventLoop* loop_;
     Logger log_;
@@ -143,6 +163,7 @@ class Balancer {
     std::string statsPath_;
     int rescanMs_;
     bool reusePort_ = false;
+    int workerId_ = 0;
 
     int udpFd_ = -1, tcpFd_ = -1, statsFd_ = -1;
     std::map<std::string, std::shared_ptr<Backend>> backends_;  // by path
@@ -740,35 +761,84 @@ void Balancer::onTcpClient(std::shared_ptr<TcpClient> c, uint32_t ev) {
     }
 }
 
+Json Balancer::snapshot() const {
+    Json out = Json::object();
+    JsonArray bes;
+    for (auto& [path, be] : backends_) {
+        Json b = Json::object();
+        b.set("id", Json((int64_t)be->id));
+        b.set("path", Json(be->path));
+        b.set("ok", Json(be->ok));
+        b.set("remotes", Json((int64_t)be->remotes));
+        b.set("queries", Json((int64_t)be->queries));
+        b.set("replies", Json((int64_t)be->replies));
+        b.set("pending", Json((int64_t)be->pending.size()));
+        bes.push_back(std::move(b));
+    }
+    out.set("backends", Json(std::move(bes)));
+    JsonArray rms;
+    for (auto& [ip, id] : remotes_) {
+        Json r = Json::object();
+        r.set("addr", Json(ip));
+        r.set("backend", Json((int64_t)id));
+        rms.push_back(std::move(r));
+    }
+    out.set("remotes", Json(std::move(rms)));
+    out.set("udp_queries", Json((int64_t)udpQueries_));
+    out.set("udp_replies", Json((int64_t)udpReplies_));
+    out.set("drops", Json((int64_t)drops_));
+    return out;
+}
+
 void Balancer::onStatsAccept() {
     while (true) {
         int fd = accept4(statsFd_, nullptr, nullptr, SOCK_CLOEXEC);
         if (fd < 0) return;
+        /* merge the latest snapshot from every worker (our own fresh;
+         * others as of their last sweep tick) */
+        g_stats.publish(workerId_, snapshot().dump());
+        std::map<std::string, Json> byPath;      // merged backends
+        JsonArray remotes;
+        int64_t udpQ = 0, udpR = 0, drops = 0;
+        int workers = 0;
+        for (const std::string& frag : g_stats.all()) {
+            auto parsed = Json::parse(frag);
+            if (!parsed) continue;
+            workers++;
+            udpQ += parsed->get("udp_queries").asInt();
+            udpR += parsed->get("udp_replies").asInt();
+            drops += parsed->get("drops").asInt();
+            for (const auto& r : parsed->get("remotes").items())
+                remotes.push_back(r);
+            for (const auto& b : parsed->get("backends").items()) {
+                const std::string& path = b.get("path").asString();
+                auto it = byPath.find(path);
+                if (it == byPath.end()) {
+                    byPath[path] = b;
+                } else {
+                    Json& m = it->second;
+                    m.set("remotes", Json(m.get("remotes").asInt() +
+                                          b.get("remotes").asInt()));
+                    m.set("queries", Json(m.get("queries").asInt() +
+                                          b.get("queries").asInt()));
+                    m.set("replies", Json(m.get("replies").asInt() +
+                                          b.get("replies").asInt()));
+                    m.set("pending", Json(m.get("pending").asInt() +
+                                          b.get("pending").asInt()));
+                    m.set("ok", Json(m.get("ok").asBool() ||
+                                     b.get("ok").asBool()));
+                }
+            }
+        }
         Json out = Json::object();
         JsonArray bes;
-        for (auto& [path, be] : backends_) {
-            Json b = Json::object();
-            b.set("id", Json((int64_t)be->id));
-            b.set("path", Json(be->path));
-            b.set("ok", Json(be->ok));
-            b.set("remotes", Json((int64_t)be->remotes));
-            b.set("queries", Json((int64_t)be->queries));
-            b.set("replies", Json((int64_t)be->replies));
-            b.set("pending", Json((int64_t)be->pending.size()));
-            bes.push_back(std::move(b));
-        }
+        for (auto& [path, b] : byPath) bes.push_back(std::move(b));
         out.set("backends", Json(std::move(bes)));
-        JsonArray rms;
-        for (auto& [ip, id] : remotes_) {
-            Json r = Json::object();
-            r.set("addr", Json(ip));
-            r.set("backend", Json((int64_t)id));
-            rms.push_back(std::move(r));
-        }
-        out.set("remotes", Json(std::move(rms)));
-        out.set("udp_queries", Json((int64_t)udpQueries_));
-        out.set("udp_replies", Json((int64_t)udpReplies_));
-        out.set("drops", Json((int64_t)drops_));
+        out.set("remotes", Json(std::move(remotes)));
+        out.set("udp_queries", Json(udpQ));
+        out.set("udp_replies", Json(udpR));
+        out.set("drops", Json(drops));
+        out.set("workers", Json((int64_t)workers));
         std::string s = out.dump();
         s.push_back('\n');
         ssize_t rv = write(fd, s.data(), s.size());
@@ -811,6 +881,7 @@ void Balancer::sweep() {
             backendDown(be.get());
         }
     }
+    g_stats.publish(workerId_, snapshot().dump());
 }
 
 }  // namespace
@@ -866,7 +937,7 @@ int main(int argc, char** argv) {
         bals.emplace_back(std::make_unique<Balancer>(
             loops[w].get(),
             log.child({{"worker", Json((int64_t)w)}}), host, port, dir,
-            w == 0 ? stats : std::string(), rescanMs, workers > 1));
+            w == 0 ? stats : std::string(), rescanMs, workers > 1, w));
         if (!bals[w]->start()) return 1;
     }
     for (int w = 0; w < workers; ++w)

@@ -141,3 +141,31 @@ def test_larger_tree_count(stack):
     srv.wait_ready("h199.foo.com", timeout=15)
     assert srv.dig("h0.foo.com").status == "NOERROR"
     assert srv.dig("h123.foo.com").answers[0]["address"] == "10.42.0.123"
+
+
+def test_ephemeral_nodes_vanish_on_session_expiry(stack):
+    """Production registrars create EPHEMERAL znodes; when their session
+    dies, ZK deletes the node and binder must stop serving it."""
+    zk, srv = stack
+    srv.wait_ready("web.bar.foo.com")
+
+    # a "registrar" session holding an ephemeral registration
+    from binder_amd.zkclient import ZkConn
+    reg = ZkConn("127.0.0.1", zk.port)
+    import json as _json
+    reg.create("/com/foo/eph", _json.dumps(
+        {"type": "host", "host": {"address": "10.66.0.1"}}).encode(),
+        flags=1)
+    srv.wait_ready("eph.foo.com")
+
+    # registrar dies (session expired server-side)
+    zk.expire_session(reg.session_id)
+    deadline = time.time() + 10
+    while time.time() < deadline:
+        if srv.dig("eph.foo.com").status == "REFUSED":
+            break
+        time.sleep(0.1)
+    else:
+        pytest.fail("ephemeral registration never vanished")
+    # PTR gone too (reverse-map cleanup)
+    assert srv.dig("1.0.66.10.in-addr.arpa", "PTR").status == "REFUSED"
