@@ -20,7 +20,9 @@ from typing import Optional
 from . import REPO_ROOT
 from .digclient import dig
 
-BINDERD = REPO_ROOT / "bin" / "binderd"
+# BINDERD_BIN overrides the server binary (e.g. an ASan build)
+BINDERD = Path(os.environ.get("BINDERD_BIN",
+                              REPO_ROOT / "bin" / "binderd"))
 BALANCERD = REPO_ROOT / "bin" / "binder-balancer"
 
 

@@ -67,6 +67,12 @@ QueryResult Engine::handle(const Message& query, Message& resp) {
         resp.header.rcode = RCODE_FORMERR;
         return qr;
     }
+    if (query.header.opcode != 0) {
+        /* only standard QUERY is implemented (IQUERY/STATUS/NOTIFY/
+         * UPDATE are not) */
+        resp.header.rcode = RCODE_NOTIMP;
+        return qr;
+    }
     const Question& q = query.questions[0];
     qr.logName = q.name;
     bool rd = query.header.rd;

@@ -391,3 +391,17 @@ def test_url_hostname():
         "ufds.coal.joyent.us"
     assert n.url_hostname("http://[::1]:8080/x") == "::1"
     assert n.url_hostname("10.0.0.1:70") == "10.0.0.1"
+
+
+def test_non_query_opcode_notimp():
+    e = mkengine()
+    put(e, "foo.com", None)
+    put(e, "web.foo.com", HOST)
+    import binder_amd
+    n = binder_amd.require_native()
+    wire = bytearray(n.encode_message(
+        {"id": 5, "questions": [{"name": "web.foo.com", "type": "A"}]}))
+    wire[2] |= 0x28  # opcode 5 (UPDATE)
+    resp_wire, action = e.query_wire(bytes(wire), 512)
+    resp = n.decode_message(resp_wire)
+    assert resp["rcode"] == "NOTIMP"
