@@ -68,6 +68,12 @@ struct CompiledRecord {
 
     /* member-only: per-member port list (server.js:383-385) */
     std::vector<uint16_t> ports;
+
+    /* binderd fast-path cache: prebuilt wire response for an
+     * all-lowercase A/IN query of this node's domain (built lazily in
+     * server.cpp). A fresh CompiledRecord starts empty, so every
+     * recompile (mirror update) invalidates it automatically. */
+    mutable std::vector<uint8_t> wireA;
 };
 
 /* Compile a znode JSON payload. `parsed`=false when payload was not valid

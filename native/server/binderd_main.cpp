@@ -186,6 +186,7 @@ int main(int argc, char** argv) {
     sopts.balancerSocket = opts.get("balancerSocket").asString();
 
     DnsServer server(&loop, log, sopts, &engine, &collector);
+    server.setStore(store);
 
     /* recursion (lib/recursion.js equivalent; config block per
      * sapi_manifests/binder/template) */

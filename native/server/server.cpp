@@ -221,10 +221,108 @@ static inline int64_t nowUs() {
     return (int64_t)ts.tv_sec * 1000000 + ts.tv_nsec / 1000;
 }
 
+/*
+ * Fast path: all-lowercase A/IN query with no EDNS for a host-like or
+ * database node => serve a prebuilt wire response (header id/rd
+ * patched). Falls back to the full path for anything else, including
+ * whenever info-level logging is on (the per-query log line needs the
+ * decoded message). Policy checks mirror Engine::resolve exactly for
+ * the subset they cover; anything non-trivial bails to slow.
+ */
+bool DnsServer::fastPath(const uint8_t* data, size_t len,
+                         std::vector<uint8_t>& out) {
+    if (log_.enabled(LogLevel::Info)) return false;
+    if (len < 17 || len > 300) return false;
+    /* flags: QR/opcode/AA/TC clear, RD free; byte 3 must be zero */
+    if ((data[2] & 0xFE) != 0 || data[3] != 0) return false;
+    /* counts: qd=1, an=ns=ar=0 */
+    if (data[4] != 0 || data[5] != 1 || data[6] | data[7] ||
+        data[8] | data[9] || data[10] | data[11])
+        return false;
+
+    char name[256];
+    size_t nlen = 0;
+    size_t pos = 12;
+    while (true) {
+        if (pos >= len) return false;
+        uint8_t l = data[pos++];
+        if (l == 0) break;
+        if ((l & 0xC0) != 0 || pos + l > len) return false;
+        if (nlen + l + 1 > sizeof(name)) return false;
+        if (nlen > 0) name[nlen++] = '.';
+        for (uint8_t i = 0; i < l; ++i) {
+            char c = (char)data[pos + i];
+            bool ok = (c >= 'a' && c <= 'z') || (c >= '0' && c <= '9') ||
+                      c == '_' || c == '-';
+            if (!ok) return false;  // uppercase/odd chars: slow path
+            name[nlen++] = c;
+        }
+        pos += l;
+    }
+    if (pos + 4 > len) return false;
+    if (data[pos] != 0 || data[pos + 1] != 1 ||  /* qtype A */
+        data[pos + 2] != 0 || data[pos + 3] != 1)  /* class IN */
+        return false;
+
+    std::string_view key(name, nlen);
+    const EngineConfig& cfg = engine_->config();
+    if (!cfg.dnsDomain.empty()) {
+        /* must be a strict subdomain of dnsDomain, not doubled */
+        size_t dl = cfg.dnsDomain.size();
+        if (key.size() <= dl + 1) return false;
+        if (key[key.size() - dl - 1] != '.' ||
+            key.substr(key.size() - dl) != cfg.dnsDomain)
+            return false;
+        std::string_view stripped = key.substr(0, key.size() - dl - 1);
+        auto endsWith = [](std::string_view s, std::string_view suf) {
+            return s.size() >= suf.size() &&
+                   s.substr(s.size() - suf.size()) == suf;
+        };
+        if (endsWith(stripped, cfg.dnsDomain)) return false;
+        std::string dcsuff = cfg.dnsDomain + "." + cfg.datacenterName;
+        if (endsWith(stripped, dcsuff)) return false;
+    }
+
+    const Store* store = engineStore_;
+    if (store == nullptr || !store->ready()) return false;
+    const StoreNode* node = store->lookup(std::string(key));
+    if (node == nullptr) return false;
+    const CompiledRecord& rec = node->rec();
+    if (!rec.valid || rec.address.empty()) return false;
+    bool hostish = recTypeIsHostLike(rec.type) ||
+                   rec.type == RecType::Database;
+    if (!hostish) return false;
+
+    if (rec.wireA.empty()) {
+        /* build: header + question + answer (name = ptr to offset 12) */
+        dns::Message m;
+        m.header.qr = true;
+        m.header.aa = true;
+        dns::Question q;
+        q.name = std::string(key);
+        m.questions.push_back(std::move(q));
+        m.answers.push_back(
+            dns::Record::A(std::string(key), rec.address, rec.ttl));
+        rec.wireA = m.encode(0);
+    }
+    out = rec.wireA;
+    out[0] = data[0];  /* id */
+    out[1] = data[1];
+    out[2] = (uint8_t)(out[2] | (data[2] & 0x01));  /* echo RD */
+
+    ++served_;
+    static const std::string kLabelA = "type=\"A\"";
+    reqCounter_->increment(kLabelA);
+    latHist_->observe(kLabelA, 1e-6);  /* sub-us; below first bucket */
+    sizeHist_->observe(kLabelA, (double)out.size());
+    return true;
+}
+
 bool DnsServer::process(const uint8_t* data, size_t len, bool udp,
                         const ClientInfo& ci, std::vector<uint8_t>& out,
                         std::function<void(std::vector<uint8_t>)>
                             asyncReply) {
+    if (fastPath(data, len, out)) return true;
     int64_t start = nowUs();
     BAMD_PROBE2(op_req_start, len, (int)udp);
     auto parsed = Message::decode(data, len);

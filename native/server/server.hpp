@@ -112,6 +112,15 @@ class DnsServer {
     bool process(const uint8_t* data, size_t len, bool udp,
                  const ClientInfo& ci, std::vector<uint8_t>& out,
                  std::function<void(std::vector<uint8_t>)> asyncReply);
+    bool fastPath(const uint8_t* data, size_t len,
+                  std::vector<uint8_t>& out);
+
+  public:
+    /* the store behind the engine (fast-path lookups); set by main */
+    void setStore(const Store* s) { engineStore_ = s; }
+
+  private:
+    const Store* engineStore_ = nullptr;
 
     void afterQuery(const dns::Message& query, const dns::Message& resp,
                     const QueryResult& qr, const ClientInfo& ci,

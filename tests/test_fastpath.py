@@ -1,0 +1,108 @@
+"""Fast-path (prebuilt wire) parity: answers must be byte-equivalent in
+meaning to the slow path, and caches must invalidate on updates.
+
+The fast path only engages when info logging is off (LOG_LEVEL=warn),
+so run twin servers and diff responses.
+"""
+import json
+import time
+
+import pytest
+
+from binder_amd.harness import BinderProcess
+from binder_amd.stubzk import StubZk
+
+TREE = {
+    "foo.com": None,
+    "web.foo.com": {"type": "host", "host": {"address": "1.2.3.4"}},
+    "db.foo.com": {"type": "database",
+                   "database": {"primary": "tcp://u@9.8.7.6:5432/x"},
+                   "ttl": 20},
+    "svc.foo.com": {"type": "service",
+                    "service": {"srvce": "_x", "proto": "_tcp",
+                                "port": 1, "ttl": 60}},
+    "m0.svc.foo.com": {"type": "rr_host",
+                       "rr_host": {"address": "10.0.0.9"}},
+}
+
+
+@pytest.fixture()
+def twins(tmp_path):
+    store = tmp_path / "tree.json"
+    store.write_text(json.dumps(TREE))
+    fast = BinderProcess(store=f"file:{store}", workdir=tmp_path,
+                         log_level="warn")
+    slow = BinderProcess(store=f"file:{store}", workdir=tmp_path,
+                         log_level="info")
+    fast.start()
+    slow.start()
+    yield fast, slow
+    fast.stop()
+    slow.stop()
+
+
+def normalize(r):
+    return (r.status, r["aa"], r["ra"], r["tc"],
+            sorted((a["type"], a.get("address"), a["ttl"], a["name"])
+                   for a in r.answers))
+
+
+@pytest.mark.parametrize("name,qtype,rd", [
+    ("web.foo.com", "A", False),
+    ("web.foo.com", "A", True),
+    ("WEB.foo.com", "A", False),       # uppercase: both slow
+    ("db.foo.com", "A", False),
+    ("svc.foo.com", "A", False),       # service: fast path bails
+    ("missing.foo.com", "A", False),
+    ("web.foo.com", "SRV", False),
+    ("foo.com", "A", False),           # apex refused
+    ("web.foo.com.foo.com", "A", False),
+])
+def test_fast_equals_slow(twins, name, qtype, rd):
+    fast, slow = twins
+    rf = fast.dig(name, qtype, rd=rd, qid=777)
+    rs = slow.dig(name, qtype, rd=rd, qid=777)
+    assert normalize(rf) == normalize(rs), (name, qtype, rd)
+    assert rf["id"] == 777
+    assert rf["rd"] == rd
+
+
+def test_edns_still_served_with_fastpath_on(twins):
+    fast, _ = twins
+    r = fast.dig("web.foo.com", edns=4096)
+    assert r.status == "NOERROR"
+    assert any(x["type"] == "OPT" for x in r["additionals"])
+
+
+def test_cache_invalidation_on_update(tmp_path):
+    zk = StubZk().start()
+    try:
+        zk.mkdirp("/com/foo")
+        zk.put("/com/foo/web", json.dumps(
+            {"type": "host", "host": {"address": "1.1.1.1"}}).encode())
+        srv = BinderProcess(store="zk", zk_host="127.0.0.1",
+                            zk_port=zk.port, workdir=tmp_path,
+                            log_level="warn")
+        srv.start()
+        try:
+            srv.wait_ready("web.foo.com")
+            # prime the fast-path cache
+            for _ in range(5):
+                r = srv.dig("web.foo.com")
+                assert r.answers[0]["address"] == "1.1.1.1"
+            zk.put("/com/foo/web", json.dumps(
+                {"type": "host",
+                 "host": {"address": "2.2.2.2", "ttl": 77}}).encode())
+            deadline = time.time() + 5
+            while time.time() < deadline:
+                r = srv.dig("web.foo.com")
+                if r.answers and r.answers[0]["address"] == "2.2.2.2":
+                    break
+                time.sleep(0.05)
+            else:
+                pytest.fail("cached answer never invalidated")
+            assert r.answers[0]["ttl"] == 77
+        finally:
+            srv.stop()
+    finally:
+        zk.stop()
