@@ -88,6 +88,7 @@ struct Config {
     std::string namesFile;
     std::string bindBase;
     int timeoutMs = 2000;
+    bool rd = false;  // set RD (exercises recursion paths)
 };
 
 void worker(const Config& cfg, int tid,
@@ -253,8 +254,9 @@ void worker(const Config& cfg, int tid,
 int main(int argc, char** argv) {
     Config cfg;
     int c;
-    while ((c = getopt(argc, argv, "hs:p:n:c:t:f:B:T:")) != -1) {
+    while ((c = getopt(argc, argv, "hs:p:n:c:t:f:B:T:R")) != -1) {
         switch (c) {
+        case 'R': cfg.rd = true; break;
         case 's': cfg.server = optarg; break;
         case 'p': cfg.port = (uint16_t)atoi(optarg); break;
         case 'n': cfg.queries = strtoull(optarg, nullptr, 10); break;
@@ -268,7 +270,7 @@ int main(int argc, char** argv) {
             fprintf(stderr,
                     "usage: dnsblast -s server -p port -n queries "
                     "[-c window] [-t threads] [-f names-file] "
-                    "[-B bind-base-ip] [-T timeout-ms]\n");
+                    "[-B bind-base-ip] [-T timeout-ms] [-R]\n");
             return c == 'h' ? 0 : 1;
         }
     }
@@ -296,7 +298,7 @@ int main(int argc, char** argv) {
     for (const auto& e : names) {
         dns::Message q;
         q.header.id = 0;
-        q.header.rd = false;
+        q.header.rd = cfg.rd;
         dns::Question qq;
         qq.name = e.name;
         qq.qtype = e.qtype;
