@@ -95,3 +95,55 @@ def test_upstream_sees_rd_cleared(dcs):
     time.sleep(0.2)
     log = (upstream.log_path and open(upstream.log_path).read()) or ""
     assert "svc.dc2.foo.com" in log
+
+
+def test_ptr_recursion_fans_out(dcs):
+    """PTR misses fan out to all known DC resolvers
+    (recursion.js:346-354) and accepted PTR answers come back under the
+    original name."""
+    local, upstream = dcs
+    r = local.dig("1.0.22.10.in-addr.arpa", "PTR", rd=True, timeout=6)
+    assert r.status == "NOERROR"
+    assert r.answers[0]["type"] == "PTR"
+    assert r.answers[0]["target"] == "svc.dc2.foo.com"
+    assert r.answers[0]["name"] == "1.0.22.10.in-addr.arpa"
+
+
+def test_srv_forwarding_passthrough(tmp_path):
+    """SRV answers from the upstream are re-materialized with
+    target/port/priority/weight (recursion.js:311-315)."""
+    import json as _json
+    up_tree = tmp_path / "up2.json"
+    up_tree.write_text(_json.dumps({
+        "dc2.foo.com": None,
+        "s.dc2.foo.com": {"type": "service",
+                          "service": {"srvce": "_m", "proto": "_tcp",
+                                      "port": 123, "ttl": 60}},
+        "m0.s.dc2.foo.com": {"type": "rr_host",
+                             "rr_host": {"address": "10.22.9.9"}},
+    }))
+    upstream = BinderProcess(dns_domain="dc2.foo.com", datacenter="dc2",
+                             host="127.0.0.3", store=f"file:{up_tree}",
+                             workdir=tmp_path)
+    upstream.start()
+    local_tree = tmp_path / "local2.json"
+    local_tree.write_text("{\"foo.com\": null}")
+    local = BinderProcess(
+        dns_domain="foo.com", datacenter="dc1",
+        store=f"file:{local_tree}", workdir=tmp_path,
+        config={"recursion": {
+            "source": "static", "regionName": "r1",
+            "dnsDomain": "foo.com", "upstreamPort": upstream.port,
+            "dcs": {"dc2": ["127.0.0.3"]}}})
+    local.start()
+    try:
+        r = local.dig("_m._tcp.s.dc2.foo.com", "SRV", rd=True, timeout=6)
+        assert r.status == "NOERROR"
+        a = r.answers[0]
+        assert a["type"] == "SRV"
+        assert a["port"] == 123
+        assert a["target"] == "m0.s.dc2.foo.com"
+        assert a["name"] == "_m._tcp.s.dc2.foo.com"
+    finally:
+        local.stop()
+        upstream.stop()
