@@ -352,8 +352,20 @@ void Recursion::resolve(const Message& query, Message& resp,
     }
 
     auto up = std::make_shared<Upstream>();
-    up->qid = nextQid_ == 0 ? ++nextQid_ : nextQid_;
-    nextQid_ += 1;
+    /* skip qids still in flight (collision would cross-cancel) */
+    uint16_t qid = nextQid_;
+    for (int tries = 0; tries < 8; ++tries) {
+        if (qid == 0) qid++;
+        if (pendingByQid_.count(qid) == 0) break;
+        qid++;
+    }
+    nextQid_ = (uint16_t)(qid + 1);
+    if (pendingByQid_.count(qid) != 0) {
+        /* >64k concurrent recursions: shed load (best effort) */
+        refuse();
+        return;
+    }
+    up->qid = qid;
     up->hosts = std::move(filtered);
     up->maxConcurrency = isPtr ? 100 : 2;  // recursion.js:64-78
     up->resp = &resp;
