@@ -272,7 +272,11 @@ def main():
                 "ms_per_step": round(elapsed * 1000 / args.steps, 3),
                 "higher_is_better": True,
                 "scaling": "weak",
-                "vs_baseline": None,
+                # measured reference-architecture Node.js baseline:
+                # 138,545 qps/process on the driver box class
+                # (BASELINE.md "Measured baseline"); weak scaling =>
+                # baseline scales with N
+                "vs_baseline": round(qps / (n * 138545.3), 2),
                 "dtype": "n/a",
                 "data": "synthetic",
                 "config": {
