@@ -123,3 +123,34 @@ def test_malformed_packet_ignored(server):
             s.recvfrom(512)
     # server still alive
     assert server.dig("web.bar.foo.com").status == "NOERROR"
+
+
+def test_tcp_pipelining(server):
+    """Multiple length-prefixed queries on one TCP connection, answered
+    in order (DNS-over-TCP pipelining)."""
+    import socket
+    import struct
+    from binder_amd import require_native
+    n = require_native()
+    with socket.socket() as s:
+        s.settimeout(3)
+        s.connect((server.host, server.port))
+        blob = b""
+        for i in range(5):
+            w = n.encode_message(
+                {"id": 100 + i,
+                 "questions": [{"name": "web.bar.foo.com",
+                                "type": "A"}]})
+            blob += struct.pack(">H", len(w)) + w
+        s.sendall(blob)
+        for i in range(5):
+            hdr = b""
+            while len(hdr) < 2:
+                hdr += s.recv(2 - len(hdr))
+            (rlen,) = struct.unpack(">H", hdr)
+            data = b""
+            while len(data) < rlen:
+                data += s.recv(rlen - len(data))
+            m = n.decode_message(data)
+            assert m["id"] == 100 + i
+            assert m["rcode"] == "NOERROR"
