@@ -180,6 +180,10 @@ def main():
         workers = min(16, max(4, 2 * n))
         threads = max(8, 4 * n)
         window = 192
+        # experiment overrides (profiling/tuning only)
+        workers = int(os.environ.get("BENCH_WORKERS", workers))
+        threads = int(os.environ.get("BENCH_THREADS", threads))
+        window = int(os.environ.get("BENCH_WINDOW", window))
     else:
         workers = 1
         threads = min(4 * n, max(2, ncpu // 2))
