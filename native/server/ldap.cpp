@@ -1,6 +1,7 @@
 #include "ldap.hpp"
 
 #include <arpa/inet.h>
+#include <netdb.h>
 #include <netinet/in.h>
 #include <netinet/tcp.h>
 #include <openssl/err.h>
@@ -167,9 +168,19 @@ bool Client::connect() {
     sa.sin_family = AF_INET;
     sa.sin_port = htons(opts_.port);
     if (inet_pton(AF_INET, opts_.host.c_str(), &sa.sin_addr) != 1) {
-        err_ = "host is not an IPv4 address: " + opts_.host;
-        close();
-        return false;
+        struct addrinfo hints {};
+        hints.ai_family = AF_INET;
+        hints.ai_socktype = SOCK_STREAM;
+        struct addrinfo* res = nullptr;
+        int rc = getaddrinfo(opts_.host.c_str(), nullptr, &hints, &res);
+        if (rc != 0 || res == nullptr) {
+            err_ = "cannot resolve host: " + opts_.host;
+            if (res) freeaddrinfo(res);
+            close();
+            return false;
+        }
+        sa.sin_addr = ((struct sockaddr_in*)res->ai_addr)->sin_addr;
+        freeaddrinfo(res);
     }
     if (::connect(fd_, (struct sockaddr*)&sa, sizeof(sa)) != 0) {
         err_ = std::string("connect: ") + strerror(errno);

@@ -1,6 +1,7 @@
 #include "client.hpp"
 
 #include <arpa/inet.h>
+#include <netdb.h>
 #include <netinet/in.h>
 #include <netinet/tcp.h>
 #include <sys/epoll.h>
@@ -94,16 +95,23 @@ void ZkClient::connectStart() {
     sa.sin_family = AF_INET;
     sa.sin_port = htons(cfg_.port);
     if (inet_pton(AF_INET, cfg_.host.c_str(), &sa.sin_addr) != 1) {
-        /* loopback name fallback */
-        if (cfg_.host == "localhost")
-            inet_pton(AF_INET, "127.0.0.1", &sa.sin_addr);
-        else {
+        /* resolve hostnames (ZK_HOST may not be an IP); blocking, but
+         * only on the (re)connect path */
+        struct addrinfo hints {};
+        hints.ai_family = AF_INET;
+        hints.ai_socktype = SOCK_STREAM;
+        struct addrinfo* res = nullptr;
+        int rc = getaddrinfo(cfg_.host.c_str(), nullptr, &hints, &res);
+        if (rc != 0 || res == nullptr) {
             log_.error({{"host", Json(cfg_.host)}},
-                       "ZK host is not an IPv4 address");
+                       "cannot resolve ZK host");
+            if (res) freeaddrinfo(res);
             teardown();
             scheduleReconnect();
             return;
         }
+        sa.sin_addr = ((struct sockaddr_in*)res->ai_addr)->sin_addr;
+        freeaddrinfo(res);
     }
     int rv = ::connect(fd_, (struct sockaddr*)&sa, sizeof(sa));
     if (rv != 0 && errno != EINPROGRESS) {
