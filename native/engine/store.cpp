@@ -280,6 +280,11 @@ const StoreNode* StubStore::lookup(const std::string& domain) const {
     return it == nodes_.end() ? nullptr : it->second.get();
 }
 
+const StoreNode* StubStore::lookupView(std::string_view domain) const {
+    auto it = nodes_.find(domain);
+    return it == nodes_.end() ? nullptr : it->second.get();
+}
+
 const StoreNode* StubStore::reverseLookup(const std::string& ip) const {
     auto it = rev_.find(ip);
     return it == rev_.end() ? nullptr : it->second;

@@ -15,6 +15,7 @@
 #include <memory>
 #include <optional>
 #include <string>
+#include <string_view>
 #include <unordered_map>
 #include <vector>
 
@@ -89,10 +90,31 @@ class StoreNode {
     virtual std::vector<const StoreNode*> children() const = 0;
 };
 
+/* heterogeneous lookup support: find by string_view without building
+ * a std::string per query */
+struct SvHash {
+    using is_transparent = void;
+    size_t operator()(std::string_view sv) const {
+        return std::hash<std::string_view>{}(sv);
+    }
+    size_t operator()(const std::string& s) const {
+        return std::hash<std::string_view>{}(s);
+    }
+};
+struct SvEq {
+    using is_transparent = void;
+    bool operator()(std::string_view a, std::string_view b) const {
+        return a == b;
+    }
+};
+
 class Store {
   public:
     virtual ~Store() = default;
     virtual const StoreNode* lookup(const std::string& domain) const = 0;
+    virtual const StoreNode* lookupView(std::string_view domain) const {
+        return lookup(std::string(domain));
+    }
     virtual const StoreNode* reverseLookup(const std::string& ip) const = 0;
     virtual bool ready() const = 0;
 };
@@ -124,6 +146,7 @@ class StubStore : public Store {
     void setReady(bool r) { ready_ = r; }
 
     const StoreNode* lookup(const std::string& domain) const override;
+    const StoreNode* lookupView(std::string_view domain) const override;
     const StoreNode* reverseLookup(const std::string& ip) const override;
     bool ready() const override { return ready_; }
 
@@ -132,7 +155,8 @@ class StubStore : public Store {
     Node* ensure(const std::string& domain);
     void reindex(Node* n, const std::string& oldAddr);
 
-    std::unordered_map<std::string, std::unique_ptr<Node>> nodes_;
+    std::unordered_map<std::string, std::unique_ptr<Node>, SvHash, SvEq>
+        nodes_;
     std::unordered_map<std::string, Node*> rev_;
     bool ready_ = true;
 };

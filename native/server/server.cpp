@@ -285,7 +285,7 @@ bool DnsServer::fastPath(const uint8_t* data, size_t len,
 
     const Store* store = engineStore_;
     if (store == nullptr || !store->ready()) return false;
-    const StoreNode* node = store->lookup(std::string(key));
+    const StoreNode* node = store->lookupView(key);
     if (node == nullptr) return false;
     const CompiledRecord& rec = node->rec();
     if (!rec.valid || rec.address.empty()) return false;

@@ -60,6 +60,11 @@ const StoreNode* ZkMirror::lookup(const std::string& domain) const {
     return it == byDomain_.end() ? nullptr : it->second;
 }
 
+const StoreNode* ZkMirror::lookupView(std::string_view domain) const {
+    auto it = byDomain_.find(domain);
+    return it == byDomain_.end() ? nullptr : it->second;
+}
+
 const StoreNode* ZkMirror::reverseLookup(const std::string& ip) const {
     auto it = rev_.find(ip);
     return it == rev_.end() ? nullptr : it->second;

@@ -61,6 +61,7 @@ class ZkMirror : public Store {
 
     /* Store */
     const StoreNode* lookup(const std::string& domain) const override;
+    const StoreNode* lookupView(std::string_view domain) const override;
     const StoreNode* reverseLookup(const std::string& ip) const override;
     bool ready() const override;
 
@@ -84,7 +85,7 @@ class ZkMirror : public Store {
     std::unique_ptr<zk::ZkClient> client_;
 
     std::unordered_map<std::string, std::unique_ptr<Node>> byPath_;
-    std::unordered_map<std::string, Node*> byDomain_;
+    std::unordered_map<std::string, Node*, SvHash, SvEq> byDomain_;
     std::unordered_map<std::string, Node*> rev_;
     Node* root_ = nullptr;
 
