@@ -78,7 +78,7 @@ class BinderProcess:
     def metrics_port(self) -> int:
         return self.config["metricsPort"]
 
-    def start(self, wait_ready=True, timeout=10.0):
+    def start(self, wait_ready=True, timeout=20.0):
         logf = open(self.log_path, "ab") if self.log_path \
             else subprocess.DEVNULL
         self.proc = subprocess.Popen(
@@ -87,7 +87,7 @@ class BinderProcess:
             self.wait_listening(timeout)
         return self
 
-    def wait_listening(self, timeout=10.0):
+    def wait_listening(self, timeout=20.0):
         """Wait until the server answers DNS at all (any rcode)."""
         deadline = time.time() + timeout
         last_err = None
@@ -104,7 +104,7 @@ class BinderProcess:
                 time.sleep(0.05)
         raise TimeoutError(f"binderd not listening: {last_err}")
 
-    def wait_ready(self, name, timeout=10.0, rcode="NOERROR", qtype="A"):
+    def wait_ready(self, name, timeout=15.0, rcode="NOERROR", qtype="A"):
         """Wait until `name` resolves with the given rcode."""
         deadline = time.time() + timeout
         while time.time() < deadline:

@@ -52,7 +52,7 @@ def cluster(tmp_path):
          "-r", "100"],
         env=dict(os.environ, LOG_LEVEL="warn"),
         stdout=open(tmp_path / "bal.log", "ab"), stderr=subprocess.STDOUT)
-    deadline = time.time() + 10
+    deadline = time.time() + 20
     while time.time() < deadline:
         try:
             st = balstat(stats)

@@ -69,7 +69,7 @@ def test_queries_track_churn(tmp_path):
 
             # after churn settles, every mutated node must converge to
             # its final address
-            deadline = time.time() + 20
+            deadline = time.time() + 30
             pending = dict(mutations)
             while pending and time.time() < deadline:
                 for h, addr in list(pending.items()):

@@ -44,7 +44,7 @@ def test_data_change_propagates(stack):
     srv.wait_ready("web.bar.foo.com")
     jput(zk, "/com/foo/bar/web",
          {"type": "host", "host": {"address": "192.168.0.99"}})
-    deadline = time.time() + 5
+    deadline = time.time() + 12
     while time.time() < deadline:
         r = srv.dig("web.bar.foo.com")
         if r.answers and r.answers[0]["address"] == "192.168.0.99":
@@ -64,7 +64,7 @@ def test_new_and_removed_children_propagate(stack):
          {"type": "host", "host": {"address": "192.168.0.2"}})
     srv.wait_ready("api.bar.foo.com")
     zk.rmr("/com/foo/bar/api")
-    deadline = time.time() + 5
+    deadline = time.time() + 12
     while time.time() < deadline:
         if srv.dig("api.bar.foo.com").status == "REFUSED":
             break
