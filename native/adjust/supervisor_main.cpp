@@ -83,6 +83,7 @@ class Supervisor {
     std::string dir_;
     std::string binderd_;
     std::map<std::string, Instance> instances_;
+    bool shuttingDown_ = false;
 };
 
 bool Supervisor::start() {
@@ -108,8 +109,25 @@ bool Supervisor::start() {
                 converge();
             } else {
                 log_.info("shutting down; stopping all instances");
+                shuttingDown_ = true;  /* converge() must not respawn */
                 for (auto& [name, in] : instances_)
-                    if (in.pid > 0) kill(in.pid, SIGTERM);
+                    if (in.pid > 0) {
+                        in.stopping = true;
+                        kill(in.pid, SIGTERM);
+                    }
+                /* wait for children so our exit means theirs (ports
+                 * free for an immediate successor) */
+                int64_t deadline = monotonicMillis() + 5000;
+                bool anyLeft = true;
+                while (anyLeft && monotonicMillis() < deadline) {
+                    onChild();
+                    anyLeft = false;
+                    for (auto& [name, in] : instances_)
+                        anyLeft = anyLeft || in.pid > 0;
+                    if (anyLeft) usleep(50 * 1000);
+                }
+                for (auto& [name, in] : instances_)
+                    if (in.pid > 0) kill(in.pid, SIGKILL);
                 loop_->stop();
             }
         }
@@ -182,6 +200,7 @@ void Supervisor::scan() {
 }
 
 void Supervisor::converge() {
+    if (shuttingDown_) return;
     int64_t now = monotonicMillis();
     for (auto it = instances_.begin(); it != instances_.end();) {
         Instance& in = it->second;
@@ -241,7 +260,7 @@ void Supervisor::startInstance(Instance& in) {
         _exit(127);
     }
     in.pid = pid;
-    in.state = "online";
+    in.state = "starting";
     in.since = wallMillis();
     log_.info({{"instance", Json(in.name)},
                {"pid", Json((int64_t)pid)}},
@@ -293,9 +312,20 @@ void Supervisor::writeStatus() {
     JsonObject insts;
     std::string metricPorts;
     for (auto& [name, in] : instances_) {
+        /* online only once the instance is actually serving: binderd
+         * creates its balancer socket AFTER binding UDP+TCP, so file
+         * presence == ready (the SMF-online equivalent; smf_adjust's
+         * -w waits for real online, smf_adjust.c:457-544). Instances
+         * without a socket_path count as online once spawned. */
+        if (in.pid > 0 && in.state == "starting") {
+            const Json& sp = in.cfg.get("socket_path");
+            if (!sp.isString() ||
+                access(sp.asString().c_str(), F_OK) == 0)
+                in.state = "online";
+        }
         Json j = Json::object();
         j.set("pid", Json((int64_t)in.pid));
-        j.set("state", Json(in.pid > 0 ? "online" : in.state));
+        j.set("state", Json(in.pid > 0 ? in.state : "offline"));
         j.set("restarts", Json((int64_t)in.restarts));
         j.set("since", Json(in.since));
         if (in.cfg.get("port").isNumber())

@@ -63,7 +63,11 @@ static int bindSocket(const std::string& host, uint16_t port, int type,
                     type | SOCK_NONBLOCK | SOCK_CLOEXEC, 0);
     if (fd < 0) return -1;
     int one = 1;
-    setsockopt(fd, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+    /* SO_REUSEADDR only for TCP (TIME_WAIT rebinds). On UDP it would
+     * allow two processes to bind the same port with packets landing
+     * on the older socket — a silent split brain during restarts. */
+    if (type == SOCK_STREAM)
+        setsockopt(fd, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
     if (type == SOCK_DGRAM) {
         int sz = 4 << 20;
         setsockopt(fd, SOL_SOCKET, SO_RCVBUF, &sz, sizeof(sz));
