@@ -235,6 +235,13 @@ void StubStore::reindex(Node* n, const std::string& oldAddr) {
         rev_[n->rec_.address] = n;
 }
 
+void StubStore::clearParentSvcCache(const std::string& domain) {
+    size_t dot = domain.find('.');
+    if (dot == std::string::npos) return;
+    auto it = nodes_.find(domain.substr(dot + 1));
+    if (it != nodes_.end()) it->second->rec_.clearWireCaches();
+}
+
 void StubStore::put(const std::string& domain, const Json& data) {
     std::string d = domain;
     dnsLower(d);
@@ -247,6 +254,7 @@ void StubStore::put(const std::string& domain, const Json& data) {
     if (rec.hasData) {
         n->rec_ = std::move(rec);
         reindex(n, oldAddr);
+        clearParentSvcCache(d);
     }
 }
 
@@ -273,6 +281,7 @@ void StubStore::remove(const std::string& domain) {
         }
     }
     nodes_.erase(d);
+    clearParentSvcCache(d);
 }
 
 const StoreNode* StubStore::lookup(const std::string& domain) const {

@@ -70,11 +70,32 @@ struct CompiledRecord {
     /* member-only: per-member port list (server.js:383-385) */
     std::vector<uint16_t> ports;
 
-    /* binderd fast-path cache: prebuilt wire response for an
-     * all-lowercase A/IN query of this node's domain (built lazily in
-     * server.cpp). A fresh CompiledRecord starts empty, so every
-     * recompile (mirror update) invalidates it automatically. */
-    mutable std::vector<uint8_t> wireA;
+    /* binderd fast-path caches (built lazily in server.cpp; a fresh
+     * CompiledRecord starts empty, so recompiles invalidate them; the
+     * stores additionally clear the PARENT's service cache when a
+     * member changes). */
+    mutable std::vector<uint8_t> wireA;  /* host-like A response */
+    /* service responses, segment-permuted per query to preserve the
+     * reference's Fisher-Yates member shuffle: */
+    struct ServiceCache {
+        bool usable = false;     /* false => always slow path */
+        std::vector<uint8_t> headA;    /* header+question, A query */
+        std::vector<uint8_t> headSrv;  /* header+question, SRV query */
+        /* per member: plain-A answer segment / SRV answer segment(s) /
+         * additional-A segment */
+        struct Member {
+            std::vector<uint8_t> aSeg;
+            std::vector<uint8_t> srvSeg;
+            std::vector<uint8_t> addSeg;
+        };
+        std::vector<Member> members;
+        uint16_t srvAnCount = 0;  /* total SRV records over members */
+    };
+    mutable std::unique_ptr<ServiceCache> svc;
+    void clearWireCaches() const {
+        wireA.clear();
+        svc.reset();
+    }
 };
 
 /* Compile a znode JSON payload. `parsed`=false when payload was not valid
@@ -154,6 +175,7 @@ class StubStore : public Store {
     friend class Node;
     Node* ensure(const std::string& domain);
     void reindex(Node* n, const std::string& oldAddr);
+    void clearParentSvcCache(const std::string& domain);
 
     std::unordered_map<std::string, std::unique_ptr<Node>, SvHash, SvEq>
         nodes_;

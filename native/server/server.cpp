@@ -233,7 +233,118 @@ static inline int64_t nowUs() {
  * decoded message). Policy checks mirror Engine::resolve exactly for
  * the subset they cover; anything non-trivial bails to slow.
  */
-bool DnsServer::fastPath(const uint8_t* data, size_t len,
+/* append one big-endian u16 */
+static inline void putU16be(std::vector<uint8_t>& v, uint16_t x) {
+    v.push_back((uint8_t)(x >> 8));
+    v.push_back((uint8_t)x);
+}
+static inline void putU32be(std::vector<uint8_t>& v, uint32_t x) {
+    v.push_back((uint8_t)(x >> 24));
+    v.push_back((uint8_t)(x >> 16));
+    v.push_back((uint8_t)(x >> 8));
+    v.push_back((uint8_t)x);
+}
+static void putName(std::vector<uint8_t>& v, std::string_view name) {
+    size_t start = 0;
+    while (start <= name.size() && !name.empty()) {
+        size_t dot = name.find('.', start);
+        size_t end = dot == std::string_view::npos ? name.size() : dot;
+        v.push_back((uint8_t)(end - start));
+        v.insert(v.end(), name.begin() + start, name.begin() + end);
+        if (dot == std::string_view::npos) break;
+        start = dot + 1;
+    }
+    v.push_back(0);
+}
+
+/* Build the permutable service-response cache (see store.hpp). Returns
+ * a cache marked unusable when any member needs slow-path handling
+ * (invalid member record => SERVFAIL-partial semantics). */
+static void buildServiceCache(const StoreNode* node,
+                              std::string_view key) {
+    const CompiledRecord& rec = node->rec();
+    /* the only SRV qname this cache can serve is the registered one
+     * (mismatches fall back to the slow path before cache use) */
+    std::string srvQname =
+        rec.srvce + "." + rec.proto + "." + std::string(key);
+    auto cache = std::make_unique<CompiledRecord::ServiceCache>();
+    cache->usable = true;
+    uint32_t svcTtl = rec.ttl;
+
+    for (const StoreNode* kid : node->children()) {
+        const CompiledRecord& kr = kid->rec();
+        if (!kr.hasData || !recTypeServesUnderService(kr.type)) continue;
+        if (!kr.valid) {
+            cache->usable = false;  /* slow path owns SERVFAIL-partial */
+            break;
+        }
+        if (kr.address.empty()) continue;
+        CompiledRecord::ServiceCache::Member m;
+        uint32_t rttl = kr.memberTtlOverride.value_or(svcTtl);
+        uint32_t aTtl = rttl < svcTtl ? rttl : svcTtl;
+        std::string target = kid->name() + "." + std::string(key);
+        uint8_t a4[4];
+        if (inet_pton(AF_INET, kr.address.c_str(), a4) != 1) {
+            cache->usable = false;
+            break;
+        }
+        /* plain-A segment: name=ptr(12), A IN, min-ttl, addr */
+        m.aSeg.push_back(0xC0);
+        m.aSeg.push_back(0x0C);
+        putU16be(m.aSeg, 1);
+        putU16be(m.aSeg, 1);
+        putU32be(m.aSeg, aTtl);
+        putU16be(m.aSeg, 4);
+        m.aSeg.insert(m.aSeg.end(), a4, a4 + 4);
+        /* SRV segment(s): one per port */
+        std::vector<uint16_t> ports = kr.ports;
+        if (ports.empty())
+            ports.push_back(rec.hasDefaultPort ? rec.defaultPort : 0);
+        for (uint16_t p : ports) {
+            m.srvSeg.push_back(0xC0);
+            m.srvSeg.push_back(0x0C);
+            putU16be(m.srvSeg, 33);
+            putU16be(m.srvSeg, 1);
+            putU32be(m.srvSeg, svcTtl);
+            putU16be(m.srvSeg, (uint16_t)(6 + target.size() + 2));
+            putU16be(m.srvSeg, 0);   /* priority */
+            putU16be(m.srvSeg, 10);  /* weight (Record::SRV default) */
+            putU16be(m.srvSeg, p);
+            putName(m.srvSeg, target);
+            cache->srvAnCount++;
+        }
+        /* additional A: full (uncompressed) target name */
+        putName(m.addSeg, target);
+        putU16be(m.addSeg, 1);
+        putU16be(m.addSeg, 1);
+        putU32be(m.addSeg, rttl);
+        putU16be(m.addSeg, 4);
+        m.addSeg.insert(m.addSeg.end(), a4, a4 + 4);
+        cache->members.push_back(std::move(m));
+    }
+
+    auto buildHead = [&](std::vector<uint8_t>& head,
+                         std::string_view qname, uint16_t qtype,
+                         uint16_t an, uint16_t ar) {
+        head.reserve(16 + qname.size() + 6);
+        putU16be(head, 0);        /* id patched per query */
+        head.push_back(0x84);     /* QR|AA */
+        head.push_back(0x00);
+        putU16be(head, 1);
+        putU16be(head, an);
+        putU16be(head, 0);
+        putU16be(head, ar);
+        putName(head, qname);
+        putU16be(head, qtype);
+        putU16be(head, 1);
+    };
+    buildHead(cache->headA, key, 1, (uint16_t)cache->members.size(), 0);
+    buildHead(cache->headSrv, srvQname, 33, cache->srvAnCount,
+              (uint16_t)cache->members.size());
+    rec.svc = std::move(cache);
+}
+
+bool DnsServer::fastPath(const uint8_t* data, size_t len, bool udp,
                          std::vector<uint8_t>& out) {
     if (log_.enabled(LogLevel::Info)) return false;
     if (len < 17 || len > 300) return false;
@@ -264,11 +375,37 @@ bool DnsServer::fastPath(const uint8_t* data, size_t len,
         pos += l;
     }
     if (pos + 4 > len) return false;
-    if (data[pos] != 0 || data[pos + 1] != 1 ||  /* qtype A */
-        data[pos + 2] != 0 || data[pos + 3] != 1)  /* class IN */
+    bool isSrv;
+    if (data[pos] == 0 && data[pos + 1] == 1)
+        isSrv = false;
+    else if (data[pos] == 0 && data[pos + 1] == 33)
+        isSrv = true;
+    else
+        return false;
+    if (data[pos + 2] != 0 || data[pos + 3] != 1)  /* class IN */
         return false;
 
-    std::string_view key(name, nlen);
+    std::string_view qname(name, nlen);
+    std::string_view key = qname;
+    if (isSrv) {
+        /* strip _svc._proto (engine srvShape rules); mismatches fall
+         * back to the slow path */
+        size_t d1 = key.find('.');
+        if (d1 == std::string_view::npos || d1 == 0 || key[0] != '_')
+            return false;
+        size_t d2 = key.find('.', d1 + 1);
+        if (d2 == std::string_view::npos || key[d1 + 1] != '_')
+            return false;
+        std::string_view l1 = key.substr(0, d1);
+        std::string_view l2 = key.substr(d1 + 1, d2 - d1 - 1);
+        if (l1.find('_', 1) != std::string_view::npos ||
+            l2.find('_', 1) != std::string_view::npos)
+            return false;
+        srvSvc_ = l1;
+        srvProto_ = l2;
+        key = key.substr(d2 + 1);
+        if (key.empty()) return false;
+    }
     const EngineConfig& cfg = engine_->config();
     if (!cfg.dnsDomain.empty()) {
         /* must be a strict subdomain of dnsDomain, not doubled */
@@ -292,7 +429,69 @@ bool DnsServer::fastPath(const uint8_t* data, size_t len,
     const StoreNode* node = store->lookupView(key);
     if (node == nullptr) return false;
     const CompiledRecord& rec = node->rec();
-    if (!rec.valid || rec.address.empty()) return false;
+    if (!rec.valid) return false;
+
+    if (rec.type == RecType::Service) {
+        if (isSrv &&
+            (srvSvc_ != rec.srvce || srvProto_ != rec.proto))
+            return false;  /* slow path: NXDOMAIN */
+        if (!rec.svc) buildServiceCache(node, key);
+        const CompiledRecord::ServiceCache& c = *rec.svc;
+        if (!c.usable) return false;
+
+        /* size check against the UDP limit; oversize => slow path
+         * (which truncates with TC exactly as before) */
+        size_t total = (isSrv ? c.headSrv.size() : c.headA.size());
+        for (const auto& m : c.members)
+            total += isSrv ? m.srvSeg.size() + m.addSeg.size()
+                           : m.aSeg.size();
+        if (udp && total > 512) return false;
+
+        /* Fisher-Yates member order (server.js:40-53, 361) */
+        size_t n = c.members.size();
+        if (n > shuffleIdx_.size()) shuffleIdx_.resize(n);
+        for (size_t i = 0; i < n; ++i) shuffleIdx_[i] = (uint32_t)i;
+        for (size_t i = n; i > 1;) {
+            --i;
+            size_t j = rng_() % (i + 1);
+            std::swap(shuffleIdx_[i], shuffleIdx_[j]);
+        }
+
+        out.clear();
+        out.reserve(total);
+        const auto& head = isSrv ? c.headSrv : c.headA;
+        out.insert(out.end(), head.begin(), head.end());
+        if (isSrv) {
+            for (size_t i = 0; i < n; ++i) {
+                const auto& m = c.members[shuffleIdx_[i]];
+                out.insert(out.end(), m.srvSeg.begin(), m.srvSeg.end());
+            }
+            for (size_t i = 0; i < n; ++i) {
+                const auto& m = c.members[shuffleIdx_[i]];
+                out.insert(out.end(), m.addSeg.begin(), m.addSeg.end());
+            }
+        } else {
+            for (size_t i = 0; i < n; ++i) {
+                const auto& m = c.members[shuffleIdx_[i]];
+                out.insert(out.end(), m.aSeg.begin(), m.aSeg.end());
+            }
+        }
+        out[0] = data[0];
+        out[1] = data[1];
+        out[2] = (uint8_t)(out[2] | (data[2] & 0x01));
+
+        ++served_;
+        static const std::string kLabelSrv = "type=\"SRV\"";
+        static const std::string kLabelA2 = "type=\"A\"";
+        const std::string& lbl = isSrv ? kLabelSrv : kLabelA2;
+        reqCounter_->increment(lbl);
+        latHist_->observe(lbl, 1e-6);
+        sizeHist_->observe(lbl, (double)out.size());
+        return true;
+    }
+
+    if (isSrv) return false;  /* SRV on non-service: slow (NODATA+SOA) */
+    if (rec.address.empty()) return false;
     bool hostish = recTypeIsHostLike(rec.type) ||
                    rec.type == RecType::Database;
     if (!hostish) return false;
@@ -326,7 +525,7 @@ bool DnsServer::process(const uint8_t* data, size_t len, bool udp,
                         const ClientInfo& ci, std::vector<uint8_t>& out,
                         std::function<void(std::vector<uint8_t>)>
                             asyncReply) {
-    if (fastPath(data, len, out)) return true;
+    if (fastPath(data, len, udp, out)) return true;
     int64_t start = nowUs();
     BAMD_PROBE2(op_req_start, len, (int)udp);
     auto parsed = Message::decode(data, len);

@@ -19,6 +19,7 @@
 #include <sys/un.h>
 
 #include <functional>
+#include <random>
 #include <map>
 #include <memory>
 #include <string>
@@ -112,8 +113,11 @@ class DnsServer {
     bool process(const uint8_t* data, size_t len, bool udp,
                  const ClientInfo& ci, std::vector<uint8_t>& out,
                  std::function<void(std::vector<uint8_t>)> asyncReply);
-    bool fastPath(const uint8_t* data, size_t len,
+    bool fastPath(const uint8_t* data, size_t len, bool udp,
                   std::vector<uint8_t>& out);
+    std::string_view srvSvc_, srvProto_;   /* scratch, loop thread only */
+    std::vector<uint32_t> shuffleIdx_;
+    std::mt19937 rng_{0xb1d3};
 
   public:
     /* the store behind the engine (fast-path lookups); set by main */

@@ -174,6 +174,7 @@ void ZkMirror::onChildren(const std::string& path,
     /* whatever remains in n->kids_ was removed */
     for (auto& [label, old] : n->kids_) unbind(old);
     n->kids_ = std::move(newKids);
+    n->rec_.clearWireCaches();  /* membership feeds service responses */
 }
 
 void ZkMirror::unbind(Node* n) {
@@ -217,6 +218,12 @@ void ZkMirror::onData(const std::string& path, const std::string& data) {
         n->ip_ = newIp;
     }
     n->rec_ = std::move(rec);
+    /* member data feeds the parent's cached service responses */
+    size_t slash = path.rfind('/');
+    if (slash != std::string::npos && slash > 0) {
+        Node* parent = nodeAt(path.substr(0, slash));
+        if (parent != nullptr) parent->rec_.clearWireCaches();
+    }
 }
 
 }  // namespace bamd

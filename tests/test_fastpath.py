@@ -106,3 +106,98 @@ def test_cache_invalidation_on_update(tmp_path):
             srv.stop()
     finally:
         zk.stop()
+
+
+def test_service_fastpath_parity(twins):
+    fast, slow = twins
+    for qtype, name in [("A", "svc.foo.com"),
+                        ("SRV", "_x._tcp.svc.foo.com")]:
+        rf = fast.dig(name, qtype, qid=555)
+        rs = slow.dig(name, qtype, qid=555)
+        assert rf.status == rs.status == "NOERROR"
+        def norm(r):
+            return sorted(
+                (a["type"], a.get("address"), a.get("target"),
+                 a.get("port"), a.get("priority"), a.get("weight"),
+                 a["ttl"], a["name"]) for a in r.answers)
+        assert norm(rf) == norm(rs), (qtype, rf, rs)
+        addf = sorted((x["name"], x.get("address"), x["ttl"])
+                      for x in rf["additionals"])
+        adds = sorted((x["name"], x.get("address"), x["ttl"])
+                      for x in rs["additionals"])
+        assert addf == adds
+
+
+def test_service_fastpath_shuffles(tmp_path):
+    import json as _json
+    tree = {"foo.com": None,
+            "s.foo.com": {"type": "service",
+                          "service": {"srvce": "_x", "proto": "_tcp",
+                                      "port": 1}}}
+    for i in range(6):
+        tree[f"m{i}.s.foo.com"] = {
+            "type": "rr_host", "rr_host": {"address": f"10.4.0.{i}"}}
+    store = tmp_path / "t.json"
+    store.write_text(_json.dumps(tree))
+    srv = BinderProcess(store=f"file:{store}", workdir=tmp_path,
+                        log_level="warn")
+    srv.start()
+    try:
+        orders = set()
+        for _ in range(25):
+            r = srv.dig("s.foo.com")
+            orders.add(tuple(a["address"] for a in r.answers))
+        assert len(orders) > 1, "fast path must keep shuffling"
+    finally:
+        srv.stop()
+
+
+def test_service_fastpath_invalidation_on_member_change(tmp_path):
+    zk = StubZk().start()
+    try:
+        zk.mkdirp("/com/foo")
+        zk.put("/com/foo/s", json.dumps(
+            {"type": "service",
+             "service": {"srvce": "_x", "proto": "_tcp",
+                         "port": 9}}).encode())
+        zk.put("/com/foo/s/m0", json.dumps(
+            {"type": "rr_host",
+             "rr_host": {"address": "10.5.0.1"}}).encode())
+        srv = BinderProcess(store="zk", zk_host="127.0.0.1",
+                            zk_port=zk.port, workdir=tmp_path,
+                            log_level="warn")
+        srv.start()
+        try:
+            srv.wait_ready("s.foo.com")
+            for _ in range(5):  # prime cache
+                r = srv.dig("_x._tcp.s.foo.com", "SRV")
+            assert r["additionals"][0]["address"] == "10.5.0.1"
+            # member address change must invalidate the parent cache
+            zk.put("/com/foo/s/m0", json.dumps(
+                {"type": "rr_host",
+                 "rr_host": {"address": "10.5.0.2"}}).encode())
+            deadline = time.time() + 10
+            while time.time() < deadline:
+                r = srv.dig("_x._tcp.s.foo.com", "SRV")
+                if r["additionals"] and \
+                        r["additionals"][0]["address"] == "10.5.0.2":
+                    break
+                time.sleep(0.05)
+            else:
+                pytest.fail("stale cached service answer")
+            # membership change too
+            zk.put("/com/foo/s/m1", json.dumps(
+                {"type": "rr_host",
+                 "rr_host": {"address": "10.5.0.3"}}).encode())
+            deadline = time.time() + 10
+            while time.time() < deadline:
+                r = srv.dig("_x._tcp.s.foo.com", "SRV")
+                if len(r.answers) == 2:
+                    break
+                time.sleep(0.05)
+            else:
+                pytest.fail("new member never appeared in cached answer")
+        finally:
+            srv.stop()
+    finally:
+        zk.stop()
