@@ -152,3 +152,59 @@ def test_original_source_preserved(cluster):
         (cluster["tmp"] / f"b{i}.log").read_text()
         for i in range(3) if (cluster["tmp"] / f"b{i}.log").exists())
     assert '"client":"127.0.0.77"' in logs
+
+
+def test_multiworker_balancer(tmp_path):
+    """SO_REUSEPORT worker mode (-w 2): queries from several source IPs
+    through both workers; aggregated stats cover all workers."""
+    sockdir = tmp_path / "socks"
+    sockdir.mkdir()
+    store = tmp_path / "tree.json"
+    store.write_text(json.dumps(TREE))
+    backends = [BinderProcess(store=f"file:{store}", workdir=tmp_path,
+                              balancer_socket=str(sockdir / f"b{i}"))
+                .start() for i in range(2)]
+    port = free_port()
+    stats = tmp_path / "stats.sock"
+    bal = subprocess.Popen(
+        [str(REPO_ROOT / "bin" / "binder-balancer"), "-p", str(port),
+         "-H", "127.0.0.1", "-s", str(sockdir), "-S", str(stats),
+         "-r", "100", "-w", "2"],
+        env=dict(os.environ, LOG_LEVEL="warn"),
+        stdout=subprocess.DEVNULL, stderr=subprocess.STDOUT)
+    try:
+        deadline = time.time() + 20
+        while time.time() < deadline:
+            try:
+                st = balstat(stats)
+                if st.get("workers") == 2 and \
+                        sum(1 for b in st["backends"] if b["ok"]) >= 2:
+                    break
+            except (OSError, ValueError):
+                pass
+            time.sleep(0.1)
+        else:
+            pytest.fail("multiworker balancer never ready")
+        from binder_amd import require_native
+        n = require_native()
+        for i in range(2, 12):
+            with socket.socket(socket.AF_INET,
+                               socket.SOCK_DGRAM) as s:
+                s.bind((f"127.0.0.{i}", 0))
+                s.settimeout(3)
+                wire = n.encode_message(
+                    {"id": i, "questions": [{"name": "web.foo.com",
+                                             "type": "A"}]})
+                s.sendto(wire, ("127.0.0.1", port))
+                data, _ = s.recvfrom(4096)
+                m = n.decode_message(data)
+                assert m["rcode"] == "NOERROR"
+                assert m["id"] == i
+        st = balstat(stats)
+        assert st["udp_queries"] >= 10
+        assert sum(b["queries"] for b in st["backends"]) >= 10
+    finally:
+        bal.terminate()
+        bal.wait(timeout=5)
+        for b in backends:
+            b.stop()
