@@ -200,7 +200,14 @@ def test_multiworker_balancer(tmp_path):
                 m = n.decode_message(data)
                 assert m["rcode"] == "NOERROR"
                 assert m["id"] == i
-        st = balstat(stats)
+        # other workers publish their snapshots on their next sweep
+        # tick; retry briefly
+        deadline = time.time() + 5
+        while time.time() < deadline:
+            st = balstat(stats)
+            if st["udp_queries"] >= 10:
+                break
+            time.sleep(0.1)
         assert st["udp_queries"] >= 10
         assert sum(b["queries"] for b in st["backends"]) >= 10
     finally:
