@@ -258,7 +258,8 @@ class StubZk:
         return struct.pack(">i", len(b)) + b
 
     def _do_create(self, path: str, data: bytes,
-                   ephemeral_owner: int = 0) -> int:
+                   ephemeral_owner: int = 0,
+                   txn_session: int = 0) -> int:
         if path in self._nodes:
             return ZNODEEXISTS
         parent = _parent(path)
@@ -272,7 +273,8 @@ class StubZk:
         self._txn(1, self._jstr(path) + self._jstr(data) +
                   struct.pack(">i", 1) + struct.pack(">i", 31) +
                   self._jstr("world") + self._jstr("anyone") +
-                  eph + struct.pack(">i", pn.cversion + 1))
+                  eph + struct.pack(">i", pn.cversion + 1),
+                  client_id=txn_session or ephemeral_owner)
         self._nodes[path] = _Node(data, ephemeral_owner)
         pn.children.add(path[path.rfind("/") + 1:])
         pn.cversion += 1
@@ -280,7 +282,8 @@ class StubZk:
         self._fire(parent, EV_CHILDREN, child=True)
         return ZOK
 
-    def _do_set(self, path: str, data: bytes) -> int:
+    def _do_set(self, path: str, data: bytes,
+                txn_session: int = 0) -> int:
         n = self._nodes.get(path)
         if n is None:
             return ZNONODE
@@ -288,18 +291,18 @@ class StubZk:
         n.data = data
         n.version += 1
         self._txn(5, self._jstr(path) + self._jstr(data) +
-                  struct.pack(">i", n.version))
+                  struct.pack(">i", n.version), client_id=txn_session)
         self._fire(path, EV_DATA)
         return ZOK
 
-    def _do_delete(self, path: str) -> int:
+    def _do_delete(self, path: str, txn_session: int = 0) -> int:
         n = self._nodes.get(path)
         if n is None:
             return ZNONODE
         if n.children:
             return ZNOTEMPTY
         self._zxid += 1
-        self._txn(2, self._jstr(path))
+        self._txn(2, self._jstr(path), client_id=txn_session)
         del self._nodes[path]
         parent = _parent(path)
         pn = self._nodes.get(parent)
@@ -549,7 +552,8 @@ class StubZk:
                 r.s()
             flags = r.i32()
             owner = c.session_id if (flags & 1) else 0  # EPHEMERAL
-            rc = self._do_create(path, data, owner)
+            rc = self._do_create(path, data, owner,
+                                 txn_session=c.session_id)
             self._reply(c, xid, rc, _s(path) if rc == ZOK else b"")
             return
 
@@ -557,7 +561,7 @@ class StubZk:
             path = r.s()
             data = r.buf() or b""
             r.i32()  # version (-1 = any; stub ignores)
-            rc = self._do_set(path, data)
+            rc = self._do_set(path, data, txn_session=c.session_id)
             n = self._nodes.get(path)
             self._reply(c, xid, rc,
                         self._stat(path, n) if rc == ZOK else b"")
@@ -566,7 +570,7 @@ class StubZk:
         if op == OP_DELETE:
             path = r.s()
             r.i32()
-            rc = self._do_delete(path)
+            rc = self._do_delete(path, txn_session=c.session_id)
             self._reply(c, xid, rc)
             return
 

@@ -74,3 +74,53 @@ def test_session_txns_from_wire_client(tmp_path):
     assert created[0]["timeout_ms"] == 30000
     sessions = [t for t in txns if "open" in t]
     assert len(sessions) >= 1
+
+
+def test_filters(tmp_path):
+    """-s (session), -z (server id), -t (window) filters."""
+    import socket
+    import struct
+    import time
+    logdir = tmp_path / "txnlog"
+    zk = StubZk(txnlog_dir=str(logdir)).start()
+    try:
+        # two wire sessions, each doing one create
+        sids = []
+        for i in range(2):
+            s = socket.socket()
+            s.connect(("127.0.0.1", zk.port))
+            req = struct.pack(">iqiq", 0, 0, 30000, 0) + \
+                struct.pack(">i", 16) + b"\x00" * 16 + b"\x00"
+            s.sendall(struct.pack(">i", len(req)) + req)
+            resp = s.recv(4096)
+            (_, _, sid) = struct.unpack_from(">iiq", resp[4:], 0)
+            sids.append(sid)
+            # create /w<i>
+            path = f"/w{i}".encode()
+            body = struct.pack(">ii", 1, 1) + \
+                struct.pack(">i", len(path)) + path + \
+                struct.pack(">i", 4) + b"null" + \
+                struct.pack(">i", 1) + struct.pack(">i", 31) + \
+                struct.pack(">i", 5) + b"world" + \
+                struct.pack(">i", 6) + b"anyone" + struct.pack(">i", 0)
+            s.sendall(struct.pack(">i", len(body)) + body)
+            s.recv(4096)
+            s.close()
+        time.sleep(0.3)
+    finally:
+        zk.stop()
+
+    log = str(logdir / "log.1")
+    # session filter: only txns of session 0
+    txns = run_zklogcat(["-s", format(sids[0], "x"), log])
+    assert all(t["session"] == format(sids[0], "x") for t in txns)
+    assert any(t.get("path") == "/w0" for t in txns)
+    assert not any(t.get("path") == "/w1" for t in txns)
+    # server-id filter: top byte of our stub sessions is 0 -> all
+    txns = run_zklogcat(["-z", "0", log])
+    assert any(t.get("path") == "/w1" for t in txns)
+    txns = run_zklogcat(["-z", "5", log])
+    assert txns == []
+    # window filter: everything is recent -> all kept
+    txns = run_zklogcat(["-t", "3600", log])
+    assert any(t.get("path") == "/w0" for t in txns)
