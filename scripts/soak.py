@@ -5,7 +5,7 @@ Catches slow leaks (pending-map growth, watch bookkeeping, buffer
 bloat) that short benches miss. Exits nonzero if RSS keeps climbing
 after warmup or if service degrades.
 
-usage: soak.py [seconds] (default 60)
+usage: soak.py [seconds] [n_backends] (default 60 2)
 """
 import json
 import subprocess
@@ -35,6 +35,7 @@ def rss_mb(pid):
 
 def main():
     seconds = int(sys.argv[1]) if len(sys.argv) > 1 else 60
+    n_backends = int(sys.argv[2]) if len(sys.argv) > 2 else 2
     tmp = Path(tempfile.mkdtemp(prefix="soak-"))
     zk = StubZk().start()
     zk.mkdirp("/com/foo")
@@ -50,7 +51,7 @@ def main():
     sockdir = tmp / "socks"
     sockdir.mkdir()
     backends = []
-    for i in range(2):
+    for i in range(n_backends):
         b = BinderProcess(store="zk", zk_host="127.0.0.1",
                           zk_port=zk.port, workdir=tmp,
                           log_level="warn",
@@ -63,7 +64,8 @@ def main():
     bport = free_port()
     bal = subprocess.Popen(
         [str(REPO / "bin" / "binder-balancer"), "-p", str(bport),
-         "-H", "127.0.0.1", "-s", str(sockdir), "-r", "200", "-w", "2"],
+         "-H", "127.0.0.1", "-s", str(sockdir), "-r", "200",
+         "-w", str(min(8, n_backends))],
         stdout=subprocess.DEVNULL, stderr=subprocess.STDOUT)
     time.sleep(1.0)
 
@@ -99,7 +101,8 @@ def main():
     blast = subprocess.Popen(
         [str(REPO / "bin" / "dnsblast"), "-s", "127.0.0.1",
          "-p", str(bport), "-n", str(10_000_000_000), "-c", "64",
-         "-t", "4", "-f", str(tmp / "names.txt"), "-B", "127.0.1.1"],
+         "-t", str(2 * n_backends), "-f", str(tmp / "names.txt"),
+         "-B", "127.0.1.1"],
         stdout=subprocess.PIPE, stderr=subprocess.STDOUT)
 
     samples = []
