@@ -195,7 +195,12 @@ bool Client::connect() {
             close();
             return false;
         }
-        SSL_CTX_set_verify(ctx, SSL_VERIFY_NONE, nullptr);
+        if (opts_.tlsVerify) {
+            SSL_CTX_set_default_verify_paths(ctx);
+            SSL_CTX_set_verify(ctx, SSL_VERIFY_PEER, nullptr);
+        } else {
+            SSL_CTX_set_verify(ctx, SSL_VERIFY_NONE, nullptr);
+        }
         sslCtx_ = ctx;
         SSL* ssl = SSL_new(ctx);
         SSL_set_fd(ssl, fd_);

@@ -25,6 +25,7 @@ struct Options {
     std::string host;
     uint16_t port = 389;
     bool tls = false;           // ldaps://
+    bool tlsVerify = false;     // verify peer cert (system CA paths)
     std::string bindDn;         // empty = anonymous
     std::string bindPassword;
     int timeoutMs = 10000;

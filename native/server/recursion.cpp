@@ -209,6 +209,7 @@ void Recursion::refreshViaUfds() {
     lopts.host = addr;
     lopts.port = port;
     lopts.tls = tls;
+    lopts.tlsVerify = ucfg.get("tlsVerify").asBool(false);
     lopts.bindDn = ucfg.get("bindDN").asString();
     lopts.bindPassword = ucfg.get("bindPassword").asString();
     std::string region = opts_.regionName;
