@@ -792,7 +792,10 @@ Json Balancer::snapshot() const {
 
 void Balancer::onStatsAccept() {
     while (true) {
-        int fd = accept4(statsFd_, nullptr, nullptr, SOCK_CLOEXEC);
+        /* non-blocking best-effort: a stalled stats reader must not
+         * stall the serving loop */
+        int fd = accept4(statsFd_, nullptr, nullptr,
+                         SOCK_CLOEXEC | SOCK_NONBLOCK);
         if (fd < 0) return;
         /* merge the latest snapshot from every worker (our own fresh;
          * others as of their last sweep tick) */
