@@ -177,6 +177,7 @@ class Balancer {
     int nextBackendId_ = 1;
     uint64_t udpQueries_ = 0, udpReplies_ = 0, drops_ = 0;
 
+    static constexpr size_t kMaxRemotes = 262144;
     static constexpr int64_t kReplyTtlMs = 3000;
     static constexpr int64_t kPingIntervalMs = 2000;
     static constexpr int64_t kPingTimeoutMs = 6000;
@@ -548,6 +549,19 @@ static uint64_t addrKey(const struct sockaddr_storage& ss) {
 }
 
 Backend* Balancer::pickBackendFast(const struct sockaddr_storage& ss) {
+    /* bound the affinity tables: a flood of spoofed source addresses
+     * must not grow memory forever. Wiping re-pins live remotes on
+     * their next packet (brief affinity reset, logged). */
+    if (remotesFast_.size() > kMaxRemotes) {
+        log_.warn({{"remotes", Json((int64_t)remotesFast_.size())}},
+                  "remote table overflow; resetting affinity pins");
+        remotesFast_.clear();
+        remotes_.clear();
+        for (auto& [id, be] : backendsById_) {
+            g_pins.add(be->path, -(int)be->remotes);
+            be->remotes = 0;
+        }
+    }
     uint64_t key = addrKey(ss);
     auto it = remotesFast_.find(key);
     if (it != remotesFast_.end()) {

@@ -23,6 +23,9 @@ TREE = {
                                 "port": 1, "ttl": 60}},
     "m0.svc.foo.com": {"type": "rr_host",
                        "rr_host": {"address": "10.0.0.9"}},
+    # member-level TTL override exercises the min()/override chains
+    "m1.svc.foo.com": {"type": "rr_host", "ttl": 7,
+                       "rr_host": {"address": "10.0.0.10"}},
 }
 
 
