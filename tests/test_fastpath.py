@@ -172,6 +172,14 @@ def test_service_fastpath_invalidation_on_member_change(tmp_path):
         srv.start()
         try:
             srv.wait_ready("s.foo.com")
+            # wait for the member itself to be mirrored (a service
+            # answers NOERROR even with zero members), then prime
+            deadline = time.time() + 10
+            while time.time() < deadline:
+                r = srv.dig("_x._tcp.s.foo.com", "SRV")
+                if r["additionals"]:
+                    break
+                time.sleep(0.05)
             for _ in range(5):  # prime cache
                 r = srv.dig("_x._tcp.s.foo.com", "SRV")
             assert r["additionals"][0]["address"] == "10.5.0.1"
