@@ -21,6 +21,7 @@
  * (deep equality via the canonical dump — the nvlist_equal analog,
  * src/nvlist_equal.c). Exit 0 on success.
  */
+#include <signal.h>
 #include <dirent.h>
 #include <sys/stat.h>
 #include <unistd.h>
@@ -47,6 +48,9 @@ static Json readJson(const std::string& path) {
 }
 
 int main(int argc, char** argv) {
+    /* a peer closing mid-write must be an EPIPE errno, not process
+     * death */
+    signal(SIGPIPE, SIG_IGN);
     const char* lvl = getenv("LOG_LEVEL");
     Logger log("binder-adjust",
                logLevelFromName(lvl ? lvl : "info", LogLevel::Info));

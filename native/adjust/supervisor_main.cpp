@@ -357,6 +357,9 @@ void Supervisor::writeStatus() {
 }  // namespace
 
 int main(int argc, char** argv) {
+    /* a peer closing mid-write must be an EPIPE errno, not process
+     * death */
+    signal(SIGPIPE, SIG_IGN);
     const char* lvl = getenv("LOG_LEVEL");
     Logger log("binder-supervisor",
                logLevelFromName(lvl ? lvl : "info", LogLevel::Info));

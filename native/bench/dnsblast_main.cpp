@@ -14,6 +14,7 @@
  *   -B 127.0.0.x base: thread i binds source ip base+i (gives the
  *      balancer distinct remotes so per-IP affinity spreads load).
  */
+#include <signal.h>
 #include <arpa/inet.h>
 #include <netinet/in.h>
 #include <poll.h>
@@ -252,6 +253,9 @@ void worker(const Config& cfg, int tid,
 }  // namespace
 
 int main(int argc, char** argv) {
+    /* a peer closing mid-write must be an EPIPE errno, not process
+     * death */
+    signal(SIGPIPE, SIG_IGN);
     Config cfg;
     int c;
     while ((c = getopt(argc, argv, "hs:p:n:c:t:f:B:T:R")) != -1) {

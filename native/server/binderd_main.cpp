@@ -66,6 +66,9 @@ static bool loadFileStore(StubStore& store, const std::string& path,
 }
 
 int main(int argc, char** argv) {
+    /* a peer closing mid-write must be an EPIPE errno, not process
+     * death */
+    signal(SIGPIPE, SIG_IGN);
     const char* lvl = getenv("LOG_LEVEL");
     LogLevel level = logLevelFromName(lvl ? lvl : "info", LogLevel::Info);
     Logger log("binder", level);
