@@ -215,3 +215,49 @@ def test_multiworker_balancer(tmp_path):
         bal.wait(timeout=5)
         for b in backends:
             b.stop()
+
+
+def test_bsock_frame_dribble(cluster, tmp_path):
+    """A bsock1 QUERY frame delivered one byte at a time must still be
+    served (stream reassembly on the backend side)."""
+    from binder_amd import require_native
+    n = require_native()
+    sock_path = None
+    for i in range(3):
+        p = cluster["sockdir"] / f"b{i}"
+        if p.exists():
+            sock_path = p
+            break
+    assert sock_path is not None
+    wire = n.encode_message(
+        {"id": 99, "questions": [{"name": "web.foo.com", "type": "A"}]})
+    payload = (99).to_bytes(4, "little") + bytes([4, 0]) + \
+        (5353).to_bytes(2, "little") + \
+        bytes([127, 0, 0, 55]) + b"\x00" * 12 + wire
+    frame = bytes([0xB5, 1]) + len(payload).to_bytes(4, "little") + \
+        payload
+    with socket.socket(socket.AF_UNIX) as s:
+        s.settimeout(5)
+        s.connect(str(sock_path))
+        for b in frame:
+            s.sendall(bytes([b]))
+            time.sleep(0.001)
+        hdr = b""
+        while len(hdr) < 6:
+            hdr += s.recv(6 - len(hdr))
+        assert hdr[0] == 0xB5 and hdr[1] == 2  # REPLY
+        plen = int.from_bytes(hdr[2:6], "little")
+        body = b""
+        while len(body) < plen:
+            body += s.recv(plen - len(body))
+        req_id = int.from_bytes(body[:4], "little")
+        assert req_id == 99
+        resp = n.decode_message(body[4:])
+        assert resp["rcode"] == "NOERROR"
+        assert resp["id"] == 99
+
+
+def test_edns_through_balancer(cluster):
+    r = dig("web.foo.com", port=cluster["port"], edns=4096)
+    assert r.status == "NOERROR"
+    assert any(x["type"] == "OPT" for x in r["additionals"])
