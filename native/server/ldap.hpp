@@ -48,6 +48,8 @@ class Client {
                 const std::string& value, std::vector<Entry>& out);
 
     void close();
+    bool isConnected() const { return fd_ >= 0; }
+    const Options& options() const { return opts_; }
     const std::string& error() const { return err_; }
 
   private:

@@ -219,16 +219,37 @@ void Recursion::refreshViaUfds() {
     EventLoop* loop = loop_;
     Recursion* self = this;
     ldapThread_ = std::thread([self, loop, lopts, region]() {
-        ldap::Client client(lopts);
-        bool ok = client.connect();
+        /* reuse the held connection when the target is unchanged */
+        ldap::Client* client = self->ldapClient_.get();
+        bool fresh = false;
+        if (client == nullptr || !client->isConnected() ||
+            client->options().host != lopts.host ||
+            client->options().port != lopts.port) {
+            self->ldapClient_ = std::make_unique<ldap::Client>(lopts);
+            client = self->ldapClient_.get();
+            fresh = true;
+        }
         std::vector<ldap::Entry> entries;
         std::string errMsg;
-        if (ok) {
-            ok = client.search("region=" + region + ", o=smartdc",
-                               "objectclass", "resolver", entries);
+        std::string base = "region=" + region + ", o=smartdc";
+        bool ok = true;
+        if (fresh) ok = client->connect();
+        if (ok)
+            ok = client->search(base, "objectclass", "resolver",
+                                entries);
+        if (!ok && !fresh) {
+            /* held connection went stale: reconnect once */
+            entries.clear();
+            self->ldapClient_ = std::make_unique<ldap::Client>(lopts);
+            client = self->ldapClient_.get();
+            ok = client->connect() &&
+                 client->search(base, "objectclass", "resolver",
+                                entries);
         }
-        if (!ok) errMsg = client.error();
-        client.close();
+        if (!ok) {
+            errMsg = client->error();
+            self->ldapClient_.reset();  /* reconnect next refresh */
+        }
         loop->postFromThread([self, ok, errMsg,
                               entries = std::move(entries)]() {
             self->ldapBusy_ = false;

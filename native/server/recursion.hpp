@@ -36,6 +36,7 @@
 #include "../common/log.hpp"
 #include "../common/loop.hpp"
 #include "../engine/store.hpp"
+#include "ldap.hpp"
 #include "server.hpp"
 
 namespace bamd {
@@ -113,9 +114,13 @@ class Recursion : public RecursionIface {
     int64_t nicCacheAtMs_ = 0;
 
     /* UFDS/LDAP refresh runs on a helper thread (blocking client);
-     * results are posted back via EventLoop::postFromThread. */
+     * results are posted back via EventLoop::postFromThread. The
+     * connection persists across 5-minute refreshes like the
+     * reference's long-lived UFDS client (recursion.js:129-148);
+     * only the helper thread touches ldapClient_ (ldapBusy_ gates). */
     std::thread ldapThread_;
     bool ldapBusy_ = false;
+    std::unique_ptr<ldap::Client> ldapClient_;
     void refreshViaUfds();
 };
 
