@@ -83,9 +83,28 @@ check: all
 	$(PYTHON) -m py_compile binder_amd/*.py bench.py __graft_entry__.py
 	@echo "check OK"
 
+# release tarball layout under /opt/binder-amd (the reference ships
+# /opt/smartdc/binder with balancer+smf_adjust in lib/ and zklog in
+# bin/; Makefile:178-229 there)
+release: all
+	rm -rf dist/binder-amd
+	mkdir -p dist/binder-amd/bin dist/binder-amd/etc
+	cp bin/binderd bin/binder-balancer bin/binder-adjust \
+	    bin/binder-supervisor bin/zklogcat bin/dnsblast \
+	    dist/binder-amd/bin/
+	cp -r deploy dist/binder-amd/
+	cp -r tools dist/binder-amd/
+	cp etc/config.json.in etc/config.sample.json dist/binder-amd/etc/
+	mkdir -p dist/binder-amd/lib/python/binder_amd
+	cp binder_amd/*.py $(PYMOD) dist/binder-amd/lib/python/binder_amd/
+	cp README.md LICENSE dist/binder-amd/
+	cp -r docs dist/binder-amd/
+	tar -C dist -czf dist/binder-amd.tar.gz binder-amd
+	@echo "release: dist/binder-amd.tar.gz"
+
 clean:
-	rm -rf $(BUILD) $(PYMOD) $(BINARIES)
+	rm -rf $(BUILD) $(PYMOD) $(BINARIES) dist
 
 -include $(shell find $(BUILD) -name '*.d' 2>/dev/null)
 
-.PHONY: all clean test check
+.PHONY: all clean test check release
