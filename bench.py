@@ -70,7 +70,7 @@ def build_tree(zk, names_path, records=10000):
 
 
 def start_backends(n, tmp, zk_port, last_name="h4999.foo.com"):
-    from binder_amd.harness import BinderProcess
+    from binder_amd.harness import BinderProcess, BALANCERD
     sockdir = tmp / "socks"
     sockdir.mkdir(exist_ok=True)
     backends = []
@@ -91,7 +91,7 @@ def start_backends(n, tmp, zk_port, last_name="h4999.foo.com"):
 def start_balancer(tmp, sockdir, port, workers=1):
     env = dict(os.environ, LOG_LEVEL="warn")
     proc = subprocess.Popen(
-        [str(REPO / "bin" / "binder-balancer"), "-p", str(port),
+        [str(BALANCERD), "-p", str(port),
          "-H", "127.0.0.1", "-s", str(sockdir),
          "-S", str(tmp / "stats.sock"), "-r", "200",
          "-w", str(workers)],

@@ -23,7 +23,8 @@ from .digclient import dig
 # BINDERD_BIN overrides the server binary (e.g. an ASan build)
 BINDERD = Path(os.environ.get("BINDERD_BIN",
                               REPO_ROOT / "bin" / "binderd"))
-BALANCERD = REPO_ROOT / "bin" / "binder-balancer"
+BALANCERD = Path(os.environ.get("BALANCER_BIN",
+                                REPO_ROOT / "bin" / "binder-balancer"))
 
 
 def free_port() -> int:

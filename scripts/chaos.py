@@ -26,7 +26,7 @@ REPO = Path(__file__).resolve().parent.parent
 sys.path.insert(0, str(REPO))
 
 from binder_amd.digclient import dig  # noqa: E402
-from binder_amd.harness import free_port  # noqa: E402
+from binder_amd.harness import BALANCERD, free_port  # noqa: E402
 from binder_amd.stubzk import StubZk  # noqa: E402
 
 BIN = REPO / "bin"
@@ -63,7 +63,7 @@ def main():
         capture_output=True).returncode == 0
     bport = free_port()
     bal = subprocess.Popen(
-        [str(BIN / "binder-balancer"), "-p", str(bport),
+        [str(BALANCERD), "-p", str(bport),
          "-H", "127.0.0.1", "-s", str(statedir / "sockets"),
          "-r", "100"], env=env,
         stdout=open(tmp / "bal.log", "ab"), stderr=subprocess.STDOUT)

@@ -19,7 +19,8 @@ from pathlib import Path
 REPO = Path(__file__).resolve().parent.parent
 sys.path.insert(0, str(REPO))
 
-from binder_amd.harness import BinderProcess, free_port  # noqa: E402
+from binder_amd.harness import (  # noqa: E402
+    BALANCERD, BinderProcess, free_port)
 
 
 def main():
@@ -68,7 +69,7 @@ def main():
             bal.wait(timeout=5)
         bal_port = free_port()
         bal = subprocess.Popen(
-            [str(REPO / "bin" / "binder-balancer"), "-p", str(bal_port),
+            [str(BALANCERD), "-p", str(bal_port),
              "-H", "127.0.0.1", "-s", str(sockdir),
              "-S", str(tmp / "stats.sock"), "-r", "200",
              "-w", str(w)],

@@ -18,7 +18,8 @@ from pathlib import Path
 REPO = Path(__file__).resolve().parent.parent
 sys.path.insert(0, str(REPO))
 
-from binder_amd.harness import BinderProcess, free_port  # noqa: E402
+from binder_amd.harness import (  # noqa: E402
+    BALANCERD, BinderProcess, free_port)
 from binder_amd.stubzk import StubZk  # noqa: E402
 
 
@@ -63,7 +64,7 @@ def main():
 
     bport = free_port()
     bal = subprocess.Popen(
-        [str(REPO / "bin" / "binder-balancer"), "-p", str(bport),
+        [str(BALANCERD), "-p", str(bport),
          "-H", "127.0.0.1", "-s", str(sockdir), "-r", "200",
          "-w", str(min(8, n_backends))],
         stdout=subprocess.DEVNULL, stderr=subprocess.STDOUT)

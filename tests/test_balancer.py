@@ -16,7 +16,7 @@ import pytest
 
 from binder_amd import REPO_ROOT
 from binder_amd.digclient import dig
-from binder_amd.harness import BinderProcess, free_port
+from binder_amd.harness import BinderProcess, free_port, BALANCERD
 
 TREE = {
     "foo.com": None,
@@ -47,7 +47,7 @@ def cluster(tmp_path):
     port = free_port()
     stats = tmp_path / "stats.sock"
     bal = subprocess.Popen(
-        [str(REPO_ROOT / "bin" / "binder-balancer"), "-p", str(port),
+        [str(BALANCERD), "-p", str(port),
          "-H", "127.0.0.1", "-s", str(sockdir), "-S", str(stats),
          "-r", "100"],
         env=dict(os.environ, LOG_LEVEL="warn"),
@@ -167,7 +167,7 @@ def test_multiworker_balancer(tmp_path):
     port = free_port()
     stats = tmp_path / "stats.sock"
     bal = subprocess.Popen(
-        [str(REPO_ROOT / "bin" / "binder-balancer"), "-p", str(port),
+        [str(BALANCERD), "-p", str(port),
          "-H", "127.0.0.1", "-s", str(sockdir), "-S", str(stats),
          "-r", "100", "-w", "2"],
         env=dict(os.environ, LOG_LEVEL="warn"),

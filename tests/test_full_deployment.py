@@ -16,7 +16,7 @@ import pytest
 
 from binder_amd import REPO_ROOT, cli
 from binder_amd.digclient import dig
-from binder_amd.harness import free_port
+from binder_amd.harness import free_port, BALANCERD
 from binder_amd.stubzk import StubZk
 from binder_amd.zkclient import ZkConn
 
@@ -66,7 +66,7 @@ def test_full_stack(tmp_path):
         bport = free_port()
         stats = tmp_path / "stats.sock"
         bal = subprocess.Popen(
-            [str(BIN / "binder-balancer"), "-p", str(bport),
+            [str(BALANCERD), "-p", str(bport),
              "-H", "127.0.0.1", "-s", str(statedir / "sockets"),
              "-S", str(stats), "-r", "100"],
             env=dict(os.environ, LOG_LEVEL="warn"),

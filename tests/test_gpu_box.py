@@ -14,7 +14,7 @@ import time
 import pytest
 
 from binder_amd import REPO_ROOT, require_native
-from binder_amd.harness import BinderProcess, free_port
+from binder_amd.harness import BinderProcess, free_port, BALANCERD
 from binder_amd.stubzk import StubZk
 
 pytestmark = pytest.mark.gpu
@@ -52,7 +52,7 @@ def test_full_stack_sustained_load(tmp_path):
             b.wait_ready("h999.foo.com", timeout=30)
             port = free_port()
             bal = subprocess.Popen(
-                [str(REPO_ROOT / "bin" / "binder-balancer"),
+                [str(BALANCERD),
                  "-p", str(port), "-H", "127.0.0.1",
                  "-s", str(sockdir), "-r", "100"],
                 env=dict(os.environ, LOG_LEVEL="warn"),
