@@ -89,6 +89,7 @@ def start_backends(n, tmp, zk_port, last_name="h4999.foo.com"):
 
 
 def start_balancer(tmp, sockdir, port, workers=1):
+    from binder_amd.harness import BALANCERD
     env = dict(os.environ, LOG_LEVEL="warn")
     proc = subprocess.Popen(
         [str(BALANCERD), "-p", str(port),
