@@ -202,6 +202,28 @@ class PyStubEngine {
             py::bytes((const char*)out.data(), out.size()), "respond");
     }
 
+    /* Microbenchmark: run the wire query N times in a tight C++ loop
+     * (BASELINE config 1 without Python call overhead). Returns
+     * queries/second. */
+    double benchWire(py::bytes wire, size_t iters, size_t maxSize) {
+        std::string_view sv = std::string_view(wire);
+        std::vector<uint8_t> out;
+        struct timespec t0, t1;
+        clock_gettime(CLOCK_MONOTONIC, &t0);
+        for (size_t i = 0; i < iters; ++i) {
+            auto q = dns::Message::decode((const uint8_t*)sv.data(),
+                                          sv.size());
+            if (!q) return 0;
+            dns::Message resp;
+            engine_->handle(*q, resp);
+            resp.encodeInto(out, maxSize);
+        }
+        clock_gettime(CLOCK_MONOTONIC, &t1);
+        double secs = (double)(t1.tv_sec - t0.tv_sec) +
+                      (double)(t1.tv_nsec - t0.tv_nsec) / 1e9;
+        return secs > 0 ? (double)iters / secs : 0;
+    }
+
     /* Convenience: query by name/type, response as dict. */
     py::dict query(const std::string& name, const std::string& type,
                    bool rd) {
@@ -284,6 +306,8 @@ PYBIND11_MODULE(_native, m) {
         .def("set_ready", &PyStubEngine::setReady)
         .def("query_wire", &PyStubEngine::queryWire, py::arg("wire"),
              py::arg("max_size") = 0)
+        .def("bench_wire", &PyStubEngine::benchWire, py::arg("wire"),
+             py::arg("iters"), py::arg("max_size") = 512)
         .def("query", &PyStubEngine::query, py::arg("name"),
              py::arg("type") = "A", py::arg("rd") = false);
 }

@@ -72,6 +72,9 @@ def config1(n_queries):
     report(1, "in_proc_lookups_per_sec", n_queries / dt, "queries/s",
            note="single A vs in-process stub store, one thread, "
                 "includes Python call overhead")
+    qps = e.bench_wire(wire, n_queries)
+    report(1, "in_proc_lookups_per_sec_native_loop", qps, "queries/s",
+           note="same path, C++ loop (decode+resolve+encode only)")
 
 
 def config2(tmp, n_queries):
