@@ -178,6 +178,7 @@ class Balancer {
     uint64_t udpQueries_ = 0, udpReplies_ = 0, drops_ = 0;
 
     static constexpr size_t kMaxRemotes = 262144;
+    static constexpr size_t kMaxPending = 16384;
     static constexpr int64_t kReplyTtlMs = 3000;
     static constexpr int64_t kPingIntervalMs = 2000;
     static constexpr int64_t kPingTimeoutMs = 6000;
@@ -626,6 +627,14 @@ void Balancer::onUdpReadable() {
             Backend* be = pickBackendFast(addrs[i]);
             udpQueries_++;
             if (be == nullptr) {
+                drops_++;
+                continue;
+            }
+            /* load shedding: a backend this far behind will answer
+             * past any client deadline — drop now (clients retry)
+             * rather than queue into a multi-ms tail */
+            if (be->pending.size() > kMaxPending ||
+                be->out.size() > (4u << 20)) {
                 drops_++;
                 continue;
             }
