@@ -534,7 +534,9 @@ bool DnsServer::process(const uint8_t* data, size_t len, bool udp,
     int64_t tParse = nowUs();
 
     Message& query = *parsed;
-    size_t limit = 0;
+    /* TCP responses must fit the u16 length prefix; oversize answers
+     * degrade to TC like UDP rather than corrupting the stream */
+    size_t limit = 65535;
     bool hasEdns = query.edns() != nullptr;
     if (udp) {
         limit = 512;

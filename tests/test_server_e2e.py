@@ -154,3 +154,32 @@ def test_tcp_pipelining(server):
             m = n.decode_message(data)
             assert m["id"] == 100 + i
             assert m["rcode"] == "NOERROR"
+
+
+def test_huge_service_tcp_fits_length_prefix(tmp_path):
+    """A service with enough members to exceed 64KB must not corrupt
+    DNS-over-TCP framing (length prefix is u16): the response is
+    truncated with TC instead."""
+    import json as _json
+    tree = {"foo.com": None,
+            "big.foo.com": {"type": "service",
+                            "service": {"srvce": "_x", "proto": "_tcp",
+                                        "port": 1}}}
+    for i in range(3000):
+        tree[f"m{i}.big.foo.com"] = {
+            "type": "rr_host",
+            "rr_host": {"address": f"10.{i % 200}.{(i // 200) % 200}.9"}}
+    store = tmp_path / "t.json"
+    store.write_text(_json.dumps(tree))
+    from binder_amd.harness import BinderProcess
+    srv = BinderProcess(store=f"file:{store}", workdir=tmp_path,
+                        log_level="warn")
+    srv.start()
+    try:
+        r = srv.dig("big.foo.com", tcp=True, timeout=5)
+        # either all answers fit under 64KB, or TC is set — never a
+        # corrupted stream (dig would raise on garbage)
+        assert r.status == "NOERROR"
+        assert r["tc"] or len(r.answers) == 3000
+    finally:
+        srv.stop()
