@@ -120,7 +120,14 @@ void ZkMirror::onWatch(int32_t type, const std::string& path) {
     if (n == nullptr) return;  // removed meanwhile
     switch (type) {
     case EV_NODE_DATA_CHANGED:
-    case EV_NODE_CREATED: {
+    case EV_NODE_CREATED:
+    case EV_NODE_DELETED: {
+        /* NodeDeleted also re-probes: if the znode was deleted and
+         * recreated before our (delete-triggered) getChildren ran, the
+         * parent diff keeps this node object — whose data watch was
+         * just consumed. Re-reading re-arms it on the recreated node;
+         * a true deletion answers ZNONODE (no watch) and the parent's
+         * childrenChanged removes the node. */
         const std::string p = path;
         client_->getData(p, true,
                          [this, p](int32_t rc, const std::string& data,
@@ -138,9 +145,6 @@ void ZkMirror::onWatch(int32_t type, const std::string& path) {
             });
         break;
     }
-    case EV_NODE_DELETED:
-        /* parent's childrenChanged handles removal; nothing to re-arm */
-        break;
     default:
         break;
     }

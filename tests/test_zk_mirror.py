@@ -169,3 +169,40 @@ def test_ephemeral_nodes_vanish_on_session_expiry(stack):
         pytest.fail("ephemeral registration never vanished")
     # PTR gone too (reverse-map cleanup)
     assert srv.dig("1.0.66.10.in-addr.arpa", "PTR").status == "REFUSED"
+
+
+def test_rapid_delete_recreate_keeps_watches(stack):
+    """delete+recreate faster than watch processing must not leave a
+    retained node with a dead data watch (stale-forever hazard)."""
+    zk, srv = stack
+    srv.wait_ready("web.bar.foo.com")
+    for cycle in range(5):
+        # delete and recreate back-to-back (faster than the mirror's
+        # round trips)
+        zk.delete("/com/foo/bar/web")
+        jput(zk, "/com/foo/bar/web",
+             {"type": "host", "host": {"address": f"192.168.1.{cycle}"}})
+        deadline = time.time() + 10
+        while time.time() < deadline:
+            r = srv.dig("web.bar.foo.com")
+            if r.answers and r.answers[0]["address"] == \
+                    f"192.168.1.{cycle}":
+                break
+            time.sleep(0.05)
+        else:
+            pytest.fail(f"cycle {cycle}: recreated node stale")
+        # and a subsequent plain update must still propagate (the
+        # watch must be alive)
+        jput(zk, "/com/foo/bar/web",
+             {"type": "host",
+              "host": {"address": f"192.168.2.{cycle}"}})
+        deadline = time.time() + 10
+        while time.time() < deadline:
+            r = srv.dig("web.bar.foo.com")
+            if r.answers and r.answers[0]["address"] == \
+                    f"192.168.2.{cycle}":
+                break
+            time.sleep(0.05)
+        else:
+            pytest.fail(f"cycle {cycle}: data watch died after "
+                        "delete+recreate")
