@@ -159,6 +159,13 @@ int main(int argc, char** argv) {
         if (!loadFileStore(fileStore, storeMode.substr(5), log)) return 1;
         store = &fileStore;
     } else if (storeMode == "zk") {
+        if (dnsDomain.empty()) {
+            /* the reference's ZKCache asserts options.domain
+             * (lib/zk.js:23); an empty domain would mirror the whole
+             * ZK root here */
+            log.fatal("dnsDomain is required with the zk store");
+            return 1;
+        }
         const char* zh = getenv("ZK_HOST");
         const char* zp = getenv("ZK_PORT");
         std::string zkHost = zh && *zh ? zh : "127.0.0.1";
