@@ -682,6 +682,10 @@ void Balancer::onTcpAccept() {
         int fd = accept4(tcpFd_, (struct sockaddr*)&ss, &sl,
                          SOCK_NONBLOCK | SOCK_CLOEXEC);
         if (fd < 0) return;
+        if (tcpClients_.size() >= 4096) {  /* fd-exhaustion guard */
+            close(fd);
+            continue;
+        }
         auto c = std::make_shared<TcpClient>();
         c->fd = fd;
         c->src = ss;

@@ -197,6 +197,10 @@ void MetricsHttpServer::onAccept() {
         int fd = accept4(listenFd_, nullptr, nullptr,
                          SOCK_NONBLOCK | SOCK_CLOEXEC);
         if (fd < 0) break;
+        if (connBufs_.size() >= 256) {  /* half-open conn guard */
+            ::close(fd);
+            continue;
+        }
         connBufs_[fd] = "";
         loop_->addFd(fd, EPOLLIN,
                      [this, fd](uint32_t ev) { onConn(fd, ev); });

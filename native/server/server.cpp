@@ -726,6 +726,10 @@ void DnsServer::onTcpAccept() {
         int fd = accept4(tcpFd_, (struct sockaddr*)&ss, &sl,
                          SOCK_NONBLOCK | SOCK_CLOEXEC);
         if (fd < 0) return;
+        if (tcpConns_.size() >= 4096) {  /* fd-exhaustion guard */
+            close(fd);
+            continue;
+        }
         auto conn = std::make_shared<TcpConn>();
         conn->fd = fd;
         conn->lastActivityMs = monotonicMillis();
