@@ -147,3 +147,12 @@ def test_srv_forwarding_passthrough(tmp_path):
     finally:
         local.stop()
         upstream.stop()
+
+
+def test_recursion_over_tcp(dcs):
+    """Misses arriving over TCP also forward (async reply on the TCP
+    conn)."""
+    local, _ = dcs
+    r = local.dig("svc.dc2.foo.com", rd=True, tcp=True, timeout=6)
+    assert r.status == "NOERROR"
+    assert r.answers[0]["address"] == "10.22.0.1"
