@@ -181,3 +181,32 @@ def test_supervisor_sighup_rescans(sup):
     sup["proc"].send_signal(signal.SIGHUP)
     wait_online(statedir, [f"binder-{port}"])
     assert dig("web.foo.com", port=port, timeout=3).status == "NOERROR"
+
+
+def test_full_scale_32_instances(sup):
+    """The reference caps at 32 processes per zone (boot/setup.sh:15);
+    converge to the cap, verify serving, then to zero."""
+    statedir, cfg, tree = sup["statedir"], sup["cfg"], sup["tree"]
+    base = BASE + 100
+    r = run_adjust(statedir, 32, base, cfg, tree, wait=60)
+    assert r.returncode == 0, r.stderr
+    names = [f"binder-{base + i}" for i in range(32)]
+    st = wait_online(statedir, names, timeout=60)
+    assert len(st) == 32
+    # spot-check serving across the range
+    for port in (base, base + 15, base + 31):
+        assert dig("web.foo.com", port=port, timeout=3).status == \
+            "NOERROR"
+    # beyond the cap is rejected (smf_adjust.c:904-909 bounds)
+    r = run_adjust(statedir, 33, base, cfg, tree)
+    assert r.returncode != 0
+    # converge to zero: all instances drained
+    r = run_adjust(statedir, 0, base, cfg, tree)
+    assert r.returncode == 0
+    deadline = time.time() + 30
+    while time.time() < deadline:
+        if not read_status(statedir):
+            break
+        time.sleep(0.3)
+    else:
+        pytest.fail(f"instances left: {read_status(statedir)}")
