@@ -33,7 +33,7 @@ PY_INCLUDES := $(shell $(PYTHON) -m pybind11 --includes)
 PYMOD := binder_amd/_native$(PY_EXT_SUFFIX)
 
 BINARIES := bin/binderd bin/binder-balancer bin/dnsblast \
-	bin/binder-adjust bin/binder-supervisor bin/zklogcat
+	bin/binder-adjust bin/binder-supervisor bin/zklogcat bin/zktool
 
 all: $(PYMOD) $(BINARIES)
 
@@ -58,6 +58,10 @@ bin/binder-supervisor: $(CORE_OBJS) $(BUILD)/native/adjust/supervisor_main.o
 	$(CXX) $(CXXFLAGS) $^ -o $@ $(LDFLAGS)
 
 bin/zklogcat: $(CORE_OBJS) $(BUILD)/native/zklog/zklogcat_main.o
+	@mkdir -p bin
+	$(CXX) $(CXXFLAGS) $^ -o $@ $(LDFLAGS)
+
+bin/zktool: $(CORE_OBJS) $(BUILD)/native/zk/client.o $(BUILD)/native/zk/zktool_main.o
 	@mkdir -p bin
 	$(CXX) $(CXXFLAGS) $^ -o $@ $(LDFLAGS)
 
@@ -90,7 +94,7 @@ release: all
 	rm -rf dist/binder-amd
 	mkdir -p dist/binder-amd/bin dist/binder-amd/etc
 	cp bin/binderd bin/binder-balancer bin/binder-adjust \
-	    bin/binder-supervisor bin/zklogcat bin/dnsblast \
+	    bin/binder-supervisor bin/zklogcat bin/dnsblast bin/zktool \
 	    dist/binder-amd/bin/
 	cp -r deploy dist/binder-amd/
 	cp -r tools dist/binder-amd/
