@@ -124,3 +124,44 @@ def test_filters(tmp_path):
     # window filter: everything is recent -> all kept
     txns = run_zklogcat(["-t", "3600", log])
     assert any(t.get("path") == "/w0" for t in txns)
+
+
+def test_multi_txn_decode(tmp_path):
+    """MULTI transactions decode recursively (zklog.c:270-414 parity);
+    crafted per the FileTxnLog jute format."""
+    import struct
+    import zlib
+    logdir = tmp_path / "txnlog"
+    zk = StubZk(txnlog_dir=str(logdir)).start()
+    zk.put("/seed", b"null")
+    zk.stop()
+
+    def jstr(b):
+        if isinstance(b, str):
+            b = b.encode()
+        return struct.pack(">i", len(b)) + b
+
+    # inner op 1: create /m1 (path, data, acl vec, ephemeral, pcver)
+    create_rec = jstr("/m1") + jstr(b"abc") + struct.pack(">i", 1) + \
+        struct.pack(">i", 31) + jstr("world") + jstr("anyone") + \
+        b"\x00" + struct.pack(">i", 1)
+    # inner op 2: delete /m0
+    delete_rec = jstr("/m0")
+    multi_body = struct.pack(">i", 2) + \
+        struct.pack(">i", 1) + jstr(create_rec) + \
+        struct.pack(">i", 2) + jstr(delete_rec)
+    hdr = struct.pack(">qiqqi", 0x77, 9, 100, 1789000000000, 14)
+    txn = hdr + multi_body
+    crc = zlib.adler32(txn) & 0xFFFFFFFF
+    with open(logdir / "log.1", "ab") as f:
+        f.write(struct.pack(">qi", crc, len(txn)) + txn + b"\x42")
+
+    txns = run_zklogcat(["-d", str(logdir / "log.1")])
+    multi = [t for t in txns if t["type"] == "multi"]
+    assert len(multi) == 1
+    ops = multi[0]["ops"]
+    assert ops[0]["type"] == "create"
+    assert ops[0]["path"] == "/m1"
+    assert ops[0]["data"] == "616263"  # "abc" hex
+    assert ops[1]["type"] == "delete"
+    assert ops[1]["path"] == "/m0"
