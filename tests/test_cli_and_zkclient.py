@@ -71,3 +71,69 @@ def test_cli_zk_ops(zk, capsys):
     assert '{"k":1}' in out
     cli.main(["zk", "ls", "/x", "--zk-port", str(zk.port)])
     assert "y" in capsys.readouterr().out
+
+
+def test_cli_dig_and_balstat_and_status(tmp_path, capsys):
+    """The operator CLI verbs end-to-end: dig against binderd, balstat
+    against a live balancer, status against a supervisor state dir."""
+    import json as _json
+    import os
+    import subprocess
+    import time
+
+    from binder_amd import REPO_ROOT
+    from binder_amd.harness import BALANCERD, BinderProcess, free_port
+
+    sockdir = tmp_path / "socks"
+    sockdir.mkdir()
+    store = tmp_path / "tree.json"
+    store.write_text(_json.dumps(
+        {"foo.com": None,
+         "web.foo.com": {"type": "host",
+                         "host": {"address": "1.2.3.4"}}}))
+    srv = BinderProcess(store=f"file:{store}", workdir=tmp_path,
+                        balancer_socket=str(sockdir / "b0"))
+    srv.start()
+    stats = tmp_path / "stats.sock"
+    port = free_port()
+    bal = subprocess.Popen(
+        [str(BALANCERD), "-p", str(port), "-H", "127.0.0.1",
+         "-s", str(sockdir), "-S", str(stats), "-r", "100"],
+        env=dict(os.environ, LOG_LEVEL="warn"),
+        stdout=subprocess.DEVNULL, stderr=subprocess.STDOUT)
+    try:
+        deadline = time.time() + 15
+        while time.time() < deadline and not stats.exists():
+            time.sleep(0.1)
+        time.sleep(0.5)
+
+        assert cli.main(["dig", "web.foo.com", "-s", "127.0.0.1",
+                         "-p", str(srv.port)]) == 0
+        out = capsys.readouterr().out
+        assert "NOERROR" in out and "1.2.3.4" in out
+
+        assert cli.main(["dig", "web.foo.com", "A", "-s", "127.0.0.1",
+                         "-p", str(port), "--tcp"]) == 0
+        assert "NOERROR" in capsys.readouterr().out
+
+        assert cli.main(["balstat", str(stats)]) == 0
+        out = capsys.readouterr().out
+        assert "PATH" in out and "b0" in out
+        assert cli.main(["balstat", str(stats), "--json"]) == 0
+        st = _json.loads(capsys.readouterr().out)
+        assert st["backends"][0]["ok"] is True
+
+        # status against a fabricated state dir
+        state = tmp_path / "state"
+        state.mkdir()
+        (state / "status.json").write_text(_json.dumps(
+            {"instances": {"binder-5301": {
+                "pid": 42, "state": "online", "restarts": 0,
+                "since": 0, "port": 5301}}}))
+        assert cli.main(["status", "-d", str(state)]) == 0
+        out = capsys.readouterr().out
+        assert "online" in out and "binder-5301" in out
+    finally:
+        bal.terminate()
+        bal.wait(timeout=5)
+        srv.stop()
