@@ -211,7 +211,13 @@ def main():
             stack.append(lambda: [b.stop() for b in backends])
             bal_port = free_port()
             bal = start_balancer(tmp, sockdir, bal_port, workers=workers)
-            stack.append(bal.terminate)
+            def stop_bal():
+                bal.terminate()
+                try:
+                    bal.wait(timeout=5)
+                except subprocess.TimeoutExpired:
+                    bal.kill()
+            stack.append(stop_bal)
             wait_balancer_ready(bal_port, n, tmp)
             log(f"balancer ready on :{bal_port}; warmup "
                 f"{args.warmup} x {q_step} queries")
