@@ -80,11 +80,19 @@ def cmd_register(args):
     service (port 53, ttl 60) and a child rr_host record per address —
     the layout binder itself is discovered by
     (sapi_manifests/registrar/template:1-30).
+
+    With --hold the child is EPHEMERAL and this command stays attached
+    as a minimal registrar daemon: the registration disappears the
+    moment the process (or its session) dies — the production
+    liveness mechanism.
     """
-    from .zkclient import ZkConn
+    import time as _time
+
+    from .zkclient import ZkConn, ZkError
     domain = args.domain
     path = "/" + "/".join(reversed(domain.split(".")))
-    with ZkConn(args.zk_host, args.zk_port) as zk:
+    zk = ZkConn(args.zk_host, args.zk_port)
+    try:
         zk.mkdirp(path)
         svc = {
             "type": "service",
@@ -99,10 +107,29 @@ def cmd_register(args):
             "ttl": 30,
         }
         child = f"{path}/{args.instance}"
+        if args.hold:
+            try:
+                zk.delete(child)
+            except ZkError:
+                pass
+            zk.create(child, json.dumps(host).encode(), flags=1)
+            print(f"registered {args.instance} ({args.address}) under "
+                  f"{domain} (ephemeral); holding session — Ctrl-C to "
+                  f"deregister", flush=True)
+            try:
+                while True:
+                    _time.sleep(5)
+                    zk.ping()
+            except KeyboardInterrupt:
+                pass
+            return 0
         zk.mkdirp(child)
         zk.set(child, json.dumps(host).encode())
-    print(f"registered {args.instance} ({args.address}) under {domain}")
-    return 0
+        print(f"registered {args.instance} ({args.address}) under "
+              f"{domain}")
+        return 0
+    finally:
+        zk.close()
 
 
 def cmd_zk(args):
@@ -154,6 +181,9 @@ def main(argv=None):
     p.add_argument("-p", "--port", type=int, default=53)
     p.add_argument("--zk-host", default="127.0.0.1")
     p.add_argument("--zk-port", type=int, default=2181)
+    p.add_argument("--hold", action="store_true",
+                   help="ephemeral registration; stay attached as a "
+                        "minimal registrar daemon")
     p.set_defaults(fn=cmd_register)
 
     p = sub.add_parser("zk")

@@ -141,6 +141,21 @@ class ZkConn:
             if e.code != ZNONODE:
                 raise
 
+    def ping(self):
+        """Keepalive (xid -2, op 11); required to hold a session open
+        past the negotiated timeout (e.g. for ephemeral registrations).
+        """
+        self._send(struct.pack(">ii", -2, 11))
+        while True:
+            resp = self._recv()
+            (rxid,) = struct.unpack_from(">i", resp, 0)
+            if rxid == -2:
+                return
+            if rxid == -1:
+                continue  # watch notification
+            # a reply to an earlier call would be a protocol bug here
+            raise ZkError(-2, "ping", "unexpected xid")
+
     def close(self):
         try:
             self._send(struct.pack(">ii", self._xid + 1, OP_CLOSE))

@@ -137,3 +137,47 @@ def test_cli_dig_and_balstat_and_status(tmp_path, capsys):
         bal.terminate()
         bal.wait(timeout=5)
         srv.stop()
+
+
+def test_register_hold_is_ephemeral(zk, tmp_path):
+    """`register --hold` keeps an ephemeral registration alive via
+    session pings; killing the holder deregisters it."""
+    import json as _json
+    import signal
+    import subprocess
+    import sys
+    import time
+
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "binder_amd", "register",
+         "api.coal.foo.com", "10.77.0.9", "-i", "h0", "--hold",
+         "--zk-host", "127.0.0.1", "--zk-port", str(zk.port)],
+        stdout=subprocess.PIPE, text=True)
+    try:
+        deadline = time.time() + 10
+        while time.time() < deadline:
+            if zk.exists("/com/foo/coal/api/h0"):
+                break
+            time.sleep(0.1)
+        else:
+            pytest.fail("ephemeral registration never appeared")
+        data = _json.loads(zk.get("/com/foo/coal/api/h0").decode())
+        assert data["rr_host"]["address"] == "10.77.0.9"
+        # holder dies => session closes => node reaped... our stub
+        # reaps on session close/expiry; SIGKILL leaves the session
+        # until timeout, so use expire to model it deterministically
+        proc.send_signal(signal.SIGKILL)
+        proc.wait(timeout=5)
+        # TCP close makes the stub drop the conn; expire the sessions
+        # (as ZK would at session timeout) and the node must go
+        zk.expire_sessions()
+        deadline = time.time() + 5
+        while time.time() < deadline:
+            if not zk.exists("/com/foo/coal/api/h0"):
+                break
+            time.sleep(0.1)
+        else:
+            pytest.fail("ephemeral registration survived expiry")
+    finally:
+        if proc.poll() is None:
+            proc.kill()
