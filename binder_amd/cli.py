@@ -9,6 +9,9 @@ Subcommands (the reference's operational tool surface):
              registrar does (sapi_manifests/registrar/template: rr_host
              node + _dns._udp SRV service, ttl 60)
   zk         mkdirp/rmr/get/ls against a ZK server (test fixture ops)
+  zkd        run the built-in single-node registry daemon (stub ZK
+             with txn-log durability) — for self-contained
+             deployments without a ZooKeeper ensemble
 """
 from __future__ import annotations
 
@@ -150,6 +153,28 @@ def cmd_zk(args):
     return 0
 
 
+def cmd_zkd(args):
+    import signal
+    import time as _time
+
+    from .stubzk import StubZk
+    zk = StubZk(host=args.host, port=args.port,
+                txnlog_dir=args.data_dir)
+    zk.start()
+    print(f"zkd listening on {args.host}:{zk.port}, "
+          f"data in {args.data_dir} ({zk.node_count()} nodes restored)",
+          flush=True)
+    stop = []
+    signal.signal(signal.SIGTERM, lambda *a: stop.append(1))
+    try:
+        while not stop:
+            _time.sleep(0.5)
+    except KeyboardInterrupt:
+        pass
+    zk.stop()
+    return 0
+
+
 def main(argv=None):
     ap = argparse.ArgumentParser(prog="binder-amd")
     sub = ap.add_subparsers(dest="cmd", required=True)
@@ -185,6 +210,12 @@ def main(argv=None):
                    help="ephemeral registration; stay attached as a "
                         "minimal registrar daemon")
     p.set_defaults(fn=cmd_register)
+
+    p = sub.add_parser("zkd")
+    p.add_argument("-H", "--host", default="127.0.0.1")
+    p.add_argument("-p", "--port", type=int, default=2181)
+    p.add_argument("-d", "--data-dir", default="/var/lib/binder-zkd")
+    p.set_defaults(fn=cmd_zkd)
 
     p = sub.add_parser("zk")
     p.add_argument("op", choices=["mkdirp", "rmr", "get", "ls", "set"])

@@ -75,15 +75,6 @@ class StubZk:
         self.host = host
         self._port = port
         self.session_timeout_ms = session_timeout_ms
-        # Optional on-disk transaction log in the real ZooKeeper
-        # FileTxnLog v2 format (magic ZKLG) — lets zklogcat be tested
-        # against logs this stub writes.
-        self._txnlog = None
-        if txnlog_dir:
-            os.makedirs(txnlog_dir, exist_ok=True)
-            self._txnlog = open(os.path.join(txnlog_dir, "log.1"), "wb")
-            self._txnlog.write(struct.pack(">iiq", 0x5A4B4C47, 2, 0))
-            self._txnlog.flush()
         self._lock = threading.RLock()
         self._nodes: Dict[str, _Node] = {"/": _Node()}
         self._conns: Set[_Conn] = set()
@@ -96,6 +87,85 @@ class StubZk:
         self._running = False
         self._wake_r, self._wake_w = socket.socketpair()
         self.stats = {"sessions": 0, "ops": 0, "watches_fired": 0}
+        # Optional on-disk transaction log in the real ZooKeeper
+        # FileTxnLog v2 format (magic ZKLG) — lets zklogcat be tested
+        # against logs this stub writes, and gives `binder-amd zkd`
+        # restart durability via replay.
+        self._txnlog = None
+        self._txnlog_path = None
+        if txnlog_dir:
+            os.makedirs(txnlog_dir, exist_ok=True)
+            self._txnlog_path = os.path.join(txnlog_dir, "log.1")
+            fresh = not os.path.exists(self._txnlog_path)
+            if not fresh:
+                self._replay(self._txnlog_path)
+            self._txnlog = open(self._txnlog_path, "ab")
+            if fresh:
+                self._txnlog.write(struct.pack(">iiq", 0x5A4B4C47, 2, 0))
+                self._txnlog.flush()
+
+    def _replay(self, path: str):
+        """Rebuild the tree from a FileTxnLog (durability for zkd).
+        Ephemeral creates are skipped: their sessions are gone, which
+        is exactly what a real ZK restart + session expiry yields."""
+        with open(path, "rb") as f:
+            data = f.read()
+        if len(data) < 16 or                 struct.unpack_from(">i", data, 0)[0] != 0x5A4B4C47:
+            return
+        off = 16
+        while off + 12 <= len(data):
+            crc, tlen = struct.unpack_from(">qi", data, off)
+            if crc == 0 or tlen <= 0 or off + 12 + tlen + 1 > len(data):
+                break
+            txn = data[off + 12:off + 12 + tlen]
+            off += 12 + tlen + 1
+            (_cid, _cxid, zxid, _t, ttype) = struct.unpack_from(
+                ">qiqqi", txn, 0)
+            body = txn[32:]  # TxnHeader is 8+4+8+8+4 bytes
+            self._zxid = max(self._zxid, zxid)
+
+            def jstr(b, o):
+                (n,) = struct.unpack_from(">i", b, o)
+                if n < 0:
+                    return b"", o + 4
+                return b[o + 4:o + 4 + n], o + 4 + n
+
+            if ttype == 1:  # create
+                p, o = jstr(body, 0)
+                d, o = jstr(body, o)
+                (nacl,) = struct.unpack_from(">i", body, o)
+                o += 4
+                for _ in range(max(nacl, 0)):
+                    o += 4
+                    _s1, o = jstr(body, o)
+                    _s2, o = jstr(body, o)
+                ephemeral = body[o] != 0
+                path_s = p.decode()
+                if ephemeral:
+                    continue
+                parent = _parent(path_s)
+                pn = self._nodes.get(parent)
+                if pn is None or path_s in self._nodes:
+                    continue
+                self._nodes[path_s] = _Node(d)
+                pn.children.add(path_s[path_s.rfind("/") + 1:])
+            elif ttype == 5:  # setData
+                p, o = jstr(body, 0)
+                d, o = jstr(body, o)
+                n = self._nodes.get(p.decode())
+                if n is not None:
+                    n.data = d
+                    n.version += 1
+            elif ttype == 2:  # delete
+                p, _ = jstr(body, 0)
+                path_s = p.decode()
+                n = self._nodes.get(path_s)
+                if n is not None and not n.children:
+                    del self._nodes[path_s]
+                    pn = self._nodes.get(_parent(path_s))
+                    if pn is not None:
+                        pn.children.discard(
+                            path_s[path_s.rfind("/") + 1:])
 
     # ------------- lifecycle -------------
 
