@@ -2,7 +2,8 @@
 
 Answers simple binds and subtree searches with a configured resolver
 list, the way UFDS answers binder's listResolvers(region) search
-(reference recursion.js:17-19, 210-219). Plain TCP only.
+(reference recursion.js:17-19, 210-219). Plain TCP, or TLS (ldaps)
+when cert/key paths are given.
 """
 from __future__ import annotations
 
@@ -57,9 +58,15 @@ def _read_tlv(data: bytes, off: int):
 
 
 class StubLdap:
-    def __init__(self, host="127.0.0.1", port=0):
+    def __init__(self, host="127.0.0.1", port=0, tls_cert=None,
+                 tls_key=None):
         self.host = host
         self._port = port
+        self._ssl_ctx = None
+        if tls_cert:
+            import ssl
+            self._ssl_ctx = ssl.SSLContext(ssl.PROTOCOL_TLS_SERVER)
+            self._ssl_ctx.load_cert_chain(tls_cert, tls_key)
         self.resolvers: List[Dict[str, str]] = []
         self.binds: List[str] = []
         self.searches: List[str] = []
@@ -106,6 +113,8 @@ class StubLdap:
         conn.settimeout(10)
         buf = b""
         try:
+            if self._ssl_ctx is not None:
+                conn = self._ssl_ctx.wrap_socket(conn, server_side=True)
             while True:
                 r = _read_tlv(buf, 0)
                 if r is None:
@@ -158,5 +167,7 @@ class StubLdap:
                     return
         except (socket.timeout, OSError):
             return
+        except Exception:
+            return  # TLS handshake failures etc.
         finally:
             conn.close()

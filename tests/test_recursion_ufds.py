@@ -120,3 +120,62 @@ def test_ufds_bad_credentials_best_effort(tmp_path):
     finally:
         zk.stop()
         ldap.stop()
+
+
+def test_ufds_over_ldaps(tmp_path):
+    """ldaps:// UFDS with a self-signed cert (tlsVerify off, like the
+    reference's internal-CA deployments): the native TLS client must
+    bind+search successfully."""
+    import subprocess
+    cert = tmp_path / "cert.pem"
+    key = tmp_path / "key.pem"
+    subprocess.run(
+        ["openssl", "req", "-x509", "-newkey", "rsa:2048",
+         "-keyout", str(key), "-out", str(cert), "-days", "2",
+         "-nodes", "-subj", "/CN=ufds.test"],
+        check=True, capture_output=True)
+    ldap = StubLdap(tls_cert=str(cert), tls_key=str(key)).start()
+    ldap.resolvers = [{"datacenter": "dc2", "ip": "127.0.0.2"}]
+
+    up_tree = tmp_path / "up.json"
+    up_tree.write_text(json.dumps({
+        "dc2.foo.com": None,
+        "tls.dc2.foo.com": {"type": "host",
+                            "host": {"address": "10.31.0.1"}}}))
+    upstream = BinderProcess(dns_domain="dc2.foo.com", datacenter="dc2",
+                             host="127.0.0.2", store=f"file:{up_tree}",
+                             workdir=tmp_path)
+    upstream.start()
+
+    local_tree = tmp_path / "local.json"
+    local_tree.write_text('{"foo.com": null}')
+    local = BinderProcess(
+        dns_domain="foo.com", datacenter="dc1",
+        store=f"file:{local_tree}", workdir=tmp_path,
+        log_path=str(tmp_path / "l.log"),
+        config={"recursion": {
+            "source": "ufds", "regionName": "r1",
+            "dnsDomain": "foo.com", "upstreamPort": upstream.port,
+            "ufds": {"url": f"ldaps://127.0.0.1:{ldap.port}",
+                     "bindDN": "cn=root", "bindPassword": "pw"},
+        }})
+    local.start()
+    try:
+        deadline = time.time() + 40
+        while time.time() < deadline:
+            try:
+                r = local.dig("tls.dc2.foo.com", rd=True, timeout=4)
+                if r.status == "NOERROR":
+                    break
+            except OSError:
+                pass
+            time.sleep(0.5)
+        else:
+            log = open(str(tmp_path / "l.log")).read()[-2000:]
+            pytest.fail(f"ldaps discovery never worked; {log}")
+        assert r.answers[0]["address"] == "10.31.0.1"
+        assert any("region=r1" in s for s in ldap.searches)
+    finally:
+        local.stop()
+        upstream.stop()
+        ldap.stop()
