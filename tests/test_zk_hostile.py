@@ -76,11 +76,17 @@ def test_binderd_survives_hostile_zk(tmp_path, mode):
                         log_level="warn")
     srv.start()
     try:
-        end = time.time() + 4
+        end = time.time() + 5
+        answered = 0
         while time.time() < end:
-            r = srv.dig("web.foo.com", timeout=2)
+            try:
+                r = srv.dig("web.foo.com", timeout=2)
+            except OSError:
+                continue  #机loaded CI box: a dropped probe is fine
             assert r.status == "SERVFAIL"  # mirror never materialized
+            answered += 1
             time.sleep(0.2)
+        assert answered >= 3, "server stopped answering"
         assert srv.proc.poll() is None, "binderd died"
         assert zk.accepted >= 1
     finally:
