@@ -82,7 +82,7 @@ def test_binderd_survives_hostile_zk(tmp_path, mode):
             try:
                 r = srv.dig("web.foo.com", timeout=2)
             except OSError:
-                continue  #机loaded CI box: a dropped probe is fine
+                continue  # loaded CI box: a dropped probe is fine
             assert r.status == "SERVFAIL"  # mirror never materialized
             answered += 1
             time.sleep(0.2)
