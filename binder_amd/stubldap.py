@@ -8,7 +8,6 @@ when cert/key paths are given.
 from __future__ import annotations
 
 import socket
-import struct
 import threading
 from typing import Dict, List, Optional
 

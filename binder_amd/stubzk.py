@@ -23,7 +23,7 @@ import struct
 import threading
 import time
 import zlib
-from typing import Dict, Optional, Set, Tuple
+from typing import Dict, Optional, Set
 
 # op codes
 OP_CREATE, OP_DELETE, OP_EXISTS, OP_GETDATA, OP_SETDATA = 1, 2, 3, 4, 5

@@ -10,7 +10,7 @@ from __future__ import annotations
 
 import socket
 import struct
-from typing import List, Optional
+from typing import List
 
 OP_CREATE, OP_DELETE, OP_EXISTS, OP_GETDATA, OP_SETDATA = 1, 2, 3, 4, 5
 OP_GETCHILDREN, OP_CLOSE = 8, -11

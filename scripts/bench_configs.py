@@ -16,7 +16,6 @@ import os
 import subprocess
 import sys
 import tempfile
-import threading
 import time
 from pathlib import Path
 
@@ -24,7 +23,7 @@ REPO = Path(__file__).resolve().parent.parent
 sys.path.insert(0, str(REPO))
 
 from binder_amd import require_native            # noqa: E402
-from binder_amd.harness import BinderProcess, free_port  # noqa: E402
+from binder_amd.harness import BinderProcess  # noqa: E402
 from binder_amd.stubzk import StubZk             # noqa: E402
 
 RESULTS = []

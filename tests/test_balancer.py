@@ -14,7 +14,6 @@ from pathlib import Path
 
 import pytest
 
-from binder_amd import REPO_ROOT
 from binder_amd.digclient import dig
 from binder_amd.harness import BinderProcess, free_port, BALANCERD
 

@@ -81,7 +81,6 @@ def test_cli_dig_and_balstat_and_status(tmp_path, capsys):
     import subprocess
     import time
 
-    from binder_amd import REPO_ROOT
     from binder_amd.harness import BALANCERD, BinderProcess, free_port
 
     sockdir = tmp_path / "socks"

@@ -3,8 +3,6 @@
 The reference has zero codec tests (its mname dep is trusted); SURVEY.md
 §7 stage 1 calls for exhaustive codec coverage as the first improvement.
 """
-import pytest
-
 from binder_amd import require_native
 
 n = require_native()

@@ -10,7 +10,6 @@ import os
 import socket
 import subprocess
 import time
-from pathlib import Path
 
 import pytest
 

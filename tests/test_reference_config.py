@@ -6,8 +6,6 @@ degrade to best-effort (server keeps serving, misses refused)."""
 import json
 import time
 
-import pytest
-
 from binder_amd.harness import BinderProcess
 from binder_amd.stubzk import StubZk
 

@@ -9,7 +9,7 @@ import time
 
 import pytest
 
-from binder_amd.harness import BinderProcess, free_port
+from binder_amd.harness import BinderProcess
 
 
 class GarbageZk:
