@@ -39,10 +39,13 @@
 
 using namespace bamd;
 
+static const char* kVersion = "binder-amd 0.1.0";
+
 static void usage(const char* name) {
     fprintf(stderr,
-            "usage: %s [-v] [-a cacheExpiry] [-s cacheSize] [-p port] "
-            "[-f file] [-b balancerSocket] [-S zk|file:<path>]\n",
+            "usage: %s [-v] [-V] [-a cacheExpiry] [-s cacheSize] "
+            "[-p port] [-f file] [-b balancerSocket] "
+            "[-S zk|file:<path>]\n",
             name);
 }
 
@@ -83,8 +86,9 @@ int main(int argc, char** argv) {
     Json cli = Json::object();
     int c;
     int verbosity = 0;
-    while ((c = getopt(argc, argv, "hva:b:s:p:f:S:")) != -1) {
+    while ((c = getopt(argc, argv, "hvVa:b:s:p:f:S:")) != -1) {
         switch (c) {
+        case 'V': printf("%s\n", kVersion); return 0;
         case 'a': cli.set("expiry", Json((int64_t)atoi(optarg))); break;
         case 'b': cli.set("balancerSocket", Json(std::string(optarg))); break;
         case 'f': configFile = optarg; break;
