@@ -183,3 +183,27 @@ def test_huge_service_tcp_fits_length_prefix(tmp_path):
         assert r["tc"] or len(r.answers) == 3000
     finally:
         srv.stop()
+
+
+def test_vestigial_flags_accepted(tmp_path):
+    """-s (cacheSize) and -a (cacheExpiry) are vestigial in the
+    reference (main.js:34-38) but must be accepted
+    for CLI compatibility; -V prints the version."""
+    import json as _json
+    import subprocess
+    from binder_amd.harness import BINDERD, BinderProcess
+    out = subprocess.run([str(BINDERD), "-V"], capture_output=True,
+                         text=True)
+    assert out.returncode == 0 and "binder-amd" in out.stdout
+
+    store = tmp_path / "t.json"
+    store.write_text(_json.dumps(
+        {"foo.com": None,
+         "w.foo.com": {"type": "host", "host": {"address": "1.1.1.1"}}}))
+    srv = BinderProcess(store=f"file:{store}", workdir=tmp_path)
+    srv.cmd += ["-s", "10000", "-a", "60000"]
+    srv.start()
+    try:
+        assert srv.dig("w.foo.com").status == "NOERROR"
+    finally:
+        srv.stop()
