@@ -25,6 +25,10 @@ BINDERD = Path(os.environ.get("BINDERD_BIN",
                               REPO_ROOT / "bin" / "binderd"))
 BALANCERD = Path(os.environ.get("BALANCER_BIN",
                                 REPO_ROOT / "bin" / "binder-balancer"))
+SUPERVISORD = Path(os.environ.get("SUPERVISOR_BIN",
+                                  REPO_ROOT / "bin" / "binder-supervisor"))
+ADJUST = Path(os.environ.get("ADJUST_BIN",
+                             REPO_ROOT / "bin" / "binder-adjust"))
 
 
 def free_port() -> int:

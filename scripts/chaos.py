@@ -26,7 +26,7 @@ REPO = Path(__file__).resolve().parent.parent
 sys.path.insert(0, str(REPO))
 
 from binder_amd.digclient import dig  # noqa: E402
-from binder_amd.harness import BALANCERD, free_port  # noqa: E402
+from binder_amd.harness import BALANCERD, free_port, SUPERVISORD, ADJUST  # noqa: E402
 from binder_amd.stubzk import StubZk  # noqa: E402
 
 BIN = REPO / "bin"
@@ -54,11 +54,11 @@ def main():
     env = dict(os.environ, LOG_LEVEL="warn", ZK_HOST="127.0.0.1",
                ZK_PORT=str(zk.port))
     sup = subprocess.Popen(
-        [str(BIN / "binder-supervisor"), "-d", str(statedir),
+        [str(SUPERVISORD), "-d", str(statedir),
          "-x", str(BIN / "binderd")], env=env,
         stdout=open(tmp / "sup.log", "ab"), stderr=subprocess.STDOUT)
     assert subprocess.run(
-        [str(BIN / "binder-adjust"), "-i", "3", "-B", str(BASE),
+        [str(ADJUST), "-i", "3", "-B", str(BASE),
          "-d", str(statedir), "-f", str(cfg), "-S", "zk", "-w", "30"],
         capture_output=True).returncode == 0
     bport = free_port()
@@ -119,7 +119,7 @@ def main():
             elif ev == "scale":
                 n = rng.choice([2, 3])
                 subprocess.run(
-                    [str(BIN / "binder-adjust"), "-i", str(n),
+                    [str(ADJUST), "-i", str(n),
                      "-B", str(BASE), "-d", str(statedir),
                      "-f", str(cfg), "-S", "zk"],
                     capture_output=True)
@@ -128,7 +128,7 @@ def main():
         time.sleep(2)
 
     # restore to 3 and let it settle
-    subprocess.run([str(BIN / "binder-adjust"), "-i", "3", "-B",
+    subprocess.run([str(ADJUST), "-i", "3", "-B",
                     str(BASE), "-d", str(statedir), "-f", str(cfg),
                     "-S", "zk", "-w", "30"], capture_output=True)
     time.sleep(3)

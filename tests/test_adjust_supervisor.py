@@ -16,12 +16,13 @@ import pytest
 
 from binder_amd import REPO_ROOT
 from binder_amd.digclient import dig
+from binder_amd.harness import ADJUST, SUPERVISORD
 
 BIN = REPO_ROOT / "bin"
 
 
 def run_adjust(statedir, count, base_port, cfg, tree, wait=0):
-    cmd = [str(BIN / "binder-adjust"), "-i", str(count),
+    cmd = [str(ADJUST), "-i", str(count),
            "-B", str(base_port), "-d", str(statedir),
            "-f", str(cfg), "-S", f"file:{tree}"]
     if wait:
@@ -52,7 +53,7 @@ def sup(tmp_path):
         "dnsDomain": "foo.com", "datacenterName": "coal",
         "host": "127.0.0.1"}))
     proc = subprocess.Popen(
-        [str(BIN / "binder-supervisor"), "-d", str(statedir),
+        [str(SUPERVISORD), "-d", str(statedir),
          "-x", str(BIN / "binderd")],
         env=dict(os.environ, LOG_LEVEL="info"),
         stdout=open(tmp_path / "sup.log", "ab"),

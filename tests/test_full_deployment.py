@@ -15,7 +15,7 @@ import pytest
 
 from binder_amd import REPO_ROOT, cli
 from binder_amd.digclient import dig
-from binder_amd.harness import free_port, BALANCERD
+from binder_amd.harness import free_port, BALANCERD, SUPERVISORD, ADJUST
 from binder_amd.stubzk import StubZk
 from binder_amd.zkclient import ZkConn
 
@@ -47,7 +47,7 @@ def test_full_stack(tmp_path):
 
         # supervisor + adjust converge 3 instances
         sup = subprocess.Popen(
-            [str(BIN / "binder-supervisor"), "-d", str(statedir),
+            [str(SUPERVISORD), "-d", str(statedir),
              "-x", str(BIN / "binderd")],
             env=dict(os.environ, LOG_LEVEL="warn",
                      ZK_HOST="127.0.0.1", ZK_PORT=str(zk.port)),
@@ -55,7 +55,7 @@ def test_full_stack(tmp_path):
             stderr=subprocess.STDOUT)
         procs.append(sup)
         rc = subprocess.run(
-            [str(BIN / "binder-adjust"), "-i", "3", "-B", str(BASE),
+            [str(ADJUST), "-i", "3", "-B", str(BASE),
              "-d", str(statedir), "-f", str(cfg), "-S", "zk",
              "-w", "30"],
             capture_output=True, text=True)
@@ -143,7 +143,7 @@ def test_full_stack(tmp_path):
 
         # scale down to 1: drained instances leave the balancer
         rc = subprocess.run(
-            [str(BIN / "binder-adjust"), "-i", "1", "-B", str(BASE),
+            [str(ADJUST), "-i", "1", "-B", str(BASE),
              "-d", str(statedir), "-f", str(cfg), "-S", "zk"],
             capture_output=True, text=True)
         assert rc.returncode == 0
