@@ -223,15 +223,34 @@ void MetricsHttpServer::onConn(int fd, uint32_t events) {
         req.find("\n\n") == std::string::npos && req.size() < 65536)
         return;  // keep reading until end of headers
 
-    std::string body = collector_->expose(staticLabels_);
-    char hdr[256];
-    snprintf(hdr, sizeof(hdr),
-             "HTTP/1.1 200 OK\r\n"
-             "Content-Type: text/plain; version=0.0.4\r\n"
-             "Content-Length: %zu\r\n"
-             "Connection: close\r\n\r\n",
-             body.size());
-    std::string resp = std::string(hdr) + body;
+    // Route like the reference's restify mount (triton-metrics serves
+    // only GET /metrics; main.js:134-152): anything else is a 404.
+    bool found = false;
+    if (req.compare(0, 4, "GET ") == 0) {
+        size_t sp = req.find(' ', 4);
+        if (sp != std::string::npos) {
+            std::string_view path(req.data() + 4, sp - 4);
+            found = (path == "/metrics");
+        }
+    }
+    std::string resp;
+    if (found) {
+        std::string body = collector_->expose(staticLabels_);
+        char hdr[256];
+        snprintf(hdr, sizeof(hdr),
+                 "HTTP/1.1 200 OK\r\n"
+                 "Content-Type: text/plain; version=0.0.4\r\n"
+                 "Content-Length: %zu\r\n"
+                 "Connection: close\r\n\r\n",
+                 body.size());
+        resp = std::string(hdr) + body;
+    } else {
+        resp = "HTTP/1.1 404 Not Found\r\n"
+               "Content-Type: application/json\r\n"
+               "Content-Length: 27\r\n"
+               "Connection: close\r\n\r\n"
+               "{\"code\":\"ResourceNotFound\"}";
+    }
     ssize_t rv = write(fd, resp.data(), resp.size());
     (void)rv;
     loop_->delFd(fd);
