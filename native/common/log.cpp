@@ -96,4 +96,29 @@ void Logger::write(LogLevel lv, const std::string& msg,
     (void)rv;
 }
 
+void Logger::logRaw(LogLevel lv, const char* msg,
+                    std::string_view rawFields) const {
+    if (!enabled(lv)) return;
+    static const std::string kHost = hostName();
+    thread_local std::string line;  // reused: no steady-state allocs
+    line.clear();
+    line += "{\"v\":0,\"name\":";
+    jsonEscape(name_, line);
+    line += ",\"hostname\":";
+    jsonEscape(kHost, line);
+    char buf[64];
+    snprintf(buf, sizeof(buf), ",\"pid\":%d,\"level\":%d", (int)getpid(),
+             (int)lv);
+    line += buf;
+    line += boundPrefix_;
+    line += rawFields;
+    line += ",\"msg\":\"";
+    line += msg;  // caller passes a literal needing no escaping
+    line += "\",\"time\":\"";
+    line += isoTimeNow();
+    line += "\"}\n";
+    ssize_t rv = ::write(fd_, line.data(), line.size());
+    (void)rv;
+}
+
 }  // namespace bamd

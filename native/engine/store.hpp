@@ -75,6 +75,7 @@ struct CompiledRecord {
      * stores additionally clear the PARENT's service cache when a
      * member changes). */
     mutable std::vector<uint8_t> wireA;  /* host-like A response */
+    mutable std::string logA;  /* pre-escaped answers element for wireA */
     /* service responses, segment-permuted per query to preserve the
      * reference's Fisher-Yates member shuffle: */
     struct ServiceCache {
@@ -82,11 +83,14 @@ struct CompiledRecord {
         std::vector<uint8_t> headA;    /* header+question, A query */
         std::vector<uint8_t> headSrv;  /* header+question, SRV query */
         /* per member: plain-A answer segment / SRV answer segment(s) /
-         * additional-A segment */
+         * additional-A segment, plus the matching pre-escaped log-line
+         * fragments (comma-joined quoted JSON elements) so the fast
+         * path can emit the per-query info log without the slow path */
         struct Member {
             std::vector<uint8_t> aSeg;
             std::vector<uint8_t> srvSeg;
             std::vector<uint8_t> addSeg;
+            std::string aLog, srvLog, addLog;
         };
         std::vector<Member> members;
         uint16_t srvAnCount = 0;  /* total SRV records over members */
@@ -94,6 +98,7 @@ struct CompiledRecord {
     mutable std::unique_ptr<ServiceCache> svc;
     void clearWireCaches() const {
         wireA.clear();
+        logA.clear();
         svc.reset();
     }
 };

@@ -114,10 +114,11 @@ class DnsServer {
                  const ClientInfo& ci, std::vector<uint8_t>& out,
                  std::function<void(std::vector<uint8_t>)> asyncReply);
     bool fastPath(const uint8_t* data, size_t len, bool udp,
-                  std::vector<uint8_t>& out);
+                  const ClientInfo& ci, std::vector<uint8_t>& out);
     std::string_view srvSvc_, srvProto_;   /* scratch, loop thread only */
     std::vector<uint32_t> shuffleIdx_;
     std::mt19937 rng_{0xb1d3};
+    std::string logFields_, logScratch_;   /* afterQuery log buffers */
 
   public:
     /* the store behind the engine (fast-path lookups); set by main */

@@ -1,8 +1,11 @@
 """Fast-path (prebuilt wire) parity: answers must be byte-equivalent in
 meaning to the slow path, and caches must invalidate on updates.
 
-The fast path only engages when info logging is off (LOG_LEVEL=warn),
-so run twin servers and diff responses.
+The fast path engages at warn AND info level (it emits the per-query
+info log from cached fragments); debug/trace still force the slow
+path, so the twin-diff runs one server at warn (fast) and one at debug
+(slow) and diffs responses. A separate test pins the fast path's
+info-level log lines against the slow path's format.
 """
 import json
 import time
@@ -36,7 +39,7 @@ def twins(tmp_path):
     fast = BinderProcess(store=f"file:{store}", workdir=tmp_path,
                          log_level="warn")
     slow = BinderProcess(store=f"file:{store}", workdir=tmp_path,
-                         log_level="info")
+                         log_level="debug")
     fast.start()
     slow.start()
     yield fast, slow
@@ -212,3 +215,60 @@ def test_service_fastpath_invalidation_on_member_change(tmp_path):
             srv.stop()
     finally:
         zk.stop()
+
+
+def test_fastpath_info_log_lines_match_slow_format(tmp_path):
+    """At info level the fast path emits the per-query bunyan line
+    itself (from cached fragments). Its fields must match the slow
+    path's line for the same query (modulo latency/timers values and
+    member shuffle order)."""
+    store = tmp_path / "tree.json"
+    store.write_text(json.dumps(TREE))
+    fast_log = tmp_path / "fast.log"
+    slow_log = tmp_path / "slow.log"
+    fast = BinderProcess(store=f"file:{store}", workdir=tmp_path,
+                         log_level="info", log_path=str(fast_log))
+    slow = BinderProcess(store=f"file:{store}", workdir=tmp_path,
+                         log_level="debug", log_path=str(slow_log))
+    fast.start()
+    slow.start()
+    try:
+        for name, qtype in [("web.foo.com", "A"),
+                            ("svc.foo.com", "A"),
+                            ("_x._tcp.svc.foo.com", "SRV")]:
+            for _ in range(2):  # second hit is served from the cache
+                fast.dig(name, qtype)
+            slow.dig(name, qtype)
+        time.sleep(0.3)
+
+        def lines(path, name):
+            out = []
+            for ln in path.read_text().splitlines():
+                try:
+                    d = json.loads(ln)
+                except ValueError:
+                    continue
+                if d.get("msg") == "DNS query" and \
+                        d.get("query", {}).get("name") == name:
+                    out.append(d)
+            return out
+
+        for name, qtype in [("web.foo.com", "A"),
+                            ("svc.foo.com", "A"),
+                            ("_x._tcp.svc.foo.com", "SRV")]:
+            fl = lines(fast_log, name)
+            sl = lines(slow_log, name)
+            assert fl and sl, name
+            f, s = fl[-1], sl[-1]
+            assert f["rcode"] == s["rcode"] == "NOERROR"
+            assert f["query"] == s["query"]
+            assert sorted(f["answers"]) == sorted(s["answers"]), name
+            assert sorted(f["additional"]) == sorted(s["additional"])
+            assert f["edns"] is False
+            for k in ("req_id", "client", "port", "latency", "timers"):
+                assert k in f, k
+            assert set(f["timers"]) == {"parse_us", "resolve_us",
+                                        "encode_us"}
+    finally:
+        fast.stop()
+        slow.stop()
