@@ -77,7 +77,9 @@ def start_backends(n, tmp, zk_port, last_name="h4999.foo.com"):
     for i in range(n):
         b = BinderProcess(dns_domain="foo.com", datacenter="coal",
                           store="zk", zk_host="127.0.0.1", zk_port=zk_port,
-                          workdir=tmp, log_level="warn",
+                          workdir=tmp,
+                          log_level=os.environ.get("BENCH_LOG_LEVEL",
+                                                   "warn"),
                           balancer_socket=str(sockdir / f"b{i}"),
                           log_path=str(tmp / f"binderd-{i}.log"))
         b.start(wait_ready=False)
