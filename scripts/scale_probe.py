@@ -109,7 +109,9 @@ def main():
                        (16, 64, 32)]
         for bal_workers, window, threads in configs:
             restart_balancer(bal_workers)
-            r = blast(150000 * n, window, threads)
+            per_point = int(os.environ.get("PROBE_QUERIES",
+                                            str(150000 * n)))
+            r = blast(per_point, window, threads)
             st = stats()
             qtot = sum(b["queries"] for b in st["backends"])
             dist = sorted(round(b["queries"] / max(qtot, 1), 3)
