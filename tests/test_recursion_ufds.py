@@ -179,3 +179,52 @@ def test_ufds_over_ldaps(tmp_path):
         local.stop()
         upstream.stop()
         ldap.stop()
+
+
+def test_ufds_ldaps_tlsverify_rejects_selfsigned(tmp_path):
+    """recursion.ufds.tlsVerify=true turns on certificate verification
+    (system CA paths): a self-signed UFDS cert must be REJECTED, and
+    binder must keep serving with recursion degraded to best-effort
+    (misses refused) rather than hang or crash."""
+    import subprocess
+    cert = tmp_path / "cert.pem"
+    key = tmp_path / "key.pem"
+    subprocess.run(
+        ["openssl", "req", "-x509", "-newkey", "rsa:2048",
+         "-keyout", str(key), "-out", str(cert), "-days", "2",
+         "-nodes", "-subj", "/CN=ufds.test"],
+        check=True, capture_output=True)
+    ldap = StubLdap(tls_cert=str(cert), tls_key=str(key)).start()
+    ldap.resolvers = [{"datacenter": "dc2", "ip": "127.0.0.2"}]
+
+    local_tree = tmp_path / "local.json"
+    local_tree.write_text(
+        '{"foo.com": null, "web.foo.com": '
+        '{"type": "host", "host": {"address": "10.0.0.1"}}}')
+    local = BinderProcess(
+        dns_domain="foo.com", datacenter="dc1",
+        store=f"file:{local_tree}", workdir=tmp_path,
+        log_path=str(tmp_path / "lv.log"),
+        config={"recursion": {
+            "source": "ufds", "regionName": "r1",
+            "dnsDomain": "foo.com",
+            "ufds": {"url": f"ldaps://127.0.0.1:{ldap.port}",
+                     "bindDN": "cn=root", "bindPassword": "pw",
+                     "tlsVerify": True},
+        }})
+    local.start()
+    try:
+        # give init + at least one retry a chance to run
+        time.sleep(2.0)
+        # no resolver list can have been fetched over the rejected TLS
+        # channel => cross-DC misses refuse
+        r = local.dig("nope.dc2.foo.com", rd=True, timeout=4)
+        assert r.status == "REFUSED"
+        # local answers unaffected, process healthy
+        assert local.dig("web.foo.com").status == "NOERROR"
+        assert local.proc.poll() is None
+        assert not any("region=r1" in s for s in ldap.searches), \
+            "search succeeded despite tlsVerify against self-signed cert"
+    finally:
+        local.stop()
+        ldap.stop()
