@@ -98,7 +98,19 @@ class StubZk:
             self._txnlog_path = os.path.join(txnlog_dir, "log.1")
             fresh = not os.path.exists(self._txnlog_path)
             if not fresh:
-                self._replay(self._txnlog_path)
+                good = self._replay(self._txnlog_path)
+                # a crash mid-append leaves a truncated/corrupt tail;
+                # appending after it would make every later entry
+                # unreachable to the next replay (which stops at the
+                # first bad record, like ZooKeeper). Truncate to the
+                # last fully-verified entry first.
+                if good is None:
+                    # header never landed: start a fresh log
+                    with open(self._txnlog_path, "wb") as tf:
+                        tf.write(struct.pack(">iiq", 0x5A4B4C47, 2, 0))
+                elif good < os.path.getsize(self._txnlog_path):
+                    with open(self._txnlog_path, "r+b") as tf:
+                        tf.truncate(good)
             self._txnlog = open(self._txnlog_path, "ab")
             if fresh:
                 self._txnlog.write(struct.pack(">iiq", 0x5A4B4C47, 2, 0))
@@ -110,14 +122,20 @@ class StubZk:
         is exactly what a real ZK restart + session expiry yields."""
         with open(path, "rb") as f:
             data = f.read()
-        if len(data) < 16 or                 struct.unpack_from(">i", data, 0)[0] != 0x5A4B4C47:
-            return
+        if len(data) < 16 or \
+                struct.unpack_from(">i", data, 0)[0] != 0x5A4B4C47:
+            return None
         off = 16
         while off + 12 <= len(data):
             crc, tlen = struct.unpack_from(">qi", data, off)
             if crc == 0 or tlen <= 0 or off + 12 + tlen + 1 > len(data):
                 break
             txn = data[off + 12:off + 12 + tlen]
+            # adler32 + 0x42 end-of-record marker, like FileTxnLog:
+            # stop at the first corrupt entry
+            if (crc & 0xFFFFFFFF) != (zlib.adler32(txn) & 0xFFFFFFFF) \
+                    or data[off + 12 + tlen] != 0x42:
+                break
             off += 12 + tlen + 1
             (_cid, _cxid, zxid, _t, ttype) = struct.unpack_from(
                 ">qiqqi", txn, 0)
@@ -166,6 +184,7 @@ class StubZk:
                     if pn is not None:
                         pn.children.discard(
                             path_s[path_s.rfind("/") + 1:])
+        return off  # first byte past the last fully-verified entry
 
     # ------------- lifecycle -------------
 
