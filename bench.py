@@ -141,7 +141,7 @@ def main():
                     help="server process count (one per rank/GPU slot)")
     ap.add_argument("--steps", type=int, default=3)
     ap.add_argument("--warmup", type=int, default=1)
-    ap.add_argument("--queries-per-proc", type=int, default=200_000)
+    ap.add_argument("--queries-per-proc", type=int, default=1_000_000)
     ap.add_argument("--window", type=int, default=64)
     ap.add_argument("--tree-records", type=int, default=10_000)
     ap.add_argument("--churn-qps", type=int, default=0,
@@ -181,8 +181,8 @@ def main():
     # operating point: scripts/scale_probe.py + profiles/SCALING.md)
     if ncpu >= 8 * n + 8:
         workers = min(16, max(4, 2 * n))
-        threads = max(8, 4 * n)
-        window = 192
+        threads = max(8, 3 * n)
+        window = 128  # probe-swept: same qps as 192 with ~20x lower p99
         # experiment overrides (profiling/tuning only)
         workers = int(os.environ.get("BENCH_WORKERS", workers))
         threads = int(os.environ.get("BENCH_THREADS", threads))
