@@ -86,9 +86,13 @@ int main(int argc, char** argv) {
     Json cli = Json::object();
     int c;
     int verbosity = 0;
-    while ((c = getopt(argc, argv, "hvVa:b:s:p:f:S:")) != -1) {
+    while ((c = getopt(argc, argv, "hvVa:b:d:s:p:f:S:")) != -1) {
         switch (c) {
         case 'V': printf("%s\n", kVersion); return 0;
+        case 'd': verbosity = atoi(optarg); break;  /* README.md:46-50
+            documents `-d 2` for debug spew; main.js actually
+            implements -v (flag drift noted in SURVEY.md §5.5) —
+            accept both */
         case 'a': cli.set("expiry", Json((int64_t)atoi(optarg))); break;
         case 'b': cli.set("balancerSocket", Json(std::string(optarg))); break;
         case 'f': configFile = optarg; break;

@@ -185,6 +185,28 @@ def test_huge_service_tcp_fits_length_prefix(tmp_path):
         srv.stop()
 
 
+def test_debug_flag_alias(tmp_path):
+    """README.md documents `-d 2` for debug; main.js implements -v
+    (flag drift). binderd accepts both."""
+    import subprocess
+    from binder_amd.harness import BINDERD
+    store = tmp_path / "t.json"
+    store.write_text('{"foo.com": null}')
+    p = subprocess.Popen(
+        [str(BINDERD), "-d", "2", "-S", f"file:{store}", "-p", "0"],
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT)
+    try:
+        import time as _t
+        _t.sleep(0.8)
+        assert p.poll() is None
+    finally:
+        p.terminate()
+        out = p.communicate(timeout=5)[0].decode()
+    # trace level (10) lines prove -d 2 raised verbosity
+    assert '"level":30' in out
+    assert p.returncode is not None
+
+
 def test_vestigial_flags_accepted(tmp_path):
     """-s (cacheSize) and -a (cacheExpiry) are vestigial in the
     reference (main.js:34-38) but must be accepted
