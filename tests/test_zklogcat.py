@@ -165,3 +165,89 @@ def test_multi_txn_decode(tmp_path):
     assert ops[0]["data"] == "616263"  # "abc" hex
     assert ops[1]["type"] == "delete"
     assert ops[1]["path"] == "/m0"
+
+
+import struct
+
+ZKLOGCAT = REPO_ROOT / "bin" / "zklogcat"
+
+
+def _entry(txn):
+    import zlib
+    return (struct.pack(">qi", zlib.adler32(txn) & 0xFFFFFFFF, len(txn))
+            + txn + b"\x42")
+
+
+def _hdr(session, cxid, zxid, t, ttype):
+    return struct.pack(">qiqqi", session, cxid, zxid, t, ttype)
+
+
+def _jstr(b):
+    return struct.pack(">i", len(b)) + b
+
+
+def test_decode_check_setacl_error_session(tmp_path):
+    """Synthetic FileTxnLog covering the txn types the stub server
+    never writes: check(13), setACL(7), error(-1), createSession(-10),
+    closeSession(-11) — pinned against the format zklog.c decodes
+    (/root/reference/src/zklog.c:270-414)."""
+    log = tmp_path / "log.77"
+    body = struct.pack(">iiq", 0x5A4B4C47, 2, 0)
+    sid = 0x11000000000000AA
+    # createSession with 30s timeout
+    body += _entry(_hdr(sid, 1, 100, 1700000000000, -10)
+                   + struct.pack(">i", 30000))
+    # check /a version 3
+    body += _entry(_hdr(sid, 2, 101, 1700000001000, 13)
+                   + _jstr(b"/a") + struct.pack(">i", 3))
+    # setACL /a with one ACL entry (perms 31, world/anyone), version 2
+    body += _entry(_hdr(sid, 3, 102, 1700000002000, 7)
+                   + _jstr(b"/a") + struct.pack(">i", 1)
+                   + struct.pack(">i", 31) + _jstr(b"world")
+                   + _jstr(b"anyone") + struct.pack(">i", 2))
+    # error txn, code -110 (NODEEXISTS)
+    body += _entry(_hdr(sid, 4, 103, 1700000003000, -1)
+                   + struct.pack(">i", -110))
+    # closeSession
+    body += _entry(_hdr(sid, 5, 104, 1700000004000, -11))
+    log.write_bytes(body)
+
+    out = subprocess.run([str(ZKLOGCAT), str(log)],
+                         capture_output=True, text=True, check=True)
+    recs = [json.loads(l) for l in out.stdout.splitlines() if l.strip()]
+    types = [r["type"] for r in recs]
+    assert types == ["createSession", "check", "setACL", "error",
+                     "closeSession"]
+    assert recs[0]["timeout_ms"] == 30000
+    assert recs[1]["path"] == "/a" and recs[1]["version"] == 3
+    assert recs[2]["acl"][0]["scheme"] == "world"
+    assert recs[2]["acl"][0]["id"] == "anyone"
+    assert recs[3]["err"] == -110
+
+    # -S session summary sees the session open+close
+    out = subprocess.run([str(ZKLOGCAT), "-S", str(log)],
+                         capture_output=True, text=True, check=True)
+    assert "11000000000000aa" in out.stdout.lower()
+
+
+def test_garbage_and_truncated_files(tmp_path):
+    bad = tmp_path / "bad.log"
+    bad.write_bytes(b"\x00" * 10)
+    r = subprocess.run([str(ZKLOGCAT), str(bad)],
+                       capture_output=True, text=True)
+    assert r.returncode != 0 or r.stdout.strip() == ""
+
+    # valid magic, then a truncated entry: decode stops cleanly
+    trunc = tmp_path / "trunc.log"
+    txn = _hdr(1, 1, 1, 1700000000000, 2) + _jstr(b"/x")
+    trunc.write_bytes(struct.pack(">iiq", 0x5A4B4C47, 2, 0)
+                      + _entry(txn)
+                      + b"\x00\x00\x00\x09\xaa")
+    r = subprocess.run([str(ZKLOGCAT), str(trunc)],
+                       capture_output=True, text=True, check=True)
+    recs = [json.loads(l) for l in r.stdout.splitlines() if l.strip()]
+    assert len(recs) == 1 and recs[0]["type"] == "delete"
+
+    missing = subprocess.run([str(ZKLOGCAT), str(tmp_path / "nope")],
+                             capture_output=True, text=True)
+    assert missing.returncode != 0
