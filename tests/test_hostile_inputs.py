@@ -210,3 +210,39 @@ def test_balancer_tcp_garbage(balancer):
     r = dig("web.foo.com", "A", server="127.0.0.1",
             port=balancer["port"], timeout=3, tcp=True)
     assert r.status == "NOERROR"
+
+
+def test_balancer_sockdir_junk_and_disappearance(balancer, tmp_path):
+    """Non-socket files in the socket dir are ignored; the dir
+    vanishing (operator error) must not crash the balancer — it keeps
+    serving already-connected backends."""
+    sockdir = balancer["sockdir"]
+    (sockdir / "README").write_text("not a socket")
+    (sockdir / "sub").mkdir()
+    time.sleep(0.6)  # a couple of rescan intervals
+    r = dig("web.foo.com", "A", server="127.0.0.1",
+            port=balancer["port"], timeout=3)
+    assert r.status == "NOERROR"
+    (sockdir / "README").unlink()
+    (sockdir / "sub").rmdir()
+    # dir vanishing entirely: socket presence IS registration, so all
+    # backends read as drained (same as unlink-on-SIGTERM) — the
+    # balancer must survive and resume when the dir comes back
+    import os as _os
+    _os.rename(sockdir, tmp_path / "gone")
+    try:
+        time.sleep(0.6)
+        assert balancer["bal"].poll() is None, "balancer crashed"
+    finally:
+        _os.rename(tmp_path / "gone", sockdir)
+    deadline = time.time() + 10
+    while time.time() < deadline:
+        try:
+            r = dig("web.foo.com", "A", server="127.0.0.1",
+                    port=balancer["port"], timeout=1)
+            if r.status == "NOERROR":
+                break
+        except OSError:
+            pass
+    else:
+        pytest.fail("balancer never resumed after dir returned")
