@@ -180,8 +180,8 @@ def main():
     # must not starve the backends on small boxes (probe-derived
     # operating point: scripts/scale_probe.py + profiles/SCALING.md)
     if ncpu >= 8 * n + 8:
-        workers = min(16, max(4, 2 * n))
-        threads = max(8, 3 * n)
+        workers = min(16, max(8, 2 * n))
+        threads = max(12, 3 * n)
         window = 128  # probe-swept: same qps as 192 with ~20x lower p99
         # experiment overrides (profiling/tuning only)
         workers = int(os.environ.get("BENCH_WORKERS", workers))
