@@ -180,9 +180,12 @@ def main():
     # must not starve the backends on small boxes (probe-derived
     # operating point: scripts/scale_probe.py + profiles/SCALING.md)
     if ncpu >= 8 * n + 8:
-        workers = min(16, max(8, 2 * n))
-        threads = max(12, 3 * n)
-        window = 128  # probe-swept: same qps as 192 with ~20x lower p99
+        # probe-swept operating points (scripts/scale_probe.py;
+        # profiles/SCALING.md): 16 balancer workers from N=2 up,
+        # window 128 (same qps as 192 at ~20x lower p99)
+        workers = min({1: 8}.get(n, 16), max(4, ncpu // 8))
+        threads = {1: 12, 2: 16, 4: 16}.get(n, max(12, 3 * n))
+        window = 128
         # experiment overrides (profiling/tuning only)
         workers = int(os.environ.get("BENCH_WORKERS", workers))
         threads = int(os.environ.get("BENCH_THREADS", threads))
