@@ -1,5 +1,7 @@
 #include "store.hpp"
 
+#include <arpa/inet.h>
+
 #include <algorithm>
 
 namespace bamd {
@@ -117,6 +119,13 @@ CompiledRecord compileRecord(const Json& data) {
     if (recTypeIsHostLike(out.type)) {
         const Json& addr = sub.get("address");
         if (addr.isString()) out.address = addr.asString();
+        /* non-IPv4 address (garbage, or IPv6 in an A-only server):
+         * treat as missing — the reference would abort constructing
+         * the ARecord (deviations ledger #3) */
+        struct in_addr a4h;
+        if (!out.address.empty() &&
+            inet_pton(AF_INET, out.address.c_str(), &a4h) != 1)
+            out.address.clear();
         const Json& portsJ = sub.get("ports");
         if (portsJ.isArray()) {
             for (const auto& p : portsJ.items())
@@ -126,6 +135,14 @@ CompiledRecord compileRecord(const Json& data) {
     } else if (out.type == RecType::Database) {
         const Json& prim = sub.get("primary");
         if (prim.isString()) out.address = urlHostname(prim.asString());
+        /* a primary URL whose host is not an IPv4 literal cannot be
+         * served as an A answer: treat like a missing address (empty
+         * NOERROR — the documented deviation for records the
+         * reference would abort on) rather than emitting 0.0.0.0 */
+        struct in_addr a4;
+        if (!out.address.empty() &&
+            inet_pton(AF_INET, out.address.c_str(), &a4) != 1)
+            out.address.clear();
     } else if (out.type == RecType::Service) {
         /* The service body may be nested one level as service.service
          * (server.js:324-332). */

@@ -405,3 +405,36 @@ def test_non_query_opcode_notimp():
     resp_wire, action = e.query_wire(bytes(wire), 512)
     resp = n.decode_message(resp_wire)
     assert resp["rcode"] == "NOTIMP"
+
+
+def test_database_bad_primary_urls_empty_noerror():
+    """Garbage/non-IPv4 database primaries: the reference would abort
+    constructing ARecord(undefined-or-hostname) (deviations ledger #3);
+    here they answer NOERROR with no records — never 0.0.0.0."""
+    e = mkengine()
+    put(e, "foo.com", None)
+    cases = {
+        "g1.foo.com": {"type": "database",
+                       "database": {"primary": "not a url"}},
+        "g2.foo.com": {"type": "database", "database": {}},
+        "g3.foo.com": {"type": "database",
+                       "database": {"primary": "tcp://u@[::1]:5432/d"}},
+        "g4.foo.com": {"type": "database",
+                       "database": {"primary": "tcp://u@pg.local:5432/d"}},
+    }
+    for name, rec in cases.items():
+        put(e, name, rec)
+        r = e.query(name, "A")
+        assert r["rcode"] == "NOERROR", name
+        assert r["answers"] == [], name
+
+
+def test_host_non_ipv4_address_empty_noerror():
+    e = mkengine()
+    put(e, "foo.com", None)
+    put(e, "v6.foo.com", {"type": "host", "host": {"address": "fd00::1"}})
+    put(e, "junk.foo.com", {"type": "host", "host": {"address": "zzz"}})
+    for name in ("v6.foo.com", "junk.foo.com"):
+        r = e.query(name, "A")
+        assert r["rcode"] == "NOERROR", name
+        assert r["answers"] == [], name
