@@ -111,15 +111,39 @@ class StubZk:
                 elif good < os.path.getsize(self._txnlog_path):
                     with open(self._txnlog_path, "r+b") as tf:
                         tf.truncate(good)
+                self._maybe_compact()
             self._txnlog = open(self._txnlog_path, "ab")
             if fresh:
                 self._txnlog.write(struct.pack(">iiq", 0x5A4B4C47, 2, 0))
                 self._txnlog.flush()
 
+    def _maybe_compact(self):
+        """Rewrite the log as one create per live node when history
+        dominates state (ZooKeeper bounds logs with snapshots +
+        rollover; a single-node dev registry can simply compact on
+        restart — the log stays valid FileTxnLog v2 for zklogcat)."""
+        live = [p for p in self._nodes if p != "/"]
+        if self._replayed_entries <= 2 * len(live) + 64:
+            return
+        tmp = self._txnlog_path + ".compact"
+        now = int(time.time() * 1000)
+        with open(tmp, "wb") as tf:
+            tf.write(struct.pack(">iiq", 0x5A4B4C47, 2, 0))
+            for p in sorted(live, key=lambda x: (x.count("/"), x)):
+                body = (self._jstr(p) + self._jstr(self._nodes[p].data)
+                        + struct.pack(">i", 0) + b"\x00")
+                hdr = struct.pack(">qiqqi", 0, 0, self._zxid, now, 1)
+                txn = hdr + body
+                crc = zlib.adler32(txn) & 0xFFFFFFFF
+                tf.write(struct.pack(">qi", crc, len(txn)) + txn
+                         + b"\x42")
+        os.replace(tmp, self._txnlog_path)
+
     def _replay(self, path: str):
         """Rebuild the tree from a FileTxnLog (durability for zkd).
         Ephemeral creates are skipped: their sessions are gone, which
         is exactly what a real ZK restart + session expiry yields."""
+        self._replayed_entries = 0
         with open(path, "rb") as f:
             data = f.read()
         if len(data) < 16 or \
@@ -137,6 +161,7 @@ class StubZk:
                     or data[off + 12 + tlen] != 0x42:
                 break
             off += 12 + tlen + 1
+            self._replayed_entries += 1
             (_cid, _cxid, zxid, _t, ttype) = struct.unpack_from(
                 ">qiqqi", txn, 0)
             body = txn[32:]  # TxnHeader is 8+4+8+8+4 bytes
