@@ -156,3 +156,96 @@ def test_recursion_over_tcp(dcs):
     r = local.dig("svc.dc2.foo.com", rd=True, tcp=True, timeout=6)
     assert r.status == "NOERROR"
     assert r.answers[0]["address"] == "10.22.0.1"
+
+
+def test_upstream_blackhole_times_out_refused(tmp_path):
+    """Resolvers that never answer: the 3 s upstream deadline
+    (recursion.js:257) must produce REFUSED, not a hang, and the
+    server stays healthy."""
+    import time
+    local_tree = tmp_path / "local.json"
+    local_tree.write_text(json.dumps({
+        "foo.com": None,
+        "web.dc1.foo.com": {"type": "host",
+                            "host": {"address": "10.11.0.1"}}}))
+    local = BinderProcess(
+        dns_domain="foo.com", datacenter="dc1",
+        store=f"file:{local_tree}", workdir=tmp_path,
+        config={"recursion": {
+            "source": "static", "regionName": "r1",
+            "dnsDomain": "foo.com", "upstreamPort": 9,  # discard port
+            "dcs": {"dc2": ["127.0.0.2"]},
+        }})
+    local.start()
+    try:
+        t0 = time.time()
+        r = local.dig("gone.dc2.foo.com", rd=True, timeout=8)
+        elapsed = time.time() - t0
+        assert r.status == "REFUSED"
+        assert elapsed < 6, f"deadline not enforced ({elapsed:.1f}s)"
+        assert local.dig("web.dc1.foo.com").status == "NOERROR"
+    finally:
+        local.stop()
+
+
+def test_upstream_answer_type_filter(tmp_path):
+    """Upstream answers outside A/AAAA/TXT/PTR/CNAME/SRV are dropped
+    (recursion.js:299-323): a response carrying A + SOA in the answer
+    section forwards only the A."""
+    import socket
+    import threading
+    from binder_amd import require_native
+    n = require_native()
+
+    usock = socket.socket(socket.AF_INET, socket.SOCK_DGRAM)
+    usock.bind(("127.0.0.2", 0))
+    uport = usock.getsockname()[1]
+    stop = threading.Event()
+
+    def fake_upstream():
+        usock.settimeout(0.3)
+        while not stop.is_set():
+            try:
+                data, addr = usock.recvfrom(4096)
+            except socket.timeout:
+                continue
+            q = n.decode_message(data)
+            if q is None:
+                continue
+            resp = {
+                "id": q["id"], "qr": True,
+                "questions": q["questions"],
+                "answers": [
+                    {"name": q["questions"][0]["name"], "type": "A",
+                     "ttl": 30, "address": "10.66.0.1"},
+                    {"name": q["questions"][0]["name"], "type": "SOA",
+                     "ttl": 30, "mname": "x.dc2.foo.com",
+                     "rname": "hostmaster.dc2.foo.com", "minimum": 30},
+                ],
+            }
+            usock.sendto(n.encode_message(resp), addr)
+
+    t = threading.Thread(target=fake_upstream, daemon=True)
+    t.start()
+
+    local_tree = tmp_path / "local.json"
+    local_tree.write_text('{"foo.com": null}')
+    local = BinderProcess(
+        dns_domain="foo.com", datacenter="dc1",
+        store=f"file:{local_tree}", workdir=tmp_path,
+        config={"recursion": {
+            "source": "static", "regionName": "r1",
+            "dnsDomain": "foo.com", "upstreamPort": uport,
+            "dcs": {"dc2": ["127.0.0.2"]},
+        }})
+    local.start()
+    try:
+        r = local.dig("mixed.dc2.foo.com", rd=True, timeout=5)
+        assert r.status == "NOERROR"
+        assert [a["type"] for a in r.answers] == ["A"]
+        assert r.answers[0]["address"] == "10.66.0.1"
+    finally:
+        local.stop()
+        stop.set()
+        t.join(timeout=3)
+        usock.close()
