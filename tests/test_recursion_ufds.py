@@ -228,3 +228,68 @@ def test_ufds_ldaps_tlsverify_rejects_selfsigned(tmp_path):
     finally:
         local.stop()
         ldap.stop()
+
+
+def test_ufds_garbage_ldap_server(tmp_path):
+    """A broken LDAP peer (random bytes instead of BER) must leave
+    binder serving with recursion degraded — no crash, no hang."""
+    import socket
+    import threading
+
+    srv = socket.socket()
+    srv.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+    srv.bind(("127.0.0.1", 0))
+    srv.listen(4)
+    port = srv.getsockname()[1]
+    stop = threading.Event()
+
+    def loop():
+        srv.settimeout(0.3)
+        import os as _os
+        while not stop.is_set():
+            try:
+                conn, _ = srv.accept()
+            except socket.timeout:
+                continue
+            except OSError:
+                return
+            try:
+                conn.settimeout(1)
+                try:
+                    conn.recv(4096)
+                except socket.timeout:
+                    pass
+                conn.sendall(_os.urandom(64))
+            except OSError:
+                pass
+            finally:
+                conn.close()
+
+    t = threading.Thread(target=loop, daemon=True)
+    t.start()
+
+    local_tree = tmp_path / "t.json"
+    local_tree.write_text(
+        '{"foo.com": null, "web.foo.com": '
+        '{"type": "host", "host": {"address": "10.0.0.1"}}}')
+    local = BinderProcess(
+        dns_domain="foo.com", datacenter="dc1",
+        store=f"file:{local_tree}", workdir=tmp_path,
+        config={"recursion": {
+            "source": "ufds", "regionName": "r1",
+            "dnsDomain": "foo.com",
+            "ufds": {"url": f"ldap://127.0.0.1:{port}",
+                     "bindDN": "cn=root", "bindPassword": "pw"},
+        }})
+    local.start()
+    try:
+        time.sleep(1.5)
+        r = local.dig("nope.dc2.foo.com", rd=True, timeout=4)
+        assert r.status == "REFUSED"
+        assert local.dig("web.foo.com").status == "NOERROR"
+        assert local.proc.poll() is None
+    finally:
+        local.stop()
+        stop.set()
+        t.join(timeout=3)
+        srv.close()
