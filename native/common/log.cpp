@@ -21,11 +21,23 @@ LogLevel logLevelFromName(const std::string& name, LogLevel dflt) {
 std::string isoTimeNow() {
     struct timespec ts;
     clock_gettime(CLOCK_REALTIME, &ts);
-    struct tm tm;
-    gmtime_r(&ts.tv_sec, &tm);
+    /* gmtime_r + strftime per line is measurable at >1M logged
+     * queries/s; the second-granular prefix only changes once per
+     * second, so cache it per thread. */
+    thread_local time_t cachedSec = 0;
+    thread_local char prefix[32];
+    thread_local size_t prefixLen = 0;
+    if (ts.tv_sec != cachedSec || prefixLen == 0) {
+        struct tm tm;
+        gmtime_r(&ts.tv_sec, &tm);
+        prefixLen = strftime(prefix, sizeof(prefix),
+                             "%Y-%m-%dT%H:%M:%S", &tm);
+        cachedSec = ts.tv_sec;
+    }
     char buf[40];
-    size_t n = strftime(buf, sizeof(buf), "%Y-%m-%dT%H:%M:%S", &tm);
-    snprintf(buf + n, sizeof(buf) - n, ".%03ldZ", ts.tv_nsec / 1000000);
+    memcpy(buf, prefix, prefixLen);
+    snprintf(buf + prefixLen, sizeof(buf) - prefixLen, ".%03ldZ",
+             ts.tv_nsec / 1000000);
     return buf;
 }
 
