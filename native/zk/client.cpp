@@ -213,8 +213,17 @@ void ZkClient::onConnectResponse(const uint8_t* data, size_t len) {
     int32_t timeout = r.i32();
     int64_t sid = r.i64();
     std::string pw = r.str();
-    (void)proto;
-    if (!r.ok || timeout <= 0 || sid == 0) {
+    /* A real server always answers protocolVersion 0 with a sane
+     * negotiated timeout; anything else is a broken/hostile peer, not
+     * a session verdict — reconnect WITHOUT discarding the session so
+     * a later healthy server can still resume it. */
+    if (!r.ok || proto != 0 || timeout > 600000) {
+        log_.warn("malformed ZK connect response; reconnecting");
+        teardown();
+        scheduleReconnect();
+        return;
+    }
+    if (timeout <= 0 || sid == 0) {
         /* session expired (server refuses resume): start a fresh one */
         log_.warn("ZK session expired; creating a new session");
         sessionId_ = 0;
