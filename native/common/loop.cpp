@@ -26,11 +26,7 @@ EventLoop::EventLoop() {
                                  strerror(errno));
     wakeFd_ = eventfd(0, EFD_NONBLOCK | EFD_CLOEXEC);
     if (wakeFd_ >= 0) {
-        struct epoll_event ev {};
-        ev.events = EPOLLIN;
-        ev.data.fd = wakeFd_;
-        epoll_ctl(epfd_, EPOLL_CTL_ADD, wakeFd_, &ev);
-        fds_[wakeFd_] = [this](uint32_t) {
+        addFd(wakeFd_, EPOLLIN, [this](uint32_t) {
             uint64_t v;
             while (read(wakeFd_, &v, sizeof(v)) == sizeof(v)) {
             }
@@ -40,7 +36,7 @@ EventLoop::EventLoop() {
                 tasks.swap(posted_);
             }
             for (auto& t : tasks) t();
-        };
+        });
     }
 }
 
@@ -60,19 +56,22 @@ void EventLoop::postFromThread(TimerCallback cb) {
 }
 
 void EventLoop::addFd(int fd, uint32_t events, FdCallback cb) {
+    uint32_t gen = nextFdGen_++;
     struct epoll_event ev {};
     ev.events = events;
-    ev.data.fd = fd;
+    ev.data.u64 = ((uint64_t)gen << 32) | (uint32_t)fd;
     if (epoll_ctl(epfd_, EPOLL_CTL_ADD, fd, &ev) != 0)
         throw std::runtime_error(std::string("epoll_ctl add: ") +
                                  strerror(errno));
-    fds_[fd] = std::move(cb);
+    fds_[fd] = FdReg{gen, std::move(cb)};
 }
 
 void EventLoop::modFd(int fd, uint32_t events) {
+    auto it = fds_.find(fd);
+    if (it == fds_.end()) return;
     struct epoll_event ev {};
     ev.events = events;
-    ev.data.fd = fd;
+    ev.data.u64 = ((uint64_t)it->second.gen << 32) | (uint32_t)fd;
     if (epoll_ctl(epfd_, EPOLL_CTL_MOD, fd, &ev) != 0)
         throw std::runtime_error(std::string("epoll_ctl mod: ") +
                                  strerror(errno));
@@ -133,11 +132,13 @@ void EventLoop::runOnce(int64_t maxWaitMs) {
                                  strerror(errno));
     }
     for (int i = 0; i < n; ++i) {
-        int fd = evs[i].data.fd;
+        int fd = (int)(uint32_t)evs[i].data.u64;
+        uint32_t gen = (uint32_t)(evs[i].data.u64 >> 32);
         auto it = fds_.find(fd);
         if (it == fds_.end()) continue;  // removed by earlier callback
+        if (it->second.gen != gen) continue;  // fd reused in this batch
         // Copy: callback may delFd itself.
-        FdCallback cb = it->second;
+        FdCallback cb = it->second.cb;
         cb(evs[i].events);
     }
     fireTimers();

@@ -65,9 +65,19 @@ class EventLoop {
         }
     };
 
+    /* Each registration carries a generation; epoll events are tagged
+     * with it (data.u64 = gen<<32 | fd) so a stale queued event for a
+     * closed-and-reused fd within one epoll_wait batch is dropped
+     * instead of being delivered to the new registration (fd-reuse ABA). */
+    struct FdReg {
+        uint32_t gen;
+        FdCallback cb;
+    };
+
     int epfd_;
     std::atomic<bool> running_{false};
-    std::map<int, FdCallback> fds_;
+    std::map<int, FdReg> fds_;
+    uint32_t nextFdGen_ = 1;
     std::priority_queue<Timer, std::vector<Timer>, std::greater<Timer>> heap_;
     std::map<uint64_t, TimerCallback> timers_;  // id -> cb (absent=cancelled)
     uint64_t nextTimerId_ = 1;

@@ -162,6 +162,7 @@ namespace {
 
 struct Encoder {
     std::vector<uint8_t>& buf;
+    bool ok = true;  // false: a name had a >63-byte label (unencodable)
     explicit Encoder(std::vector<uint8_t>& b) : buf(b) { buf.clear(); }
     // Compression table: (suffix view, wire offset). Views reference
     // the record name strings, which outlive the encode; linear scan
@@ -217,9 +218,16 @@ struct Encoder {
             rest = dot == std::string_view::npos
                        ? std::string_view()
                        : rest.substr(dot + 1);
-            size_t n = label.size() > 63 ? 63 : label.size();
-            u8((uint8_t)n);
-            raw(label.data(), n);
+            if (label.size() > 63) {
+                /* DNS labels cap at 63 bytes; an over-long label from
+                 * store/config data must fail the encode rather than be
+                 * silently altered on the wire. */
+                ok = false;
+                u8(0);
+                return;
+            }
+            u8((uint8_t)label.size());
+            raw(label.data(), label.size());
         }
         u8(0);
     }
@@ -294,7 +302,7 @@ uint16_t flagsWord(const Header& h) {
     return f;
 }
 
-void encodeImpl(const Message& m, bool truncated,
+bool encodeImpl(const Message& m, bool truncated,
                 std::vector<uint8_t>& out) {
     Encoder e(out);
     e.buf.reserve(512);
@@ -329,12 +337,28 @@ void encodeImpl(const Message& m, bool truncated,
         for (const auto& r : m.additionals)
             if (r.type == TYPE_OPT) e.encodeRecord(r);
     }
+    return e.ok;
 }
 
 }  // namespace
 
 void Message::encodeInto(std::vector<uint8_t>& out, size_t maxSize) const {
-    encodeImpl(*this, false, out);
+    if (!encodeImpl(*this, false, out)) {
+        /* A record carried a >63-byte label: answer SERVFAIL instead of
+         * emitting a silently altered name. Client-decoded questions
+         * can never hit this (wire labels cap at 63), but re-check and
+         * drop them too if a synthesized question is itself bad. */
+        Message fail;
+        fail.header = header;
+        fail.header.qr = true;
+        fail.header.rcode = RCODE_SERVFAIL;
+        fail.questions = questions;
+        if (!encodeImpl(fail, false, out)) {
+            fail.questions.clear();
+            encodeImpl(fail, false, out);
+        }
+        return;
+    }
     if (maxSize > 0 && out.size() > maxSize)
         encodeImpl(*this, true, out);
 }

@@ -51,7 +51,7 @@ struct QueryResult {
 class Engine {
   public:
     Engine(const EngineConfig& cfg, const Store* store)
-        : cfg_(cfg), store_(store), rng_(0xb1d3) {}
+        : cfg_(cfg), store_(store), rng_(std::random_device{}()) {}
 
     /*
      * Handle one decoded query; fills `resp` (header/question echo done

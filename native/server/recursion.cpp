@@ -23,23 +23,12 @@ Recursion::Recursion(EventLoop* loop, Logger log, RecursionOptions opts,
                      const Store* store)
     : loop_(loop),
       log_(log.child({{"component", Json("Recursion")}})),
-      opts_(std::move(opts)), store_(store) {}
+      opts_(std::move(opts)), store_(store),
+      rng_(std::random_device{}()) {}
 
 Recursion::~Recursion() {
     if (ldapThread_.joinable()) ldapThread_.join();
-    if (fd_ >= 0) {
-        loop_->delFd(fd_);
-        close(fd_);
-    }
     if (refreshTimer_) loop_->cancelTimer(refreshTimer_);
-}
-
-bool Recursion::openSocket() {
-    if (fd_ >= 0) return true;
-    fd_ = socket(AF_INET, SOCK_DGRAM | SOCK_NONBLOCK | SOCK_CLOEXEC, 0);
-    if (fd_ < 0) return false;
-    loop_->addFd(fd_, EPOLLIN, [this](uint32_t) { onSockReadable(); });
-    return true;
 }
 
 void Recursion::emitReady() {
@@ -50,7 +39,6 @@ void Recursion::emitReady() {
 
 void Recursion::init() {
     log_.info("Initing Clients...");
-    openSocket();
     refresh();
 }
 
@@ -368,50 +356,54 @@ void Recursion::resolve(const Message& query, Message& resp,
         for (const auto& m : mine) self = self || m == u;
         if (!self) filtered.push_back(u);
     }
-    if (filtered.empty() || fd_ < 0) {
+    if (filtered.empty()) {
+        refuse();
+        return;
+    }
+    if (activeLookups_ >= 1024) {
+        /* bound per-lookup fd usage: shed load (best effort) */
         refuse();
         return;
     }
 
     auto up = std::make_shared<Upstream>();
-    /* skip qids still in flight (collision would cross-cancel) */
-    uint16_t qid = nextQid_;
-    for (int tries = 0; tries < 8; ++tries) {
-        if (qid == 0) qid++;
-        if (pendingByQid_.count(qid) == 0) break;
-        qid++;
-    }
-    nextQid_ = (uint16_t)(qid + 1);
-    if (pendingByQid_.count(qid) != 0) {
-        /* >64k concurrent recursions: shed load (best effort) */
+    /* Fresh ephemeral-port socket per lookup: the kernel assigns a
+     * random source port, so a forged reply must hit both that port
+     * and the random qid, and pass the source/question checks below. */
+    up->fd = socket(AF_INET, SOCK_DGRAM | SOCK_NONBLOCK | SOCK_CLOEXEC, 0);
+    if (up->fd < 0) {
         refuse();
         return;
     }
-    up->qid = qid;
+    up->qid = (uint16_t)(rng_() & 0xffff);
+    if (up->qid == 0) up->qid = 1;
     up->hosts = std::move(filtered);
     up->maxConcurrency = isPtr ? 100 : 2;  // recursion.js:64-78
     up->resp = &resp;
     up->done = std::move(done);
     up->qname = domain;
-    pendingByQid_[up->qid] = up;
+    up->question = query.questions.empty() ? Question{} : query.questions[0];
 
     /* outgoing query: same question, rd cleared (recursion.js:258-261) */
     Message out;
     out.header.id = up->qid;
     out.header.rd = false;
     out.questions = query.questions;
-    auto wire = out.encode(0);
+    up->wire = out.encode(0);
+
+    activeLookups_++;
+    loop_->addFd(up->fd, EPOLLIN,
+                 [this, up](uint32_t) { onSockReadable(up); });
 
     up->timeoutTimer = loop_->addTimer(kUpstreamTimeoutMs, [this, up]() {
         up->timeoutTimer = 0;
         finish(up, nullptr);
     });
 
-    sendNext(up, wire);
+    sendNext(up);
 }
 
-void Recursion::sendNext(const std::shared_ptr<Upstream>& up,
-                         const std::vector<uint8_t>& wire) {
+void Recursion::sendNext(const std::shared_ptr<Upstream>& up) {
     while (up->inFlight < up->maxConcurrency && !up->hosts.empty()) {
         std::string host = up->hosts.front();
         up->hosts.erase(up->hosts.begin());
@@ -422,27 +414,48 @@ void Recursion::sendNext(const std::shared_ptr<Upstream>& up,
             up->errors++;
             continue;
         }
-        ssize_t rv = sendto(fd_, wire.data(), wire.size(), 0,
+        ssize_t rv = sendto(up->fd, up->wire.data(), up->wire.size(), 0,
                             (struct sockaddr*)&sa, sizeof(sa));
         if (rv < 0) {
             up->errors++;
             continue;
         }
+        up->queried.push_back(sa.sin_addr.s_addr);
         up->inFlight++;
     }
     if (up->inFlight == 0) finish(up, nullptr);
 }
 
-void Recursion::onSockReadable() {
+void Recursion::onSockReadable(const std::shared_ptr<Upstream>& up) {
     uint8_t buf[4096];
-    while (true) {
-        ssize_t nr = recv(fd_, buf, sizeof(buf), 0);
+    while (!up->finished) {
+        struct sockaddr_in src {};
+        socklen_t slen = sizeof(src);
+        ssize_t nr = recvfrom(up->fd, buf, sizeof(buf), 0,
+                              (struct sockaddr*)&src, &slen);
         if (nr <= 0) return;
+        /* Source must be an upstream we actually queried, replying
+         * from the DNS port we sent to. */
+        if (src.sin_family != AF_INET ||
+            src.sin_port != htons(opts_.upstreamPort))
+            continue;
+        bool known = false;
+        for (uint32_t a : up->queried)
+            known = known || a == src.sin_addr.s_addr;
+        if (!known) continue;
         auto msg = Message::decode(buf, (size_t)nr);
         if (!msg || !msg->header.qr) continue;
-        auto it = pendingByQid_.find(msg->header.id);
-        if (it == pendingByQid_.end()) continue;
-        auto up = it->second;
+        if (msg->header.id != up->qid) continue;
+        /* The echoed question must match what we asked. */
+        if (msg->questions.size() != 1) continue;
+        std::string qn = msg->questions[0].name;
+        toLowerAscii(qn);
+        std::string expect = up->question.name;
+        toLowerAscii(expect);
+        if (qn != expect ||
+            msg->questions[0].qtype != up->question.qtype ||
+            msg->questions[0].qclass != up->question.qclass)
+            continue;
         up->inFlight--;
         if (msg->header.rcode == RCODE_NOERROR && !msg->answers.empty()) {
             finish(up, &*msg);
@@ -450,16 +463,8 @@ void Recursion::onSockReadable() {
             up->errors++;
             if (up->inFlight == 0 && up->hosts.empty())
                 finish(up, nullptr);
-            else if (!up->hosts.empty()) {
-                /* keep trying further resolvers */
-                Message out;
-                out.header.id = up->qid;
-                out.header.rd = false;
-                /* reconstruct question from resp (same as original) */
-                out.questions = up->resp->questions;
-                auto wire = out.encode(0);
-                sendNext(up, wire);
-            }
+            else if (!up->hosts.empty())
+                sendNext(up);  /* keep trying further resolvers */
         }
     }
 }
@@ -472,7 +477,12 @@ void Recursion::finish(const std::shared_ptr<Upstream>& up,
         loop_->cancelTimer(up->timeoutTimer);
         up->timeoutTimer = 0;
     }
-    pendingByQid_.erase(up->qid);
+    if (up->fd >= 0) {
+        loop_->delFd(up->fd);
+        close(up->fd);
+        up->fd = -1;
+        activeLookups_--;
+    }
 
     Message& resp = *up->resp;
     size_t accepted = 0;

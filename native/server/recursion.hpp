@@ -28,6 +28,7 @@
 #include <functional>
 #include <map>
 #include <memory>
+#include <random>
 #include <string>
 #include <thread>
 #include <vector>
@@ -72,9 +73,22 @@ class Recursion : public RecursionIface {
     }
 
   private:
+    /*
+     * One in-flight lookup. Each lookup owns a fresh ephemeral-port UDP
+     * socket (kernel-assigned source port), a CSPRNG-seeded random qid,
+     * and the list of upstream addresses actually queried — replies are
+     * only accepted when source address+port, qid AND question section
+     * all match, so an off-path attacker must guess both the 16-bit qid
+     * and the 16-bit ephemeral port (the reference gets the same
+     * properties from mname-client's per-lookup sockets).
+     */
     struct Upstream {
+        int fd = -1;
         uint16_t qid;
         std::vector<std::string> hosts;  // remaining unsent
+        std::vector<uint32_t> queried;   // in_addr.s_addr actually sent to
+        std::vector<uint8_t> wire;       // encoded outgoing query
+        dns::Question question;          // for reply validation
         int inFlight = 0;
         int errors = 0;
         int maxConcurrency;
@@ -89,10 +103,8 @@ class Recursion : public RecursionIface {
     void scheduleRefresh(int64_t ms);
     void emitReady();
     std::vector<std::string> ownAddrs();
-    bool openSocket();
-    void onSockReadable();
-    void sendNext(const std::shared_ptr<Upstream>& up,
-                  const std::vector<uint8_t>& wire);
+    void onSockReadable(const std::shared_ptr<Upstream>& up);
+    void sendNext(const std::shared_ptr<Upstream>& up);
     void finish(const std::shared_ptr<Upstream>& up,
                 const dns::Message* answer);
 
@@ -106,9 +118,8 @@ class Recursion : public RecursionIface {
     std::function<void()> readyCb_;
     uint64_t refreshTimer_ = 0;
 
-    int fd_ = -1;  // shared upstream UDP socket
-    std::map<uint16_t, std::shared_ptr<Upstream>> pendingByQid_;
-    uint16_t nextQid_ = 1;
+    std::mt19937 rng_;   // seeded from std::random_device (qid entropy)
+    int activeLookups_ = 0;  // bounds per-lookup fd usage
 
     std::vector<std::string> nicCache_;
     int64_t nicCacheAtMs_ = 0;

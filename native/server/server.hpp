@@ -117,7 +117,7 @@ class DnsServer {
                   const ClientInfo& ci, std::vector<uint8_t>& out);
     std::string_view srvSvc_, srvProto_;   /* scratch, loop thread only */
     std::vector<uint32_t> shuffleIdx_;
-    std::mt19937 rng_{0xb1d3};
+    std::mt19937 rng_{std::random_device{}()};
     std::string logFields_, logScratch_;   /* afterQuery log buffers */
 
   public:

@@ -165,3 +165,30 @@ def test_txt_longer_than_255_splits():
         {"name": "t.foo.com", "type": "TXT", "ttl": 5, "target": text}]
     m = n.decode_message(n.encode_message(msg))
     assert m["answers"][0]["target"] == text
+
+
+def test_overlong_label_is_encode_error_not_truncation():
+    """A >63-byte label in a record name (bad store/config data) must
+    not be silently truncated on the wire; the encoder answers SERVFAIL
+    with the RR sections dropped instead (codec.cpp)."""
+    bad = "x" * 70 + ".foo.com"
+    msg = q("ok.foo.com", "A")
+    msg["qr"] = True
+    msg["answers"] = [
+        {"name": bad, "type": "A", "ttl": 30, "address": "1.2.3.4"}]
+    m = n.decode_message(n.encode_message(msg))
+    assert m["rcode"] == "SERVFAIL"
+    assert m["answers"] == []
+    # the (valid) question survives so the client can correlate
+    assert m["questions"][0]["name"] == "ok.foo.com"
+
+
+def test_63_byte_label_still_encodes():
+    name = "y" * 63 + ".foo.com"
+    msg = q(name, "A")
+    msg["qr"] = True
+    msg["answers"] = [
+        {"name": name, "type": "A", "ttl": 30, "address": "1.2.3.4"}]
+    m = n.decode_message(n.encode_message(msg))
+    assert m["rcode"] == "NOERROR"
+    assert m["answers"][0]["name"] == name
