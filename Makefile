@@ -83,8 +83,12 @@ $(PYMOD): $(CORE_OBJS) $(BUILD)/native/pybind/module.o
 test: all
 	$(PYTHON) -m pytest tests -q -m "not gpu"
 
+# Lint/style gate (parity with the reference's eslint/jsstyle/cstyle
+# gates, /root/reference/Makefile:17-20): C++ and Python style checks in
+# tools/lint.py; -Werror already covers compiler diagnostics.
 check: all
 	$(PYTHON) -m py_compile binder_amd/*.py bench.py __graft_entry__.py
+	$(PYTHON) tools/lint.py
 	@echo "check OK"
 
 # release tarball layout under /opt/binder-amd (the reference ships

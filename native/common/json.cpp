@@ -52,7 +52,11 @@ struct Parser {
         ++p;  // '{'
         JsonObject obj;
         skipWs();
-        if (!eof() && *p == '}') { ++p; out = Json(std::move(obj)); return true; }
+        if (!eof() && *p == '}') {
+            ++p;
+            out = Json(std::move(obj));
+            return true;
+        }
         while (true) {
             skipWs();
             if (eof() || *p != '"') return false;
@@ -76,7 +80,11 @@ struct Parser {
         ++p;  // '['
         JsonArray arr;
         skipWs();
-        if (!eof() && *p == ']') { ++p; out = Json(std::move(arr)); return true; }
+        if (!eof() && *p == ']') {
+            ++p;
+            out = Json(std::move(arr));
+            return true;
+        }
         while (true) {
             Json v;
             if (!parseValue(v)) return false;

@@ -12,7 +12,7 @@ usage: bench_configs.py [--quick] [--out FILE]
 """
 import argparse
 import json
-import os
+
 import subprocess
 import sys
 import tempfile

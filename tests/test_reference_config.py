@@ -4,7 +4,6 @@ config-agent renders must be accepted, including the full UFDS block
 with its cache/retry/timeout knobs, and an unreachable ldaps UFDS must
 degrade to best-effort (server keeps serving, misses refused)."""
 import json
-import time
 
 from binder_amd.harness import BinderProcess
 from binder_amd.stubzk import StubZk
