@@ -554,6 +554,7 @@ bool DnsServer::fastPath(const uint8_t* data, size_t len, bool udp,
             f += kLogClose;
             log_.logRaw(LogLevel::Info, "DNS query", f);
         }
+        BAMD_PROBE2("op-req-done", out.size(), 0);
         return true;
     }
 
@@ -594,6 +595,7 @@ bool DnsServer::fastPath(const uint8_t* data, size_t len, bool udp,
         f += kLogClose;
         log_.logRaw(LogLevel::Info, "DNS query", f);
     }
+    BAMD_PROBE2("op-req-done", out.size(), 0);
     return true;
 }
 
@@ -601,9 +603,9 @@ bool DnsServer::process(const uint8_t* data, size_t len, bool udp,
                         const ClientInfo& ci, std::vector<uint8_t>& out,
                         std::function<void(std::vector<uint8_t>)>
                             asyncReply) {
+    BAMD_PROBE2("op-req-start", len, (int)udp);
     if (fastPath(data, len, udp, ci, out)) return true;
     int64_t start = nowUs();
-    BAMD_PROBE2(op_req_start, len, (int)udp);
     auto parsed = Message::decode(data, len);
     if (!parsed) return true;  // drop malformed (out empty)
     if (parsed->header.qr) return true;  // ignore responses
@@ -666,7 +668,7 @@ void DnsServer::afterQuery(const Message& query, const Message& resp,
     ++served_;
     int64_t latUs = nowUs() - startUs;
     int64_t lat = latUs / 1000;
-    BAMD_PROBE2(op_req_done, bytesSent, (int)resp.header.rcode);
+    BAMD_PROBE2("op-req-done", bytesSent, (int)resp.header.rcode);
 
     const char* qtype = query.questions.empty()
                             ? nullptr

@@ -58,3 +58,16 @@ def test_seeded_python_violations_fail(tmp_path):
 def test_make_check_runs_lint():
     text = (REPO / "Makefile").read_text()
     assert "tools/lint.py" in text.split("check:")[1].split("\n\n")[0]
+
+
+def test_binderd_has_stapsdt_probes():
+    """op-req-start/op-req-done USDT probes must be compiled into the
+    binary as .note.stapsdt ELF notes (reference: dtrace-provider
+    probes, lib/server.js:25-29)."""
+    r = subprocess.run(["readelf", "-n", str(REPO / "bin" / "binderd")],
+                       capture_output=True, text=True)
+    assert r.returncode == 0
+    assert "stapsdt" in r.stdout
+    assert "Provider: binder" in r.stdout
+    assert "op-req-start" in r.stdout
+    assert "op-req-done" in r.stdout
