@@ -196,17 +196,44 @@ bool Client::connect() {
             return false;
         }
         if (opts_.tlsVerify) {
-            SSL_CTX_set_default_verify_paths(ctx);
+            bool caOk;
+            if (!opts_.caFile.empty())
+                caOk = SSL_CTX_load_verify_locations(
+                           ctx, opts_.caFile.c_str(), nullptr) == 1;
+            else
+                caOk = SSL_CTX_set_default_verify_paths(ctx) == 1;
+            if (!caOk) {
+                err_ = "cannot load CA certificates";
+                SSL_CTX_free(ctx);
+                close();
+                return false;
+            }
             SSL_CTX_set_verify(ctx, SSL_VERIFY_PEER, nullptr);
         } else {
             SSL_CTX_set_verify(ctx, SSL_VERIFY_NONE, nullptr);
         }
         sslCtx_ = ctx;
         SSL* ssl = SSL_new(ctx);
+        if (opts_.tlsVerify) {
+            /* Bind the verified chain to the peer we dialed: IP SAN
+             * check when host is an address, DNS SAN/CN otherwise. */
+            X509_VERIFY_PARAM* vp = SSL_get0_param(ssl);
+            struct in_addr ia;
+            if (inet_pton(AF_INET, opts_.host.c_str(), &ia) == 1)
+                X509_VERIFY_PARAM_set1_ip_asc(vp, opts_.host.c_str());
+            else
+                X509_VERIFY_PARAM_set1_host(vp, opts_.host.c_str(),
+                                            opts_.host.size());
+        }
         SSL_set_fd(ssl, fd_);
         ssl_ = ssl;
         if (SSL_connect(ssl) != 1) {
+            long vr = SSL_get_verify_result(ssl);
             err_ = "TLS handshake failed";
+            if (vr != X509_V_OK) {
+                err_ += ": ";
+                err_ += X509_verify_cert_error_string(vr);
+            }
             close();
             return false;
         }

@@ -25,7 +25,11 @@ struct Options {
     std::string host;
     uint16_t port = 389;
     bool tls = false;           // ldaps://
-    bool tlsVerify = false;     // verify peer cert (system CA paths)
+    /* Certificate verification is ON by default (the reference trusts
+     * the ldapjs/ufds stack's default posture; an internal-CA deploy
+     * sets caFile, and tlsVerify=false is an explicit opt-out). */
+    bool tlsVerify = true;
+    std::string caFile;         // PEM bundle; empty = system CA paths
     std::string bindDn;         // empty = anonymous
     std::string bindPassword;
     int timeoutMs = 10000;

@@ -197,7 +197,8 @@ void Recursion::refreshViaUfds() {
     lopts.host = addr;
     lopts.port = port;
     lopts.tls = tls;
-    lopts.tlsVerify = ucfg.get("tlsVerify").asBool(false);
+    lopts.tlsVerify = ucfg.get("tlsVerify").asBool(true);
+    lopts.caFile = ucfg.get("caFile").asString();
     lopts.bindDn = ucfg.get("bindDN").asString();
     lopts.bindPassword = ucfg.get("bindPassword").asString();
     std::string region = opts_.regionName;

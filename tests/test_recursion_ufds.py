@@ -123,9 +123,9 @@ def test_ufds_bad_credentials_best_effort(tmp_path):
 
 
 def test_ufds_over_ldaps(tmp_path):
-    """ldaps:// UFDS with a self-signed cert (tlsVerify off, like the
-    reference's internal-CA deployments): the native TLS client must
-    bind+search successfully."""
+    """ldaps:// UFDS with a self-signed cert and the explicit
+    tlsVerify=false opt-out (verification is ON by default now): the
+    native TLS client must bind+search successfully."""
     import subprocess
     cert = tmp_path / "cert.pem"
     key = tmp_path / "key.pem"
@@ -157,7 +157,8 @@ def test_ufds_over_ldaps(tmp_path):
             "source": "ufds", "regionName": "r1",
             "dnsDomain": "foo.com", "upstreamPort": upstream.port,
             "ufds": {"url": f"ldaps://127.0.0.1:{ldap.port}",
-                     "bindDN": "cn=root", "bindPassword": "pw"},
+                     "bindDN": "cn=root", "bindPassword": "pw",
+                     "tlsVerify": False},
         }})
     local.start()
     try:
@@ -182,10 +183,10 @@ def test_ufds_over_ldaps(tmp_path):
 
 
 def test_ufds_ldaps_tlsverify_rejects_selfsigned(tmp_path):
-    """recursion.ufds.tlsVerify=true turns on certificate verification
-    (system CA paths): a self-signed UFDS cert must be REJECTED, and
-    binder must keep serving with recursion degraded to best-effort
-    (misses refused) rather than hang or crash."""
+    """Certificate verification is the DEFAULT (no tlsVerify key in the
+    config): a self-signed UFDS cert must be REJECTED, and binder must
+    keep serving with recursion degraded to best-effort (misses
+    refused) rather than hang or crash."""
     import subprocess
     cert = tmp_path / "cert.pem"
     key = tmp_path / "key.pem"
@@ -209,8 +210,7 @@ def test_ufds_ldaps_tlsverify_rejects_selfsigned(tmp_path):
             "source": "ufds", "regionName": "r1",
             "dnsDomain": "foo.com",
             "ufds": {"url": f"ldaps://127.0.0.1:{ldap.port}",
-                     "bindDN": "cn=root", "bindPassword": "pw",
-                     "tlsVerify": True},
+                     "bindDN": "cn=root", "bindPassword": "pw"},
         }})
     local.start()
     try:
@@ -293,3 +293,85 @@ def test_ufds_garbage_ldap_server(tmp_path):
         stop.set()
         t.join(timeout=3)
         srv.close()
+
+
+def test_ufds_ldaps_with_cafile(tmp_path):
+    """A private-CA deployment pins trust with recursion.ufds.caFile:
+    verification stays on, the presented cert chains to the configured
+    CA, and the SAN must cover the dialed address (IP SAN here)."""
+    import subprocess
+    cert = tmp_path / "ca-cert.pem"
+    key = tmp_path / "ca-key.pem"
+    subprocess.run(
+        ["openssl", "req", "-x509", "-newkey", "rsa:2048",
+         "-keyout", str(key), "-out", str(cert), "-days", "2",
+         "-nodes", "-subj", "/CN=ufds.test",
+         "-addext", "subjectAltName=IP:127.0.0.1"],
+        check=True, capture_output=True)
+    ldap = StubLdap(tls_cert=str(cert), tls_key=str(key)).start()
+    ldap.resolvers = [{"datacenter": "dc2", "ip": "127.0.0.2"}]
+
+    local_tree = tmp_path / "local.json"
+    local_tree.write_text('{"foo.com": null}')
+    local = BinderProcess(
+        dns_domain="foo.com", datacenter="dc1",
+        store=f"file:{local_tree}", workdir=tmp_path,
+        log_path=str(tmp_path / "lc.log"),
+        config={"recursion": {
+            "source": "ufds", "regionName": "r1",
+            "dnsDomain": "foo.com",
+            "ufds": {"url": f"ldaps://127.0.0.1:{ldap.port}",
+                     "bindDN": "cn=root", "bindPassword": "pw",
+                     "caFile": str(cert)},
+        }})
+    local.start()
+    try:
+        deadline = time.time() + 20
+        while time.time() < deadline:
+            if any("region=r1" in s for s in ldap.searches):
+                break
+            time.sleep(0.25)
+        assert any("region=r1" in s for s in ldap.searches), \
+            "verified LDAPS search never happened; " + \
+            open(str(tmp_path / "lc.log")).read()[-1500:]
+    finally:
+        local.stop()
+        ldap.stop()
+
+
+def test_ufds_ldaps_cafile_rejects_wrong_host(tmp_path):
+    """Even with the right CA, a cert whose SAN does not cover the
+    dialed address must be rejected (hostname/IP binding)."""
+    import subprocess
+    cert = tmp_path / "other-cert.pem"
+    key = tmp_path / "other-key.pem"
+    subprocess.run(
+        ["openssl", "req", "-x509", "-newkey", "rsa:2048",
+         "-keyout", str(key), "-out", str(cert), "-days", "2",
+         "-nodes", "-subj", "/CN=ufds.test",
+         "-addext", "subjectAltName=IP:10.99.99.99"],
+        check=True, capture_output=True)
+    ldap = StubLdap(tls_cert=str(cert), tls_key=str(key)).start()
+    ldap.resolvers = [{"datacenter": "dc2", "ip": "127.0.0.2"}]
+
+    local_tree = tmp_path / "local.json"
+    local_tree.write_text('{"foo.com": null}')
+    local = BinderProcess(
+        dns_domain="foo.com", datacenter="dc1",
+        store=f"file:{local_tree}", workdir=tmp_path,
+        config={"recursion": {
+            "source": "ufds", "regionName": "r1",
+            "dnsDomain": "foo.com",
+            "ufds": {"url": f"ldaps://127.0.0.1:{ldap.port}",
+                     "bindDN": "cn=root", "bindPassword": "pw",
+                     "caFile": str(cert)},
+        }})
+    local.start()
+    try:
+        time.sleep(2.0)
+        assert not any("region=r1" in s for s in ldap.searches), \
+            "search succeeded despite SAN/host mismatch"
+        assert local.proc.poll() is None
+    finally:
+        local.stop()
+        ldap.stop()
