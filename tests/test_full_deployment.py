@@ -168,3 +168,75 @@ def test_full_stack(tmp_path):
             except subprocess.TimeoutExpired:
                 p.kill()
         zk.stop()
+
+
+@pytest.mark.timeout(120)
+def test_sdc_single_process_flavor(tmp_path):
+    """The Triton/SDC deployment flavor (reference boot/setup.sh:146-178):
+    deploy/setup.sh FLAVOR=sdc boots ONE binderd on $PORT, writes the
+    single metric port (port+1000, the `mdata-put metricPorts 1053`
+    equivalent), and self-registers in the registry as an rr_host +
+    _dns._udp SRV service — then answers for its own domain."""
+    import urllib.request
+
+    zk = StubZk().start()
+    statedir = tmp_path / "state"
+    statedir.mkdir()
+    port = free_port()
+    cfg = tmp_path / "binder.json"
+    cfg.write_text(json.dumps({
+        "dnsDomain": "foo.com", "datacenterName": "coal",
+        "host": "127.0.0.1", "store": "zk",
+        "metricsPort": free_port()}))
+    pid = None
+    try:
+        zk.mkdirp("/com/foo")
+        env = dict(os.environ,
+                   FLAVOR="sdc", PORT=str(port),
+                   STATEDIR=str(statedir), PREFIX=str(REPO_ROOT),
+                   CONFIG=str(cfg), REGISTER_ADDR="127.0.0.1",
+                   ZK_HOST="127.0.0.1", ZK_PORT=str(zk.port),
+                   LOG_LEVEL="info")
+        rc = subprocess.run(
+            ["sh", str(REPO_ROOT / "deploy" / "setup.sh")],
+            env=env, cwd=str(REPO_ROOT), capture_output=True, text=True,
+            timeout=60)
+        assert rc.returncode == 0, rc.stdout + rc.stderr
+
+        # single metric port written for cmon-agent discovery
+        mp = (statedir / "metric_ports").read_text().strip()
+        assert mp == str(port + 1000)
+        pid = int((statedir / "binderd.pid").read_text().strip())
+
+        # self-registration: binder answers A + SRV for its own domain
+        deadline = time.time() + 20
+        r = None
+        while time.time() < deadline:
+            try:
+                r = dig("binder.coal.foo.com", "A", server="127.0.0.1",
+                        port=port, timeout=0.5)
+                if r.status == "NOERROR":
+                    break
+            except (socket.timeout, OSError):
+                pass
+            time.sleep(0.2)
+        assert r is not None and r.status == "NOERROR", \
+            (statedir / "log" / "binder.log").read_text()[-2000:]
+        assert r.answers[0]["address"] == "127.0.0.1"
+        rs = dig("_dns._udp.binder.coal.foo.com", "SRV",
+                 server="127.0.0.1", port=port, timeout=2)
+        assert rs.status == "NOERROR"
+        assert any(a.get("port") == port for a in rs.answers)
+
+        # metrics exposition is up on the configured port
+        mport = json.loads(cfg.read_text())["metricsPort"]
+        with urllib.request.urlopen(
+                f"http://127.0.0.1:{mport}/metrics", timeout=3) as resp:
+            assert b"binder_requests_completed" in resp.read()
+    finally:
+        if pid is not None:
+            try:
+                os.kill(pid, 15)
+            except ProcessLookupError:
+                pass
+        zk.stop()
