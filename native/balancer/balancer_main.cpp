@@ -31,6 +31,7 @@
 #include <sys/un.h>
 #include <unistd.h>
 
+#include <cstdlib>
 #include <cstring>
 #include <map>
 #include <unordered_map>
@@ -113,7 +114,9 @@ struct PendingSlot {
 };
 static_assert(sizeof(PendingSlot) <= 32, "keep the ring compact");
 
-static constexpr size_t kPendingSlots = 32768;  /* > kMaxPending */
+/* Ring size > kMaxPending by default; -q overrides (power of two;
+ * tiny values let tests exercise the overwrite accounting). */
+size_t g_pendingSlots = 32768;
 
 struct Backend {
     int id;
@@ -129,19 +132,25 @@ struct Backend {
     int64_t pingSentAt = 0;
     uint64_t queries = 0;
     uint64_t replies = 0;
+    /* live pending-ring slots lost to an overwrite (ring collision
+     * with >kPendingSlots in flight): the reply for the overwritten
+     * request will be silently dropped, so overloads must be visible
+     * on the stats socket, not silent */
+    uint64_t overwrites = 0;
+    size_t outOff = 0;  /* consumed prefix of `out` (flush cursor) */
     size_t remotes = 0;
 
     PendingSlot* slotFor(uint32_t reqId) {
         if (!pending) {
-            pending = std::make_unique<PendingSlot[]>(kPendingSlots);
-            for (size_t i = 0; i < kPendingSlots; ++i)
+            pending = std::make_unique<PendingSlot[]>(g_pendingSlots);
+            for (size_t i = 0; i < g_pendingSlots; ++i)
                 pending[i].reqId = 0;
         }
-        return &pending[reqId & (kPendingSlots - 1)];
+        return &pending[reqId & (g_pendingSlots - 1)];
     }
     void clearPending() {
         if (pending)
-            for (size_t i = 0; i < kPendingSlots; ++i)
+            for (size_t i = 0; i < g_pendingSlots; ++i)
                 pending[i].reqId = 0;
         pendingCount = 0;
     }
@@ -394,6 +403,7 @@ void Balancer::backendDown(Backend* be) {
     be->ok = false;
     be->in.clear();
     be->out.clear();
+    be->outOff = 0;
     be->writeBlocked = false;
     be->clearPending();
     /* unpin remotes so they re-pick a healthy backend */
@@ -414,13 +424,21 @@ void Balancer::backendDown(Backend* be) {
 }
 
 void Balancer::backendFlush(Backend* be) {
-    while (!be->out.empty() && be->fd >= 0) {
-        ssize_t nw = write(be->fd, be->out.data(), be->out.size());
+    while (be->outOff < be->out.size() && be->fd >= 0) {
+        ssize_t nw = write(be->fd, be->out.data() + be->outOff,
+                           be->out.size() - be->outOff);
         if (nw > 0) {
-            be->out.erase(0, (size_t)nw);
+            be->outOff += (size_t)nw;
             continue;
         }
         if (nw < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) {
+            /* compact occasionally so a long-blocked backend does not
+             * keep a consumed prefix resident (amortized O(1) vs the
+             * O(n) memmove-per-write of erase(0, nw)) */
+            if (be->outOff > (1u << 20)) {
+                be->out.erase(0, be->outOff);
+                be->outOff = 0;
+            }
             if (!be->writeBlocked) {
                 be->writeBlocked = true;
                 loop_->modFd(be->fd, EPOLLIN | EPOLLOUT);
@@ -429,6 +447,10 @@ void Balancer::backendFlush(Backend* be) {
         }
         backendDown(be);
         return;
+    }
+    if (be->outOff >= be->out.size()) {
+        be->out.clear();
+        be->outOff = 0;
     }
     if (be->writeBlocked && be->fd >= 0) {
         be->writeBlocked = false;
@@ -683,13 +705,14 @@ void Balancer::onUdpReadable() {
              * past any client deadline — drop now (clients retry)
              * rather than queue into a multi-ms tail */
             if (be->pendingCount > kMaxPending ||
-                be->out.size() > (4u << 20)) {
+                be->out.size() - be->outOff > (4u << 20)) {
                 drops_++;
                 continue;
             }
             uint32_t reqId = be->nextReq++;
             PendingSlot* pr = be->slotFor(reqId);
             if (pr->reqId == 0) be->pendingCount++;
+            else be->overwrites++;
             pr->reqId = reqId;
             pr->expiresAtMs = (uint32_t)expiry;
             pr->tcp = false;
@@ -838,6 +861,7 @@ void Balancer::onTcpClient(std::shared_ptr<TcpClient> c, uint32_t ev) {
             uint32_t reqId = be->nextReq++;
             PendingSlot* pr = be->slotFor(reqId);
             if (pr->reqId == 0) be->pendingCount++;
+            else be->overwrites++;
             pr->reqId = reqId;
             pr->tcp = true;
             pr->tcpFd = c->fd;
@@ -870,6 +894,7 @@ Json Balancer::snapshot() const {
         b.set("queries", Json((int64_t)be->queries));
         b.set("replies", Json((int64_t)be->replies));
         b.set("pending", Json((int64_t)be->pendingCount));
+        b.set("overwrites", Json((int64_t)be->overwrites));
         bes.push_back(std::move(b));
     }
     out.set("backends", Json(std::move(bes)));
@@ -925,6 +950,9 @@ void Balancer::onStatsAccept() {
                                           b.get("replies").asInt()));
                     m.set("pending", Json(m.get("pending").asInt() +
                                           b.get("pending").asInt()));
+                    m.set("overwrites",
+                          Json(m.get("overwrites").asInt() +
+                               b.get("overwrites").asInt()));
                     m.set("ok", Json(m.get("ok").asBool() ||
                                      b.get("ok").asBool()));
                 }
@@ -965,7 +993,7 @@ void Balancer::sweep() {
         /* expire stale pendings (uint32 wrap-safe comparison) */
         if (be->pending && be->pendingCount > 0) {
             uint32_t now32 = (uint32_t)now;
-            for (size_t i = 0; i < kPendingSlots; ++i) {
+            for (size_t i = 0; i < g_pendingSlots; ++i) {
                 PendingSlot& s = be->pending[i];
                 if (s.reqId != 0 &&
                     (int32_t)(now32 - s.expiresAtMs) > 0) {
@@ -1005,8 +1033,15 @@ int main(int argc, char** argv) {
     int rescanMs = 1000;
     int workers = 1;
     int c;
-    while ((c = getopt(argc, argv, "hp:H:s:S:r:w:")) != -1) {
+    while ((c = getopt(argc, argv, "hp:H:s:S:r:w:q:")) != -1) {
         switch (c) {
+        case 'q': {
+            size_t v = (size_t)strtoull(optarg, nullptr, 10);
+            size_t p2 = 1;
+            while (p2 < v) p2 <<= 1;
+            g_pendingSlots = p2 < 8 ? 8 : p2;
+            break;
+        }
         case 'p': port = (uint16_t)atoi(optarg); break;
         case 'H': host = optarg; break;
         case 's': dir = optarg; break;
@@ -1018,7 +1053,7 @@ int main(int argc, char** argv) {
             fprintf(stderr,
                     "usage: binder-balancer [-p port] [-H host] "
                     "[-s socket-dir] [-S stats-socket] [-r rescan-ms] "
-                    "[-w workers]\n");
+                    "[-w workers] [-q ring-slots]\n");
             return c == 'h' ? 0 : 1;
         }
     }

@@ -2,14 +2,26 @@
  * dnsblast: UDP DNS load generator for the benchmark suite
  * (BASELINE.md's qps + p50/p99 measurement).
  *
- * Each worker thread keeps a sliding window of in-flight queries on its
- * own socket (sendmmsg/recvmmsg batches), cycling through a name list
- * loaded from a file. Query wire images are prebuilt; per-send we patch
- * only the DNS id. Latencies land in log-spaced microsecond buckets for
+ * Each worker thread keeps a sliding window of in-flight queries
+ * spread over K sockets (-P), cycling through a name list loaded from
+ * a file. Query wire images are prebuilt; per-send we patch only the
+ * DNS id. Latencies land in log-spaced microsecond buckets for
  * p50/p99 extraction. Output: ONE JSON line on stdout.
  *
+ * Multiple sockets per thread matter against a SO_REUSEPORT-sharded
+ * balancer: the kernel hashes FLOWS (4-tuples) across shards, so the
+ * flow count must comfortably exceed the shard count or shard load is
+ * balls-in-bins uneven and the busiest shard sets the ceiling.
+ *
+ * -r <qps> switches from closed-loop saturation to a fixed offered
+ * rate (token-paced per thread): the measurement protocol is then
+ * "qps at SLO" — the caller binary-searches the max rate whose p99
+ * stays under its SLO, which is far less noisy step-to-step than
+ * running at the chaotic saturation equilibrium.
+ *
  * usage: dnsblast -s <server-ip> -p <port> -n <queries> [-c window]
- *        [-t threads] [-f names-file] [-B bind-ip-base] [-T timeout-ms]
+ *        [-t threads] [-P socks/thread] [-r offered-qps]
+ *        [-f names-file] [-B bind-ip-base] [-T timeout-ms] [-R]
  *   names-file: lines of "<name> <qtype>"; default a single test name.
  *   -B 127.0.0.x base: thread i binds source ip base+i (gives the
  *      balancer distinct remotes so per-IP affinity spreads load).
@@ -86,6 +98,8 @@ struct Config {
     uint64_t queries = 100000;
     int window = 64;
     int threads = 1;
+    int socksPerThread = 1;
+    double rateQps = 0;  // total offered rate; 0 = closed loop
     std::string namesFile;
     std::string bindBase;
     int timeoutMs = 2000;
@@ -95,30 +109,37 @@ struct Config {
 void worker(const Config& cfg, int tid,
             const std::vector<std::vector<uint8_t>>& wires,
             ThreadResult* out, std::atomic<bool>* abort) {
-    int fd = socket(AF_INET, SOCK_DGRAM, 0);
-    if (fd < 0) return;
-    int sz = 4 << 20;
-    setsockopt(fd, SOL_SOCKET, SO_RCVBUF, &sz, sizeof(sz));
-    setsockopt(fd, SOL_SOCKET, SO_SNDBUF, &sz, sizeof(sz));
-    if (!cfg.bindBase.empty()) {
-        /* bind distinct loopback source ip per thread: base + tid */
-        struct in_addr base;
-        inet_pton(AF_INET, cfg.bindBase.c_str(), &base);
-        uint32_t ip = ntohl(base.s_addr) + (uint32_t)tid;
-        struct sockaddr_in src {};
-        src.sin_family = AF_INET;
-        src.sin_addr.s_addr = htonl(ip);
-        src.sin_port = 0;
-        bind(fd, (struct sockaddr*)&src, sizeof(src));
+    const int K = cfg.socksPerThread < 1 ? 1 : cfg.socksPerThread;
+    std::vector<int> fds;
+    for (int k = 0; k < K; ++k) {
+        int fd = socket(AF_INET, SOCK_DGRAM, 0);
+        if (fd < 0) break;
+        int sz = 4 << 20;
+        setsockopt(fd, SOL_SOCKET, SO_RCVBUF, &sz, sizeof(sz));
+        setsockopt(fd, SOL_SOCKET, SO_SNDBUF, &sz, sizeof(sz));
+        if (!cfg.bindBase.empty()) {
+            /* bind distinct loopback source ip per thread: base + tid */
+            struct in_addr base;
+            inet_pton(AF_INET, cfg.bindBase.c_str(), &base);
+            uint32_t ip = ntohl(base.s_addr) + (uint32_t)tid;
+            struct sockaddr_in src {};
+            src.sin_family = AF_INET;
+            src.sin_addr.s_addr = htonl(ip);
+            src.sin_port = 0;  // distinct ephemeral port per socket
+            bind(fd, (struct sockaddr*)&src, sizeof(src));
+        }
+        struct sockaddr_in dst {};
+        dst.sin_family = AF_INET;
+        dst.sin_port = htons(cfg.port);
+        inet_pton(AF_INET, cfg.server.c_str(), &dst.sin_addr);
+        if (connect(fd, (struct sockaddr*)&dst, sizeof(dst)) != 0) {
+            close(fd);
+            break;
+        }
+        fds.push_back(fd);
     }
-    struct sockaddr_in dst {};
-    dst.sin_family = AF_INET;
-    dst.sin_port = htons(cfg.port);
-    inet_pton(AF_INET, cfg.server.c_str(), &dst.sin_addr);
-    if (connect(fd, (struct sockaddr*)&dst, sizeof(dst)) != 0) {
-        close(fd);
-        return;
-    }
+    if (fds.empty()) return;
+    const int nSock = (int)fds.size();
 
     const int W = cfg.window;
     std::vector<int64_t> sentAt(W, 0);       // 0 = slot idle
@@ -129,6 +150,11 @@ void worker(const Config& cfg, int tid,
     uint64_t target = cfg.queries;
     uint64_t launched = 0, completed = 0;
     const int64_t timeoutUs = (int64_t)cfg.timeoutMs * 1000;
+    /* fixed-rate pacing: this thread's share of the offered rate */
+    const double rate = cfg.rateQps > 0
+                            ? cfg.rateQps / (double)cfg.threads
+                            : 0;
+    const int64_t tStart = nowUs();
 
     /* batched RX (recvmmsg) and TX (sendmmsg): syscall count, not
      * packet handling, bounds the generator at high QPS */
@@ -142,45 +168,70 @@ void worker(const Config& cfg, int tid,
     std::vector<int> freeSlots;
     freeSlots.reserve(W);
 
+    /* How many more queries may launch right now (rate pacing). */
+    auto allowance = [&]() -> uint64_t {
+        if (launched >= target) return 0;
+        uint64_t left = target - launched;
+        if (rate <= 0) return left;
+        uint64_t paced = (uint64_t)((double)(nowUs() - tStart) * rate /
+                                    1e6);
+        if (paced <= launched) return 0;
+        uint64_t a = paced - launched;
+        return a < left ? a : left;
+    };
+
+    /* Send queries for free slots, grouped per socket (slot % nSock)
+     * so each socket's burst goes out in one sendmmsg. */
     auto batchSend = [&](std::vector<int>& slots) {
-        int nTx = 0;
-        for (int slot : slots) {
-            if (launched >= target) break;
-            size_t ni = rng() % wires.size();
-            const auto& w = wires[ni];
-            slotSeq[slot]++;
-            uint16_t qid = (uint16_t)((slot & 0xFF) |
-                                      ((slotSeq[slot] & 0xFF) << 8));
-            memcpy(txBufs[nTx].data(), w.data(), w.size());
-            txBufs[nTx][0] = (uint8_t)(qid >> 8);
-            txBufs[nTx][1] = (uint8_t)qid;
-            txIovs[nTx] = {txBufs[nTx].data(), w.size()};
-            memset(&txHdrs[nTx], 0, sizeof(txHdrs[nTx]));
-            txHdrs[nTx].msg_hdr.msg_iov = &txIovs[nTx];
-            txHdrs[nTx].msg_hdr.msg_iovlen = 1;
-            sentAt[slot] = nowUs();
-            slotName[slot] = ni;
-            out->sent++;
-            launched++;
-            nTx++;
-            if (nTx == kRxBatch) {
+        uint64_t budget = allowance();
+        size_t used = 0;
+        for (int k = 0; k < nSock && budget > 0; ++k) {
+            int nTx = 0;
+            auto flush = [&]() {
                 int done = 0;
                 while (done < nTx) {
-                    int rv = sendmmsg(fd, txHdrs.data() + done,
+                    int rv = sendmmsg(fds[k], txHdrs.data() + done,
                                       nTx - done, 0);
                     if (rv <= 0) break;
                     done += rv;
                 }
                 nTx = 0;
+            };
+            for (size_t si = 0; si < slots.size() && budget > 0; ++si) {
+                int slot = slots[si];
+                if (slot < 0 || slot % nSock != k) continue;
+                size_t ni = rng() % wires.size();
+                const auto& w = wires[ni];
+                slotSeq[slot]++;
+                uint16_t qid = (uint16_t)((slot & 0xFF) |
+                                          ((slotSeq[slot] & 0xFF) << 8));
+                memcpy(txBufs[nTx].data(), w.data(), w.size());
+                txBufs[nTx][0] = (uint8_t)(qid >> 8);
+                txBufs[nTx][1] = (uint8_t)qid;
+                txIovs[nTx] = {txBufs[nTx].data(), w.size()};
+                memset(&txHdrs[nTx], 0, sizeof(txHdrs[nTx]));
+                txHdrs[nTx].msg_hdr.msg_iov = &txIovs[nTx];
+                txHdrs[nTx].msg_hdr.msg_iovlen = 1;
+                sentAt[slot] = nowUs();
+                slotName[slot] = ni;
+                out->sent++;
+                launched++;
+                budget--;
+                used++;
+                slots[si] = -1;  // consumed
+                if (++nTx == kRxBatch) flush();
             }
+            flush();
         }
-        int done = 0;
-        while (done < nTx) {
-            int rv = sendmmsg(fd, txHdrs.data() + done, nTx - done, 0);
-            if (rv <= 0) break;
-            done += rv;
+        if (used == slots.size() || budget == 0) {
+            /* compact: drop consumed entries */
+            size_t w = 0;
+            for (size_t i = 0; i < slots.size(); ++i)
+                if (slots[i] >= 0) slots[w++] = slots[i];
+            slots.resize(w);
+        } else {
+            slots.clear();
         }
-        slots.clear();
     };
 
     /* prime the window */
@@ -188,50 +239,57 @@ void worker(const Config& cfg, int tid,
         freeSlots.push_back(s);
     batchSend(freeSlots);
 
-    struct pollfd pfd {fd, POLLIN, 0};
+    std::vector<struct pollfd> pfds(nSock);
+    for (int k = 0; k < nSock; ++k) pfds[k] = {fds[k], POLLIN, 0};
     int64_t lastSweep = nowUs();
     while (completed < target && !abort->load()) {
-        int rv = poll(&pfd, 1, 50);
+        /* paced mode: short poll so the token bucket is serviced even
+         * while the window has room */
+        int pollMs = rate > 0 ? 2 : 50;
+        int rv = poll(pfds.data(), (nfds_t)nSock, pollMs);
         if (rv > 0) {
-            while (true) {
-                for (int i = 0; i < kRxBatch; ++i) {
-                    rxIovs[i] = {rxBufs[i].data(), rxBufs[i].size()};
-                    memset(&rxHdrs[i], 0, sizeof(rxHdrs[i]));
-                    rxHdrs[i].msg_hdr.msg_iov = &rxIovs[i];
-                    rxHdrs[i].msg_hdr.msg_iovlen = 1;
+            for (int k = 0; k < nSock; ++k) {
+                if (!(pfds[k].revents & POLLIN)) continue;
+                while (true) {
+                    for (int i = 0; i < kRxBatch; ++i) {
+                        rxIovs[i] = {rxBufs[i].data(), rxBufs[i].size()};
+                        memset(&rxHdrs[i], 0, sizeof(rxHdrs[i]));
+                        rxHdrs[i].msg_hdr.msg_iov = &rxIovs[i];
+                        rxHdrs[i].msg_hdr.msg_iovlen = 1;
+                    }
+                    int nr = recvmmsg(fds[k], rxHdrs.data(), kRxBatch,
+                                      MSG_DONTWAIT, nullptr);
+                    if (nr <= 0) break;
+                    int64_t now = nowUs();
+                    for (int i = 0; i < nr; ++i) {
+                        const uint8_t* rb = rxBufs[i].data();
+                        if (rxHdrs[i].msg_len < 12) continue;
+                        uint16_t qid = (uint16_t)((rb[0] << 8) | rb[1]);
+                        int slot = qid & 0xFF;
+                        uint8_t seq = (uint8_t)(qid >> 8);
+                        if (slot >= W || sentAt[slot] == 0 ||
+                            (uint8_t)(slotSeq[slot] & 0xFF) != seq)
+                            continue;  // stale/duplicate
+                        int64_t lat = now - sentAt[slot];
+                        out->latBuckets[latBucket2(lat)]++;
+                        out->received++;
+                        uint8_t rcode = rb[3] & 0x0F;
+                        if (rcode == 0)
+                            out->rcodeNoerror++;
+                        else
+                            out->rcodeOther++;
+                        out->answers +=
+                            (uint64_t)((rb[6] << 8) | rb[7]);
+                        sentAt[slot] = 0;
+                        completed++;
+                        freeSlots.push_back(slot);
+                    }
+                    if (nr < kRxBatch) break;
                 }
-                int nr = recvmmsg(fd, rxHdrs.data(), kRxBatch,
-                                  MSG_DONTWAIT, nullptr);
-                if (nr <= 0) break;
-                int64_t now = nowUs();
-                for (int i = 0; i < nr; ++i) {
-                    const uint8_t* rb = rxBufs[i].data();
-                    if (rxHdrs[i].msg_len < 12) continue;
-                    uint16_t qid = (uint16_t)((rb[0] << 8) | rb[1]);
-                    int slot = qid & 0xFF;
-                    uint8_t seq = (uint8_t)(qid >> 8);
-                    if (slot >= W || sentAt[slot] == 0 ||
-                        (uint8_t)(slotSeq[slot] & 0xFF) != seq)
-                        continue;  // stale/duplicate
-                    int64_t lat = now - sentAt[slot];
-                    out->latBuckets[latBucket2(lat)]++;
-                    out->received++;
-                    uint8_t rcode = rb[3] & 0x0F;
-                    if (rcode == 0)
-                        out->rcodeNoerror++;
-                    else
-                        out->rcodeOther++;
-                    out->answers +=
-                        (uint64_t)((rb[6] << 8) | rb[7]);
-                    sentAt[slot] = 0;
-                    completed++;
-                    freeSlots.push_back(slot);
-                }
-                if (launched < target) batchSend(freeSlots);
-                else freeSlots.clear();
-                if (nr < kRxBatch) break;
             }
         }
+        if (launched < target) batchSend(freeSlots);
+        else freeSlots.clear();
         int64_t now = nowUs();
         if (now - lastSweep > 100000) {  // sweep timeouts every 100ms
             lastSweep = now;
@@ -247,7 +305,7 @@ void worker(const Config& cfg, int tid,
             else freeSlots.clear();
         }
     }
-    close(fd);
+    for (int fd : fds) close(fd);
 }
 
 }  // namespace
@@ -258,7 +316,7 @@ int main(int argc, char** argv) {
     signal(SIGPIPE, SIG_IGN);
     Config cfg;
     int c;
-    while ((c = getopt(argc, argv, "hs:p:n:c:t:f:B:T:R")) != -1) {
+    while ((c = getopt(argc, argv, "hs:p:n:c:t:P:r:f:B:T:R")) != -1) {
         switch (c) {
         case 'R': cfg.rd = true; break;
         case 's': cfg.server = optarg; break;
@@ -266,6 +324,8 @@ int main(int argc, char** argv) {
         case 'n': cfg.queries = strtoull(optarg, nullptr, 10); break;
         case 'c': cfg.window = atoi(optarg); break;
         case 't': cfg.threads = atoi(optarg); break;
+        case 'P': cfg.socksPerThread = atoi(optarg); break;
+        case 'r': cfg.rateQps = atof(optarg); break;
         case 'f': cfg.namesFile = optarg; break;
         case 'B': cfg.bindBase = optarg; break;
         case 'T': cfg.timeoutMs = atoi(optarg); break;
@@ -273,7 +333,8 @@ int main(int argc, char** argv) {
         default:
             fprintf(stderr,
                     "usage: dnsblast -s server -p port -n queries "
-                    "[-c window] [-t threads] [-f names-file] "
+                    "[-c window] [-t threads] [-P socks/thread] "
+                    "[-r offered-qps] [-f names-file] "
                     "[-B bind-base-ip] [-T timeout-ms] [-R]\n");
             return c == 'h' ? 0 : 1;
         }
