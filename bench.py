@@ -125,14 +125,40 @@ def wait_balancer_ready(port, n_backends, tmp, timeout=30):
     raise TimeoutError("balancer never became ready")
 
 
-def run_blast(port, queries, names_file, threads, window):
-    out = subprocess.run(
-        [str(REPO / "bin" / "dnsblast"), "-s", "127.0.0.1",
-         "-p", str(port), "-n", str(queries), "-c", str(window),
-         "-t", str(threads), "-f", str(names_file), "-B", "127.0.1.1",
-         "-T", "10000"],
-        capture_output=True, text=True, check=True)
+def run_blast(port, queries, names_file, threads, window, socks=8,
+              rate=0, timeout_ms=10000):
+    cmd = [str(REPO / "bin" / "dnsblast"), "-s", "127.0.0.1",
+           "-p", str(port), "-n", str(queries), "-c", str(window),
+           "-t", str(threads), "-P", str(socks),
+           "-f", str(names_file), "-B", "127.0.1.1",
+           "-T", str(timeout_ms)]
+    if rate > 0:
+        cmd += ["-r", str(int(rate))]
+    out = subprocess.run(cmd, capture_output=True, text=True, check=True)
     return json.loads(out.stdout.strip().splitlines()[-1])
+
+
+def calibrate_rate(port, names_file, threads, window, socks, capacity,
+                   slo_us):
+    """qps@SLO discovery (untimed): walk a descending rate ladder from
+    the closed-loop capacity; the first offered rate that sustains
+    p99 <= SLO with zero timeouts and full delivery is the operating
+    point. Fixed-rate steps at that point are comparable round over
+    round, unlike the chaotic closed-loop saturation equilibrium
+    (profiles/SCALING.md round-1: steps swung +/-40% at saturation)."""
+    for frac in (1.0, 0.95, 0.9, 0.85, 0.8, 0.72, 0.64, 0.55):
+        rate = capacity * frac
+        q = max(200_000, int(rate * 1.2))
+        r = run_blast(port, q, names_file, threads, window, socks,
+                      rate=rate)
+        ok = (r["timeouts"] == 0 and r["p99_us"] <= slo_us and
+              r["qps"] >= 0.97 * rate)
+        log(f"calibrate {frac:.2f}x: offered {rate:.0f} -> "
+            f"{r['qps']:.0f} qps, p99 {r['p99_us']}us, "
+            f"timeouts {r['timeouts']} {'OK' if ok else 'over SLO'}")
+        if ok:
+            return rate
+    return capacity * 0.5
 
 
 def main():
@@ -186,14 +212,19 @@ def main():
         workers = min({1: 8}.get(n, 16), max(4, ncpu // 8))
         threads = {1: 12, 2: 16, 4: 16}.get(n, max(12, 3 * n))
         window = 128
+        socks = 8  # flows = threads*socks >> balancer reuseport shards
         # experiment overrides (profiling/tuning only)
         workers = int(os.environ.get("BENCH_WORKERS", workers))
         threads = int(os.environ.get("BENCH_THREADS", threads))
         window = int(os.environ.get("BENCH_WINDOW", window))
+        socks = int(os.environ.get("BENCH_SOCKS", socks))
     else:
         workers = 1
         threads = min(4 * n, max(2, ncpu // 2))
         window = args.window
+        socks = 4
+    slo_us = int(os.environ.get("BENCH_SLO_US", "2000"))
+    closed_loop = os.environ.get("BENCH_CLOSED_LOOP", "0") == "1"
     result = {}
 
     tmp = None
@@ -230,10 +261,22 @@ def main():
             # measured +11% at N=8 vs a single warmup step
             log(f"balancer ready on :{bal_port}; ramp + warmup "
                 f"{args.warmup} x {q_step} queries")
+            capacity = 0.0
             for _ in range(2):
-                run_blast(bal_port, q_step, names_file, threads, window)
+                r = run_blast(bal_port, q_step, names_file, threads,
+                              window, socks)
+                capacity = max(capacity, r["qps"])
+            rate = 0
+            if not closed_loop:
+                # qps@SLO protocol: fixed offered rate for the timed
+                # steps, discovered against the SLO (untimed)
+                rate = calibrate_rate(bal_port, names_file, threads,
+                                      window, socks, capacity, slo_us)
+                log(f"operating point: {rate:.0f} qps offered "
+                    f"(capacity {capacity:.0f}, SLO p99<={slo_us}us)")
             for _ in range(args.warmup):
-                run_blast(bal_port, q_step, names_file, threads, window)
+                run_blast(bal_port, q_step, names_file, threads, window,
+                          socks, rate=rate)
 
         churn_stop = None
         if rank == 0 and args.churn_qps > 0:
@@ -267,7 +310,7 @@ def main():
         if rank == 0:
             for s in range(args.steps):
                 last = run_blast(bal_port, q_step, names_file, threads,
-                                 window)
+                                 window, socks, rate=rate)
                 log(f"step {s + 1}/{args.steps}: "
                     f"{last['qps']:.0f} qps, p99 {last['p99_us']}us")
         barrier()
@@ -303,6 +346,9 @@ def main():
                 "data": "synthetic",
                 "config": {
                     "model": "binder-dns-zk",
+                    "protocol": ("closed-loop" if closed_loop else
+                                 f"fixed-rate@SLO(p99<={slo_us}us)"),
+                    "offered_qps": int(rate) if rank == 0 else None,
                     "tree_records": args.tree_records,
                     "churn_qps": args.churn_qps,
                     "query_mix": "A+SRV uniform",
