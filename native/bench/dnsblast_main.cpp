@@ -243,10 +243,17 @@ void worker(const Config& cfg, int tid,
     for (int k = 0; k < nSock; ++k) pfds[k] = {fds[k], POLLIN, 0};
     int64_t lastSweep = nowUs();
     while (completed < target && !abort->load()) {
-        /* paced mode: short poll so the token bucket is serviced even
-         * while the window has room */
-        int pollMs = rate > 0 ? 2 : 50;
-        int rv = poll(pfds.data(), (nfds_t)nSock, pollMs);
+        /* paced mode: sub-ms ppoll so the token bucket drains in
+         * small quanta (poll()'s 1 ms floor makes every thread dump a
+         * whole 1-2 ms token backlog in one synchronized burst, which
+         * shows up as multi-ms p99 at the server) */
+        int rv;
+        if (rate > 0) {
+            struct timespec ts {0, 250 * 1000};  /* 250 us */
+            rv = ppoll(pfds.data(), (nfds_t)nSock, &ts, nullptr);
+        } else {
+            rv = poll(pfds.data(), (nfds_t)nSock, 50);
+        }
         if (rv > 0) {
             for (int k = 0; k < nSock; ++k) {
                 if (!(pfds[k].revents & POLLIN)) continue;

@@ -1,0 +1,99 @@
+#!/usr/bin/env python3
+"""Paced-mode diagnosis probe: where do the multi-ms tails at N=8 come
+from? Builds the bench topology once, then sweeps offered rate, client
+threads/sockets, and balancer workers, printing one JSON line per
+experiment plus balancer drop/overwrite deltas.
+
+usage: paced_probe.py [--procs 8] [--workers 16] [--tree 10000]
+"""
+import argparse
+import json
+import os
+import socket
+import subprocess
+import sys
+import tempfile
+import time
+from pathlib import Path
+
+REPO = Path(__file__).resolve().parent.parent
+sys.path.insert(0, str(REPO))
+
+import bench  # noqa: E402  (reuse the harness pieces)
+
+
+def balstat(path):
+    with socket.socket(socket.AF_UNIX) as s:
+        s.settimeout(2)
+        s.connect(str(path))
+        return json.loads(s.recv(1 << 20).decode())
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--procs", type=int, default=8)
+    ap.add_argument("--workers", type=int, default=16)
+    ap.add_argument("--tree", type=int, default=10000)
+    ap.add_argument("--quick", action="store_true")
+    args = ap.parse_args()
+
+    from binder_amd.harness import free_port
+    from binder_amd.stubzk import StubZk
+
+    tmp = Path(tempfile.mkdtemp(prefix="paced-probe-"))
+    names_file = tmp / "names.txt"
+    zk = StubZk().start()
+    bench.build_tree(zk, names_file, args.tree)
+    backends, sockdir = bench.start_backends(
+        args.procs, tmp, zk.port,
+        last_name=f"h{args.tree // 2 - 1}.foo.com")
+    port = free_port()
+    bal = bench.start_balancer(tmp, sockdir, port, workers=args.workers)
+    bench.wait_balancer_ready(port, args.procs, tmp)
+    stats_path = tmp / "stats.sock"
+
+    def stat_delta(before, after):
+        return {k: after[k] - before[k]
+                for k in ("udp_queries", "udp_replies", "drops")} | {
+            "overwrites": sum(b.get("overwrites", 0)
+                              for b in after["backends"]) -
+            sum(b.get("overwrites", 0) for b in before["backends"])}
+
+    def run(tag, queries, threads, window, socks, rate):
+        b0 = balstat(stats_path)
+        r = bench.run_blast(port, queries, names_file, threads, window,
+                            socks=socks, rate=rate)
+        b1 = balstat(stats_path)
+        out = {"tag": tag, "threads": threads, "window": window,
+               "socks": socks, "rate": rate,
+               "qps": r["qps"], "p50_us": r["p50_us"],
+               "p90_us": r["p90_us"], "p99_us": r["p99_us"],
+               "timeouts": r["timeouts"],
+               "bal": stat_delta(b0, b1)}
+        print(json.dumps(out), flush=True)
+        return r
+
+    n = args.procs
+    q = 2_000_000 if not args.quick else 400_000
+    try:
+        # ramp
+        run("ramp1", q, 24, 128, 8, 0)
+        cap = run("closed-P8", q, 24, 128, 8, 0)["qps"]
+        run("closed-P1", q, 24, 128, 1, 0)
+        for frac in (0.9, 0.8, 0.7):
+            rate = int(cap * frac)
+            run(f"paced-{frac}-t24-P8", q, 24, 128, 8, rate)
+        rate = int(cap * 0.8)
+        run("paced-0.8-t48-P4", q, 48, 128, 4, rate)
+        run("paced-0.8-t24-P1", q, 24, 128, 1, rate)
+        run("paced-0.8-t24-P8-w256", q, 24, 256, 8, rate)
+        run("paced-0.8-t12-P8", q, 12, 256, 8, rate)
+    finally:
+        bal.terminate()
+        for b in backends:
+            b.stop()
+        zk.stop()
+
+
+if __name__ == "__main__":
+    main()
