@@ -8,12 +8,9 @@ usage: paced_probe.py [--procs 8] [--workers 16] [--tree 10000]
 """
 import argparse
 import json
-import os
 import socket
-import subprocess
 import sys
 import tempfile
-import time
 from pathlib import Path
 
 REPO = Path(__file__).resolve().parent.parent
