@@ -76,15 +76,13 @@ def main():
         # ramp
         run("ramp1", q, 24, 128, 8, 0)
         cap = run("closed-P8", q, 24, 128, 8, 0)["qps"]
-        run("closed-P1", q, 24, 128, 1, 0)
-        for frac in (0.9, 0.8, 0.7):
+        for t in (8, 12, 16, 24):
+            rate = int(cap * 0.85)
+            run(f"paced-0.85-t{t}-P8-w256", q, t, 256, 8, rate)
+        for frac in (0.95, 0.9, 1.0):
             rate = int(cap * frac)
-            run(f"paced-{frac}-t24-P8", q, 24, 128, 8, rate)
-        rate = int(cap * 0.8)
-        run("paced-0.8-t48-P4", q, 48, 128, 4, rate)
-        run("paced-0.8-t24-P1", q, 24, 128, 1, rate)
-        run("paced-0.8-t24-P8-w256", q, 24, 256, 8, rate)
-        run("paced-0.8-t12-P8", q, 12, 256, 8, rate)
+            run(f"paced-{frac}-t16-P8-w256", q, 16, 256, 8, rate)
+        run("closed-t16-P8-w256", q, 16, 256, 8, 0)
     finally:
         bal.terminate()
         for b in backends:
