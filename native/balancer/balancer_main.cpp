@@ -138,6 +138,7 @@ struct Backend {
      * on the stats socket, not silent */
     uint64_t overwrites = 0;
     size_t outOff = 0;  /* consumed prefix of `out` (flush cursor) */
+    size_t sweepCursor = 0;  /* incremental pending-expiry scan */
     size_t remotes = 0;
 
     PendingSlot* slotFor(uint32_t reqId) {
@@ -990,17 +991,30 @@ void Balancer::sweep() {
     }
     for (auto& [path, be] : backends_) {
         if (be->fd < 0) continue;
-        /* expire stale pendings (uint32 wrap-safe comparison) */
+        /* Expire stale pendings (uint32 wrap-safe comparison) in a
+         * bounded chunk per tick: a full walk of the ring is ~8 MB of
+         * cold cache per backend and was measured as a multi-ms stall
+         * that put the sweep duration straight into client p99
+         * (profiles/ probe data). Correctness doesn't need the full
+         * walk — an un-reaped slot is reclaimed on its next ring
+         * collision (counted as overwrite) — the chunked scan only
+         * keeps pendingCount from drifting for load-shedding. */
         if (be->pending && be->pendingCount > 0) {
             uint32_t now32 = (uint32_t)now;
-            for (size_t i = 0; i < g_pendingSlots; ++i) {
-                PendingSlot& s = be->pending[i];
+            size_t chunk = g_pendingSlots / 16 + 1;
+            if (chunk > 4096) chunk = 4096;
+            for (size_t k = 0; k < chunk; ++k) {
+                PendingSlot& s =
+                    be->pending[(be->sweepCursor + k) &
+                                (g_pendingSlots - 1)];
                 if (s.reqId != 0 &&
                     (int32_t)(now32 - s.expiresAtMs) > 0) {
                     s.reqId = 0;
                     be->pendingCount--;
                 }
             }
+            be->sweepCursor =
+                (be->sweepCursor + chunk) & (g_pendingSlots - 1);
         }
         /* health probe */
         if (now - be->pingSentAt >= kPingIntervalMs) {

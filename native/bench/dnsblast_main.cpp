@@ -168,15 +168,21 @@ void worker(const Config& cfg, int tid,
     std::vector<int> freeSlots;
     freeSlots.reserve(W);
 
-    /* How many more queries may launch right now (rate pacing). */
+    /* How many more queries may launch right now (rate pacing).
+     * Launches are quantized (min 8 or 100 us worth) so the spinning
+     * paced loop amortizes sendmmsg instead of degenerating into one
+     * syscall per packet at high per-thread rates. */
+    int64_t lastLaunchUs = 0;
     auto allowance = [&]() -> uint64_t {
         if (launched >= target) return 0;
         uint64_t left = target - launched;
         if (rate <= 0) return left;
-        uint64_t paced = (uint64_t)((double)(nowUs() - tStart) * rate /
-                                    1e6);
+        int64_t now = nowUs();
+        uint64_t paced = (uint64_t)((double)(now - tStart) * rate / 1e6);
         if (paced <= launched) return 0;
         uint64_t a = paced - launched;
+        if (a < left && a < 8 && now - lastLaunchUs < 100) return 0;
+        lastLaunchUs = now;
         return a < left ? a : left;
     };
 
