@@ -32,6 +32,10 @@ def main():
     ap.add_argument("--workers", type=int, default=16)
     ap.add_argument("--tree", type=int, default=10000)
     ap.add_argument("--quick", action="store_true")
+    ap.add_argument("--sweep", action="store_true",
+                    help="sweep balancer workers x client threads "
+                         "(closed loop) for the quota-bound operating "
+                         "point, then pace the best")
     args = ap.parse_args()
 
     from binder_amd.harness import free_port
@@ -70,19 +74,50 @@ def main():
         print(json.dumps(out), flush=True)
         return r
 
-    n = args.procs
     q = 2_000_000 if not args.quick else 400_000
     try:
-        # ramp
-        run("ramp1", q, 24, 128, 8, 0)
-        cap = run("closed-P8", q, 24, 128, 8, 0)["qps"]
-        for t in (8, 12, 16, 24):
-            rate = int(cap * 0.85)
-            run(f"paced-0.85-t{t}-P8-w256", q, t, 256, 8, rate)
-        for frac in (0.95, 0.9, 1.0):
-            rate = int(cap * frac)
-            run(f"paced-{frac}-t16-P8-w256", q, 16, 256, 8, rate)
-        run("closed-t16-P8-w256", q, 16, 256, 8, 0)
+        if not args.sweep:
+            run("ramp1", q, 24, 128, 8, 0)
+            cap = run("closed-P8", q, 24, 128, 8, 0)["qps"]
+            for t in (8, 12, 16, 24):
+                rate = int(cap * 0.85)
+                run(f"paced-0.85-t{t}-P8-w256", q, t, 256, 8, rate)
+            for frac in (0.95, 0.9, 1.0):
+                rate = int(cap * frac)
+                run(f"paced-{frac}-t16-P8-w256", q, 16, 256, 8, rate)
+            run("closed-t16-P8-w256", q, 16, 256, 8, 0)
+            return
+        # quota-aware sweep: restart the balancer per worker count
+        nonlocal_best = {"qps": 0}
+        for w in (4, 6, 8, 12, 16):
+            bal.terminate()
+            bal.wait(timeout=5)
+            port2 = free_port()
+            bal = bench.start_balancer(tmp, sockdir, port2,
+                                       workers=w)
+            bench.wait_balancer_ready(port2, args.procs, tmp)
+            port = port2
+
+            def run2(tag, threads, window, socks, rate):
+                b0 = balstat(stats_path)
+                r = bench.run_blast(port, q, names_file, threads,
+                                    window, socks=socks, rate=rate)
+                b1 = balstat(stats_path)
+                out = {"tag": tag, "workers": w, "threads": threads,
+                       "window": window, "socks": socks, "rate": rate,
+                       "qps": r["qps"], "p50_us": r["p50_us"],
+                       "p90_us": r["p90_us"], "p99_us": r["p99_us"],
+                       "timeouts": r["timeouts"],
+                       "bal": stat_delta(b0, b1)}
+                print(json.dumps(out), flush=True)
+                return r
+
+            run2(f"w{w}-ramp", 8, 256, 8, 0)
+            for t in (4, 6, 8, 12):
+                r = run2(f"w{w}-t{t}", t, 256, 8, 0)
+                if r["qps"] > nonlocal_best["qps"]:
+                    nonlocal_best = {"qps": r["qps"], "w": w, "t": t}
+        print(json.dumps({"best": nonlocal_best}), flush=True)
     finally:
         bal.terminate()
         for b in backends:
