@@ -37,6 +37,21 @@ def log(msg):
     print(f"# bench: {msg}", file=sys.stderr, flush=True)
 
 
+def effective_cpus():
+    """CPUs actually available to this cgroup: the bench boxes expose
+    256 hw threads but cap cpu.max at 16 cores — sizing against
+    os.cpu_count() oversubscribes the quota and CFS throttling puts
+    10-100 ms whole-group freezes straight into p99 (measured,
+    profiles/SCALING.md round 2)."""
+    try:
+        parts = Path("/sys/fs/cgroup/cpu.max").read_text().split()
+        if parts[0] != "max":
+            return max(2, int(parts[0]) // int(parts[1]))
+    except (OSError, ValueError, IndexError):
+        pass
+    return os.cpu_count() or 8
+
+
 def build_tree(zk, names_path, records=10000):
     """R records: R/2 hosts + R/10 services x 4 members (=R nodes with
     payloads), uniform A+SRV query mix. R=10000 is the BASELINE
@@ -201,17 +216,15 @@ def main():
             pass
 
     q_step = args.queries_per_proc * n
-    ncpu = os.cpu_count() or 8
-    # size the harness to the machine: the load generator + balancer
-    # must not starve the backends on small boxes (probe-derived
-    # operating point: scripts/scale_probe.py + profiles/SCALING.md)
-    if ncpu >= 8 * n + 8:
-        # probe-swept operating points (scripts/scale_probe.py;
-        # profiles/SCALING.md): 16 balancer workers from N=2 up,
-        # window 128 (same qps as 192 at ~20x lower p99)
-        workers = min({1: 8}.get(n, 16), max(4, ncpu // 8))
-        threads = {1: 12, 2: 16, 4: 16}.get(n, max(12, 3 * n))
-        window = 128
+    ncpu = min(os.cpu_count() or 8, effective_cpus())
+    # Size the harness to the CPU actually available (cgroup quota, not
+    # hw threads): operating points are probe-swept on the driver box
+    # class (16-core quota; scripts/paced_probe.py --sweep, committed
+    # under profiles/).
+    if ncpu >= 12:
+        workers = max(2, min(8, (ncpu * 3) // 8))      # 6 @ 16 cpus
+        threads = max(4, min(8, ncpu // 2))            # 8 @ 16 cpus
+        window = 256
         socks = 8  # flows = threads*socks >> balancer reuseport shards
         # experiment overrides (profiling/tuning only)
         workers = int(os.environ.get("BENCH_WORKERS", workers))

@@ -172,11 +172,12 @@ class Balancer {
   public:
     Balancer(EventLoop* loop, Logger log, std::string host, uint16_t port,
              std::string sockDir, std::string statsPath, int rescanMs,
-             bool reusePort = false, int workerId = 0)
+             bool reusePort = false, int workerId = 0, int nWorkers = 1)
         : loop_(loop), log_(std::move(log)), host_(std::move(host)),
           port_(port), sockDir_(std::move(sockDir)),
           statsPath_(std::move(statsPath)), rescanMs_(rescanMs),
-          reusePort_(reusePort), workerId_(workerId) {}
+          reusePort_(reusePort), workerId_(workerId),
+          nWorkers_(nWorkers) {}
 
     bool start();
     void stop();
@@ -207,6 +208,7 @@ class Balancer {
     int rescanMs_;
     bool reusePort_ = false;
     int workerId_ = 0;
+    int nWorkers_ = 1;
 
     int udpFd_ = -1, tcpFd_ = -1, statsFd_ = -1;
     std::map<std::string, std::shared_ptr<Backend>> backends_;  // by path
@@ -229,7 +231,7 @@ class Balancer {
     /* per-instance batch I/O arenas — workers are threads, so these
      * must NOT be static (a shared-static race here collapsed
      * multi-worker throughput) */
-    static constexpr int kBatch = 64;
+    static constexpr int kBatch = 128;
     uint8_t rxBufs_[kBatch][4096];
     struct mmsghdr rxHdrs_[kBatch];
     struct iovec rxIovs_[kBatch];
@@ -575,15 +577,34 @@ void Balancer::onBackendEvent(std::shared_ptr<Backend> be, uint32_t ev) {
 }
 
 Backend* Balancer::chooseLeastLoaded() {
-    /* healthy backend with fewest remotes ACROSS ALL WORKERS */
+    /*
+     * Affinity-aware worker->backend mapping: each SO_REUSEPORT worker
+     * pins its remotes to a slice of the backends (still falling back
+     * to any healthy backend when the slice is down). Concentrating a
+     * worker's traffic on few backends makes every backend write/read
+     * carry a full batch instead of 1/Nth of one — the chain is
+     * syscall-bound under the box's CPU quota, so emptier batches
+     * directly cost throughput. Backend load stays balanced because
+     * the kernel spreads flows evenly across workers and slices cover
+     * all backends.
+     */
     Backend* best = nullptr;
     int bestCount = 0;
+    bool bestPreferred = false;
+    int nb = (int)backendsById_.size();
+    int idx = 0;
     for (auto& [id, be] : backendsById_) {
+        int i = idx++;
         if (!be->ok || be->fd < 0) continue;
+        bool preferred =
+            nWorkers_ >= nb ? (workerId_ % (nb > 0 ? nb : 1)) == i
+                            : (i % nWorkers_) == workerId_;
         int c = g_pins.get(be->path);
-        if (best == nullptr || c < bestCount) {
+        if (best == nullptr || (preferred && !bestPreferred) ||
+            (preferred == bestPreferred && c < bestCount)) {
             best = be.get();
             bestCount = c;
+            bestPreferred = preferred;
         }
     }
     return best;
@@ -1094,7 +1115,8 @@ int main(int argc, char** argv) {
         bals.emplace_back(std::make_unique<Balancer>(
             loops[w].get(),
             log.child({{"worker", Json((int64_t)w)}}), host, port, dir,
-            w == 0 ? stats : std::string(), rescanMs, workers > 1, w));
+            w == 0 ? stats : std::string(), rescanMs, workers > 1, w,
+            workers));
         if (!bals[w]->start()) return 1;
     }
     for (int w = 0; w < workers; ++w)

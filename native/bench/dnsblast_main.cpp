@@ -249,26 +249,26 @@ void worker(const Config& cfg, int tid,
     for (int k = 0; k < nSock; ++k) pfds[k] = {fds[k], POLLIN, 0};
     int64_t lastSweep = nowUs();
     while (completed < target && !abort->load()) {
-        /* paced mode: SPIN, never sleep. Sleeping generator threads
-         * (poll/ppoll) measurably poison the tail: scheduler wakeup
-         * latency lands in the middle of the latency measurement and
-         * the token bucket drains in synchronized bursts — probe data
-         * (profiles/: paced p99 9-59 ms while closed loop showed
-         * 2 ms, worse with MORE threads). A load generator may burn
-         * its cores; non-blocking poll keeps launches smooth and
-         * replies drained at microsecond granularity. */
+        /* Paced mode: wait only until the next 8-packet launch
+         * quantum accrues (clamped to 250 us). The box runs under a
+         * CPU quota, so busy-spinning generator threads steal quota
+         * from the server chain and trigger CFS throttling — whose
+         * whole-group freezes are exactly the multi-ms p99 artifacts
+         * the paced protocol exists to avoid. Sub-quantum sleeps keep
+         * launches smooth at microsecond granularity without burning
+         * idle cycles. */
         int rv;
         if (rate > 0) {
-            struct timespec ts {0, 0};
-            rv = ppoll(pfds.data(), (nfds_t)nSock, &ts, nullptr);
-            if (rv == 0) {
-                for (int i = 0; i < 64; ++i)
-#if defined(__x86_64__)
-                    __builtin_ia32_pause();
-#else
-                    ;
-#endif
+            uint64_t backlog = allowance();
+            long waitNs;
+            if (backlog >= 8) {
+                waitNs = 0;
+            } else {
+                double need = (double)(8 - backlog) / rate * 1e9;
+                waitNs = need < 250000 ? (long)need : 250000;
             }
+            struct timespec ts {0, waitNs};
+            rv = ppoll(pfds.data(), (nfds_t)nSock, &ts, nullptr);
         } else {
             rv = poll(pfds.data(), (nfds_t)nSock, 50);
         }
