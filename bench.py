@@ -222,8 +222,10 @@ def main():
     # class (16-core quota; scripts/paced_probe.py --sweep, committed
     # under profiles/).
     if ncpu >= 12:
-        workers = max(2, min(8, (ncpu * 3) // 8))      # 6 @ 16 cpus
-        threads = max(4, min(8, ncpu // 2))            # 8 @ 16 cpus
+        # probe-swept best on the 16-core-quota box class: w8/t6 won
+        # at every N (profiles/, sweep2_n*.jsonl)
+        workers = max(2, min(8, ncpu // 2))            # 8 @ 16 cpus
+        threads = max(4, min(6, (ncpu * 3) // 8))      # 6 @ 16 cpus
         window = 256
         socks = 8  # flows = threads*socks >> balancer reuseport shards
         # experiment overrides (profiling/tuning only)
