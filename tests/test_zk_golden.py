@@ -329,10 +329,12 @@ def test_client_emits_golden_bytes(tmp_path):
 
 def test_client_handles_golden_expiry_response(tmp_path):
     """A spec-built expiry ConnectResponse (timeOut=0, sessionId=0)
-    must make the client report failure, not hang or crash."""
+    must make the client drop the dead session and reconnect with a
+    FRESH handshake (sessionId=0, zeroed passwd) — the rebuild path
+    real ZooKeeper triggers on session expiry."""
     srv = socket.socket()
     srv.bind(("127.0.0.1", 0))
-    srv.listen(1)
+    srv.listen(2)
     port = srv.getsockname()[1]
     proc = subprocess.Popen(
         [str(ZKTOOL), "-s", f"127.0.0.1:{port}", "get", "/com"],
@@ -342,12 +344,19 @@ def test_client_handles_golden_expiry_response(tmp_path):
         conn, _ = srv.accept()
         conn.settimeout(10)
         read_packet(conn)
+        # spec expiry response: proto 0, timeOut 0, sessionId 0
         conn.sendall(packet(be32(0) + be32(0) + be64(0) +
                             jstr(b"\x00" * 16) + b"\x00"))
-        out, err = proc.communicate(timeout=15)
-        assert proc.returncode != 0
-    except subprocess.TimeoutExpired:
-        pytest.fail("client hung on expiry ConnectResponse")
+        # the client must come back with a brand-new session handshake
+        conn2, _ = srv.accept()
+        conn2.settimeout(10)
+        got = read_packet(conn2)
+        golden_fresh = connect_request(timeout_ms=10000,
+                                       read_only=False)[4:]
+        assert got == golden_fresh, \
+            f"post-expiry handshake drifted: {got.hex()}"
+        conn2.close()
+        conn.close()
     finally:
         proc.kill()
         srv.close()
