@@ -33,7 +33,8 @@ PY_INCLUDES := $(shell $(PYTHON) -m pybind11 --includes)
 PYMOD := binder_amd/_native$(PY_EXT_SUFFIX)
 
 BINARIES := bin/binderd bin/binder-balancer bin/dnsblast \
-	bin/binder-adjust bin/binder-supervisor bin/zklogcat bin/zktool
+	bin/binder-adjust bin/binder-supervisor bin/zklogcat bin/zktool \
+	bin/zkd
 
 all: $(PYMOD) $(BINARIES)
 
@@ -62,6 +63,10 @@ bin/zklogcat: $(CORE_OBJS) $(BUILD)/native/zklog/zklogcat_main.o
 	$(CXX) $(CXXFLAGS) $^ -o $@ $(LDFLAGS)
 
 bin/zktool: $(CORE_OBJS) $(BUILD)/native/zk/client.o $(BUILD)/native/zk/zktool_main.o
+	@mkdir -p bin
+	$(CXX) $(CXXFLAGS) $^ -o $@ $(LDFLAGS)
+
+bin/zkd: $(CORE_OBJS) $(BUILD)/native/zkd/zkd_main.o
 	@mkdir -p bin
 	$(CXX) $(CXXFLAGS) $^ -o $@ $(LDFLAGS)
 

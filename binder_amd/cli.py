@@ -21,6 +21,8 @@ import socket
 import sys
 from pathlib import Path
 
+from . import REPO_ROOT
+
 
 def cmd_balstat(args):
     with socket.socket(socket.AF_UNIX) as s:
@@ -154,10 +156,24 @@ def cmd_zk(args):
 
 
 def cmd_zkd(args):
+    """Run the registry. The native bin/zkd is the supported
+    single-node registry; the in-process Python stub (stubzk) remains
+    a CI/test fixture and serves as fallback when the native binary
+    has not been built."""
+    import os as _os
+
+    native = REPO_ROOT / "bin" / "zkd"
+    if native.exists():
+        _os.execv(str(native), [str(native), "-H", args.host,
+                                "-p", str(args.port),
+                                "-d", args.data_dir])
+
     import signal
     import time as _time
 
     from .stubzk import StubZk
+    print("warning: bin/zkd not built; running the Python stub "
+          "(dev fallback)", flush=True)
     zk = StubZk(host=args.host, port=args.port,
                 txnlog_dir=args.data_dir)
     zk.start()

@@ -154,3 +154,50 @@ class BinderProcess:
 
     def __exit__(self, *exc):
         self.stop()
+
+
+ZKD = Path(os.environ.get("ZKD_BIN", REPO_ROOT / "bin" / "zkd"))
+
+
+class NativeZkd:
+    """bin/zkd (the native single-node registry) as a test fixture."""
+
+    def __init__(self, port=0, data_dir=None, session_timeout_ms=30000,
+                 host="127.0.0.1"):
+        self.host = host
+        self.port = port
+        self.data_dir = data_dir
+        self.session_timeout_ms = session_timeout_ms
+        self.proc: Optional[subprocess.Popen] = None
+        self.nodes_restored = 0
+
+    def start(self, timeout=10.0):
+        cmd = [str(ZKD), "-H", self.host, "-p", str(self.port),
+               "-t", str(self.session_timeout_ms)]
+        if self.data_dir:
+            cmd += ["-d", str(self.data_dir)]
+        self.proc = subprocess.Popen(
+            cmd, stdout=subprocess.PIPE, stderr=subprocess.DEVNULL,
+            text=True)
+        # "zkd listening on H:P, N nodes restored"
+        deadline = time.time() + timeout
+        line = ""
+        while time.time() < deadline:
+            line = self.proc.stdout.readline()
+            if "listening on" in line:
+                break
+        if "listening on" not in line:
+            raise RuntimeError("zkd never reported listening")
+        self.port = int(line.split(":")[1].split(",")[0])
+        self.nodes_restored = int(line.split(",")[1].split()[0])
+        return self
+
+    def stop(self):
+        if self.proc and self.proc.poll() is None:
+            self.proc.terminate()
+            try:
+                self.proc.wait(timeout=5)
+            except subprocess.TimeoutExpired:
+                self.proc.kill()
+                self.proc.wait()
+        self.proc = None

@@ -28,91 +28,34 @@ pzxid (int64) — 68 bytes.
 import socket
 import struct
 import subprocess
+import time
 
 import pytest
 
+from zkwire import (ACL_OPEN, be32, be64, connect_request, jstr, packet,
+                    parse_reply_header, parse_stat, read_packet, req)
+
 from binder_amd import REPO_ROOT
+from binder_amd.harness import NativeZkd
 from binder_amd.stubzk import StubZk
 
 ZKTOOL = REPO_ROOT / "bin" / "zktool"
 
 
-# --- independent jute packers (do NOT reuse stubzk's) ---------------
-
-def be32(v):
-    return struct.pack(">i", v)
-
-
-def be64(v):
-    return struct.pack(">q", v)
-
-
-def jstr(s):
-    b = s.encode() if isinstance(s, str) else s
-    return be32(len(b)) + b
-
-
-def packet(body):
-    """Every ZK packet is a 4-byte BE length prefix + body."""
-    return be32(len(body)) + body
-
-
-def read_packet(sock):
-    hdr = b""
-    while len(hdr) < 4:
-        chunk = sock.recv(4 - len(hdr))
-        assert chunk, "peer closed"
-        hdr += chunk
-    (n,) = struct.unpack(">i", hdr)
-    body = b""
-    while len(body) < n:
-        chunk = sock.recv(n - len(body))
-        assert chunk, "peer closed mid-packet"
-        body += chunk
-    return body
-
-
-# ConnectRequest (zookeeper.jute proto.ConnectRequest):
-#   int protocolVersion; long lastZxidSeen; int timeOut;
-#   long sessionId; buffer passwd;  [+ optional boolean readOnly 3.4.6+]
-def connect_request(timeout_ms=30000, session_id=0,
-                    passwd=b"\x00" * 16, read_only=None):
-    body = be32(0) + be64(0) + be32(timeout_ms) + be64(session_id) + \
-        jstr(passwd)
-    if read_only is not None:
-        body += bytes([1 if read_only else 0])
-    return packet(body)
-
-
-# RequestHeader: int xid; int type  (fixed opcodes: create=1, delete=2,
-# getData=4, setData=5, getChildren2=12, ping=11 w/ xid -2)
-def req(xid, op, payload=b""):
-    return packet(be32(xid) + be32(op) + payload)
-
-
-# default ACL world:anyone with ALL perms (0x1f); jute:
-# vector<ACL>{ int perms; Id id { string scheme; string id; } }
-ACL_OPEN = be32(1) + be32(0x1F) + jstr("world") + jstr("anyone")
-
-
-def parse_reply_header(body):
-    xid, zxid, err = struct.unpack(">iqi", body[:16])
-    return xid, zxid, err, body[16:]
-
-
-def parse_stat(b):
-    names = ("czxid", "mzxid", "ctime", "mtime", "version", "cversion",
-             "aversion", "ephemeralOwner", "dataLength", "numChildren",
-             "pzxid")
-    vals = struct.unpack(">qqqqiiiqiiq", b[:68])
-    return dict(zip(names, vals)), b[68:]
-
-
 # --- server side: zkd must accept/answer the spec bytes -------------
 
-@pytest.fixture()
-def zk():
-    z = StubZk().start()
+@pytest.fixture(params=["stub", "native"])
+def zk(request, tmp_path):
+    """Both registry implementations must accept/answer the spec
+    bytes: the in-proc Python stub (CI fixture) AND the native zkd
+    (the supported single-node registry)."""
+    if request.param == "stub":
+        z = StubZk().start()
+        z.expire_all = z.expire_sessions
+    else:
+        # short session timeout so natural expiry is testable
+        z = NativeZkd(session_timeout_ms=800).start()
+        z.expire_all = lambda: time.sleep(2.2)
     yield z
     z.stop()
 
@@ -162,7 +105,7 @@ def test_server_session_resume_and_expiry_bytes(zk):
     s2.close()
     # expiry: after the server expires the session, the spec response
     # is sessionId=0 AND timeOut=0 (clients detect expiry by that)
-    zk.expire_sessions()
+    zk.expire_all()
     s3, proto, timeout, sid3, _ = connect_raw(zk, session_id=sid,
                                               passwd=passwd)
     assert (timeout, sid3) == (0, 0)
