@@ -246,18 +246,20 @@ def main():
     stack = []
     try:
         if rank == 0:
-            from binder_amd.harness import free_port
-            from binder_amd.stubzk import StubZk
+            from binder_amd.harness import free_port, NativeZkd
+            from binder_amd.zkclient import ZkConn
             tmp = Path(tempfile.mkdtemp(prefix="binder-bench-"))
             names_file = tmp / "names.txt"
-            log(f"starting stub ZK + building "
+            log(f"starting native zkd + building "
                 f"{args.tree_records}-record tree")
-            zk = StubZk().start()
-            stack.append(zk.stop)
+            zkd = NativeZkd().start()
+            stack.append(zkd.stop)
+            zk = ZkConn("127.0.0.1", zkd.port)
+            stack.append(zk.close)
             nrec = build_tree(zk, names_file, args.tree_records)
             log(f"{nrec} records; starting {n} binderd process(es)")
             backends, sockdir = start_backends(
-                n, tmp, zk.port,
+                n, tmp, zkd.port,
                 last_name=f"h{args.tree_records // 2 - 1}.foo.com")
             stack.append(lambda: [b.stop() for b in backends])
             bal_port = free_port()
@@ -299,6 +301,8 @@ def main():
             import threading
             churn_stop = threading.Event()
             n_hosts = args.tree_records // 2
+            churn_zk = ZkConn("127.0.0.1", zkd.port)
+            stack.append(churn_zk.close)
 
             def churner():
                 rng = random.Random(7)
@@ -306,7 +310,7 @@ def main():
                 i = 0
                 while not churn_stop.is_set():
                     h = rng.randrange(n_hosts)
-                    zk.put(f"/com/foo/h{h}", json.dumps(
+                    churn_zk.put(f"/com/foo/h{h}", json.dumps(
                         {"type": "host",
                          "host": {"address":
                                   f"10.{(h >> 8) & 255}.{h & 255}."
@@ -372,7 +376,7 @@ def main():
                     "seq_len": None,
                     "parallelism":
                         f"{n} binderd procs behind binder-balancer",
-                    "store": "stub-zk mirror",
+                    "store": "native zkd mirror",
                     "p50_us": last["p50_us"] if last else None,
                     "p99_us": last["p99_us"] if last else None,
                     "timeouts_last_step": last["timeouts"] if last else None,

@@ -88,6 +88,18 @@ class ZkConn:
             struct.pack(">i", flags)
         self._call(OP_CREATE, body)
 
+    def put(self, path: str, data: bytes):
+        """mkdirp parents + create-or-set (stubzk.put parity)."""
+        parent = path.rsplit("/", 1)[0]
+        if parent and parent != "/":
+            self.mkdirp(parent)
+        try:
+            self.create(path, data)
+        except ZkError as e:
+            if e.code != -110:  # ZNODEEXISTS
+                raise
+            self.set(path, data)
+
     def set(self, path: str, data: bytes, version: int = -1):
         self._call(OP_SETDATA, _s(path) + _b(data) +
                    struct.pack(">i", version))
