@@ -106,6 +106,16 @@ CompiledRecord compileRecord(const Json& data) {
     if (!sub.isObject()) return out;  // record[type] must be an object
     out.valid = true;
 
+    /* Explicit resolver-registry schema (no reference counterpart —
+     * the reference only discovers resolvers via UFDS; this is the
+     * typed form of the registryDomain convention, recursion.hpp). */
+    if (out.typeName == "resolver") {
+        if (sub.get("address").isString())
+            out.address = sub.get("address").asString();
+        if (sub.get("datacenter").isString())
+            out.datacenter = sub.get("datacenter").asString();
+    }
+
     /* TTL precedence chain (deepest wins). */
     uint32_t ttl = 30;
     if (data.get("ttl").isNumber()) ttl = (uint32_t)data.get("ttl").asInt();

@@ -51,8 +51,12 @@ struct CompiledRecord {
     RecType type = RecType::Unknown;
     std::string typeName;
 
-    /* host-like + database (URL-parsed hostname): */
+    /* host-like + database (URL-parsed hostname) + resolver: */
     std::string address;    // empty = null/absent address
+    /* resolver records only ({"type":"resolver","resolver":
+     * {"datacenter":...,"address":...}}) — the explicit schema for
+     * the recursion resolver registry (recursion.hpp). */
+    std::string datacenter;
     /* TTL via the precedence chain: 30 -> record.ttl -> record[type].ttl
      * -> (service, nested) record.service.service.ttl
      * (lib/server.js:262-274, 324-332). */

@@ -103,19 +103,25 @@ void Recursion::refresh() {
             scheduleRefresh(kRetryInitMs);
             return;
         }
-        /* registry node children: each child's record is a host-like
-         * entry whose name is "<dc>-<n>" or whose record carries a
-         * "datacenter" — we accept {"type":"resolver","resolver":
-         * {"datacenter":...,"address":...}} via raw compile too. For the
-         * common case the registry node's own JSON payload is the map,
-         * so prefer children of type host with dc prefix. */
+        /* Registry children, two accepted shapes (schema documented
+         * in recursion.hpp):
+         *   1. typed (preferred): {"type":"resolver","resolver":
+         *      {"datacenter":"dc2","address":"10.0.0.5"}} — the
+         *      datacenter is explicit;
+         *   2. legacy host-like entries named "<dc>-<n>", where the
+         *      dc comes from the name prefix. */
         std::map<std::string, std::vector<std::string>> dcs;
         for (const StoreNode* kid : node->children()) {
             const CompiledRecord& r = kid->rec();
             if (!r.valid || r.address.empty()) continue;
-            std::string dc = kid->name();
-            size_t dash = dc.rfind('-');
-            if (dash != std::string::npos) dc = dc.substr(0, dash);
+            std::string dc;
+            if (!r.datacenter.empty()) {
+                dc = r.datacenter;
+            } else {
+                dc = kid->name();
+                size_t dash = dc.rfind('-');
+                if (dash != std::string::npos) dc = dc.substr(0, dash);
+            }
             auto& v = dcs[dc];
             bool dup = false;
             for (const auto& e : v) dup = dup || e == r.address;

@@ -16,8 +16,12 @@
  * Registry sources are pluggable where the reference hardcodes UFDS/LDAP
  * (its listResolvers(region) call, recursion.js:210-219):
  *   "static": {"source":"static","dcs":{"dc1":["10.0.0.5",...]}}
- *   "zk":     {"source":"zk","registryDomain":"resolvers.<domain>"} —
- *             host-like children keyed "<dc>-<n>" under that domain.
+ *   "zk":     {"source":"zk","registryDomain":"resolvers.<domain>"}.
+ *             Children of that domain are resolver entries; the typed
+ *             schema is {"type":"resolver","resolver":{"datacenter":
+ *             "<dc>","address":"<ip>"}} (explicit dc); host-like
+ *             children keyed "<dc>-<n>" are accepted as the legacy
+ *             name-prefix form.
  *   "ufds":   the reference's own path: resolve the UFDS address from
  *             the mirror (recursion.js:104-127), then LDAP-search
  *             resolvers for the region (ldap.hpp) every 5 minutes on a

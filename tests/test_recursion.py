@@ -249,3 +249,70 @@ def test_upstream_answer_type_filter(tmp_path):
         stop.set()
         t.join(timeout=3)
         usock.close()
+
+
+def test_zk_registry_typed_and_legacy_resolver_schema(tmp_path):
+    """The zk resolver-registry source accepts both shapes documented
+    in recursion.hpp: the typed {"type":"resolver","resolver":
+    {"datacenter","address"}} record (explicit dc) and the legacy
+    "<dc>-<n>" host-named form."""
+    import time
+
+    from binder_amd.stubzk import StubZk
+
+    # upstream binder answering for dc2 and dc3 names
+    up_tree = tmp_path / "up.json"
+    up_tree.write_text(json.dumps({
+        "foo.com": None,
+        "svc.dc2.foo.com": {"type": "host",
+                            "host": {"address": "10.22.0.2"}},
+        "svc.dc3.foo.com": {"type": "host",
+                            "host": {"address": "10.33.0.3"}}}))
+    upstream = BinderProcess(dns_domain="foo.com", datacenter="up",
+                             host="127.0.0.2", store=f"file:{up_tree}",
+                             workdir=tmp_path)
+    upstream.start()
+
+    zk = StubZk().start()
+    try:
+        zk.mkdirp("/com/foo/resolvers")
+        # typed form: dc comes from the record, name is arbitrary
+        zk.put("/com/foo/resolvers/anything", json.dumps(
+            {"type": "resolver",
+             "resolver": {"datacenter": "dc2",
+                          "address": "127.0.0.2"}}).encode())
+        # legacy form: dc parsed from the "<dc>-<n>" name
+        zk.put("/com/foo/resolvers/dc3-0", json.dumps(
+            {"type": "host",
+             "host": {"address": "127.0.0.2"}}).encode())
+
+        local = BinderProcess(
+            dns_domain="foo.com", datacenter="dc1",
+            store="zk", zk_host="127.0.0.1", zk_port=zk.port,
+            workdir=tmp_path, log_path=str(tmp_path / "tl.log"),
+            config={"recursion": {
+                "source": "zk", "regionName": "r1",
+                "dnsDomain": "foo.com",
+                "registryDomain": "resolvers.foo.com",
+                "upstreamPort": upstream.port,
+            }})
+        local.start()
+        try:
+            deadline = time.time() + 20
+            r = None
+            while time.time() < deadline:
+                r = local.dig("svc.dc2.foo.com", rd=True, timeout=5)
+                if r.status == "NOERROR":
+                    break
+                time.sleep(0.5)
+            assert r.status == "NOERROR", \
+                open(str(tmp_path / "tl.log")).read()[-1500:]
+            assert r.answers[0]["address"] == "10.22.0.2"
+            r3 = local.dig("svc.dc3.foo.com", rd=True, timeout=5)
+            assert r3.status == "NOERROR"
+            assert r3.answers[0]["address"] == "10.33.0.3"
+        finally:
+            local.stop()
+    finally:
+        zk.stop()
+        upstream.stop()
