@@ -281,3 +281,53 @@ def test_binderd_mirror_over_native_zkd(tmp_path):
         if b:
             b.stop()
         z.stop()
+
+
+@pytest.mark.timeout(120)
+def test_binderd_survives_zkd_restart(tmp_path):
+    """Registry crash+restart: binderd serves stale answers during the
+    outage, and when zkd comes back on the SAME data the
+    mirror rebuilds (zkd restarts have no live sessions, so clients go
+    through the expiry->fresh-session path) and new writes propagate."""
+    d = tmp_path / "data"
+    port = None
+    z = NativeZkd(data_dir=str(d)).start()
+    port = z.port
+    b = None
+    try:
+        c = ZkConn("127.0.0.1", z.port)
+        c.mkdirp("/com/foo")
+        c.create("/com/foo/web", json.dumps(
+            {"type": "host", "host": {"address": "10.0.0.9"}}).encode())
+        c.close()
+
+        b = BinderProcess(dns_domain="foo.com", store="zk",
+                          zk_host="127.0.0.1", zk_port=z.port,
+                          workdir=tmp_path,
+                          log_path=str(tmp_path / "b.log"))
+        b.start()
+        b.wait_ready("web.foo.com", timeout=20)
+
+        # hard-kill the registry
+        z.proc.kill()
+        z.proc.wait()
+        time.sleep(0.5)
+        # stale-serve during the outage
+        assert b.dig("web.foo.com").answers[0]["address"] == "10.0.0.9"
+
+        # restart on the same data and the SAME port
+        z = NativeZkd(port=port, data_dir=str(d)).start()
+        assert z.nodes_restored == 3
+        # new write must reach the (rebuilt) mirror
+        c = ZkConn("127.0.0.1", z.port)
+        c.create("/com/foo/neu", json.dumps(
+            {"type": "host",
+             "host": {"address": "10.0.0.10"}}).encode())
+        r = b.wait_ready("neu.foo.com", timeout=40)
+        assert r.answers[0]["address"] == "10.0.0.10"
+        assert b.dig("web.foo.com").answers[0]["address"] == "10.0.0.9"
+        c.close()
+    finally:
+        if b:
+            b.stop()
+        z.stop()
