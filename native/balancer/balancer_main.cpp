@@ -23,6 +23,7 @@
 #include <arpa/inet.h>
 #include <dirent.h>
 #include <netinet/in.h>
+#include <netinet/udp.h>
 #include <signal.h>
 #include <sys/epoll.h>
 #include <sys/signalfd.h>
@@ -192,6 +193,9 @@ class Balancer {
     Backend* pickBackendFast(const struct sockaddr_storage& ss);
     Backend* chooseLeastLoaded();
     void onUdpReadable();
+    void handleUdpQuery(const uint8_t* dns, size_t dnsLen,
+                        const struct sockaddr_storage& src,
+                        int64_t expiry, std::set<Backend*>& touched);
     void onTcpAccept();
     void onTcpClient(std::shared_ptr<TcpClient> c, uint32_t ev);
     void tcpClientFlush(TcpClient* c);
@@ -230,12 +234,17 @@ class Balancer {
 
     /* per-instance batch I/O arenas — workers are threads, so these
      * must NOT be static (a shared-static race here collapsed
-     * multi-worker throughput) */
+     * multi-worker throughput). RX buffers are GRO-sized: with UDP_GRO
+     * on the ingress socket one "message" may be a coalesced
+     * super-packet of up to 64 KB of equal-size datagrams. */
     static constexpr int kBatch = 128;
-    uint8_t rxBufs_[kBatch][4096];
+    static constexpr size_t kRxBufSz = 65536;
+    std::vector<uint8_t> rxArena_ =
+        std::vector<uint8_t>((size_t)kBatch * kRxBufSz);
     struct mmsghdr rxHdrs_[kBatch];
     struct iovec rxIovs_[kBatch];
     struct sockaddr_storage rxAddrs_[kBatch];
+    char rxCtrl_[kBatch][CMSG_SPACE(sizeof(uint16_t))];
     struct mmsghdr replyHdrs_[kBatch];
     struct iovec replyIovs_[kBatch];
     struct sockaddr_storage replyAddrs_[kBatch];
@@ -253,6 +262,11 @@ bool Balancer::start() {
     int sz = 8 << 20;
     setsockopt(udpFd_, SOL_SOCKET, SO_RCVBUF, &sz, sizeof(sz));
     setsockopt(udpFd_, SOL_SOCKET, SO_SNDBUF, &sz, sizeof(sz));
+    /* Accept GRO-coalesced batches: a GSO-batching client (dnsblast,
+     * or any sender using UDP_SEGMENT) then costs the kernel one
+     * loopback traversal per BURST instead of per query. Harmless if
+     * unsupported or no sender uses it. */
+    setsockopt(udpFd_, SOL_UDP, UDP_GRO, &one, sizeof(one));
     tcpFd_ = socket(v6 ? AF_INET6 : AF_INET,
                     SOCK_STREAM | SOCK_NONBLOCK | SOCK_CLOEXEC, 0);
     setsockopt(tcpFd_, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
@@ -698,96 +712,123 @@ static void ipOf(const struct sockaddr_storage& ss, char* out, size_t n,
 }
 
 void Balancer::onUdpReadable() {
-    uint8_t (*bufs)[4096] = rxBufs_;
     struct mmsghdr* hdrs = rxHdrs_;
     struct iovec* iovs = rxIovs_;
     struct sockaddr_storage* addrs = rxAddrs_;
 
     while (true) {
         for (int i = 0; i < kBatch; ++i) {
-            iovs[i] = {bufs[i], sizeof(bufs[i])};
+            iovs[i] = {rxArena_.data() + (size_t)i * kRxBufSz,
+                       kRxBufSz};
             memset(&hdrs[i], 0, sizeof(hdrs[i]));
             hdrs[i].msg_hdr.msg_iov = &iovs[i];
             hdrs[i].msg_hdr.msg_iovlen = 1;
             hdrs[i].msg_hdr.msg_name = &addrs[i];
             hdrs[i].msg_hdr.msg_namelen = sizeof(addrs[i]);
+            hdrs[i].msg_hdr.msg_control = rxCtrl_[i];
+            hdrs[i].msg_hdr.msg_controllen = sizeof(rxCtrl_[i]);
         }
         int n = recvmmsg(udpFd_, hdrs, kBatch, 0, nullptr);
         if (n <= 0) return;
         int64_t expiry = monotonicMillis() + kReplyTtlMs;
         std::set<Backend*> touched;
         for (int i = 0; i < n; ++i) {
-            Backend* be = pickBackendFast(addrs[i]);
-            udpQueries_++;
-            if (be == nullptr) {
-                drops_++;
-                continue;
+            /* GRO: one message may be several equal-size datagrams;
+             * the segment size arrives as a UDP_GRO cmsg */
+            size_t seg = hdrs[i].msg_len;
+            for (struct cmsghdr* cm = CMSG_FIRSTHDR(&hdrs[i].msg_hdr);
+                 cm != nullptr;
+                 cm = CMSG_NXTHDR(&hdrs[i].msg_hdr, cm)) {
+                if (cm->cmsg_level == SOL_UDP &&
+                    cm->cmsg_type == UDP_GRO) {
+                    uint16_t g;
+                    memcpy(&g, CMSG_DATA(cm), sizeof(g));
+                    if (g > 0) seg = g;
+                }
             }
-            /* load shedding: a backend this far behind will answer
-             * past any client deadline — drop now (clients retry)
-             * rather than queue into a multi-ms tail */
-            if (be->pendingCount > kMaxPending ||
-                be->out.size() - be->outOff > (4u << 20)) {
-                drops_++;
-                continue;
+            const uint8_t* base = rxArena_.data() + (size_t)i * kRxBufSz;
+            size_t total = hdrs[i].msg_len;
+            for (size_t off = 0; off < total; off += seg) {
+                size_t dnsLen = std::min(seg, total - off);
+                handleUdpQuery(base + off, dnsLen, addrs[i], expiry,
+                               touched);
             }
-            uint32_t reqId = be->nextReq++;
-            PendingSlot* pr = be->slotFor(reqId);
-            if (pr->reqId == 0) be->pendingCount++;
-            else be->overwrites++;
-            pr->reqId = reqId;
-            pr->expiresAtMs = (uint32_t)expiry;
-            pr->tcp = false;
-            pr->tcpFd = -1;
-            /* frame header assembled on the stack: one append for the
-             * 30-byte head + one for the DNS payload (was 7 string
-             * appends per query — 13% of user CPU in _M_append) */
-            size_t dnsLen = hdrs[i].msg_len;
-            uint32_t plen = (uint32_t)(bsock::kQueryHeadLen + dnsLen);
-            uint8_t head[bsock::kHeaderLen + bsock::kQueryHeadLen];
-            head[0] = bsock::kMagic;
-            head[1] = bsock::FRAME_QUERY;
-            /* bsock1 integers are little-endian (protocol.hpp) */
-            head[2] = (uint8_t)plen;
-            head[3] = (uint8_t)(plen >> 8);
-            head[4] = (uint8_t)(plen >> 16);
-            head[5] = (uint8_t)(plen >> 24);
-            head[6] = (uint8_t)reqId;
-            head[7] = (uint8_t)(reqId >> 8);
-            head[8] = (uint8_t)(reqId >> 16);
-            head[9] = (uint8_t)(reqId >> 24);
-            head[11] = 0;  // udp
-            if (addrs[i].ss_family == AF_INET) {
-                const auto* sa = (const struct sockaddr_in*)&addrs[i];
-                uint16_t p = ntohs(sa->sin_port);
-                head[10] = 4;
-                head[12] = (uint8_t)p;
-                head[13] = (uint8_t)(p >> 8);
-                memcpy(head + 14, &sa->sin_addr, 4);
-                memset(head + 18, 0, 12);
-                pr->family = 4;
-                pr->srcPort = p;
-                memcpy(pr->addr, &sa->sin_addr, 4);
-            } else {
-                const auto* sa = (const struct sockaddr_in6*)&addrs[i];
-                uint16_t p = ntohs(sa->sin6_port);
-                head[10] = 6;
-                head[12] = (uint8_t)p;
-                head[13] = (uint8_t)(p >> 8);
-                memcpy(head + 14, &sa->sin6_addr, 16);
-                pr->family = 6;
-                pr->srcPort = p;
-                memcpy(pr->addr, &sa->sin6_addr, 16);
-            }
-            std::string& o = be->out;
-            o.append((const char*)head, sizeof(head));
-            o.append((const char*)bufs[i], dnsLen);
-            be->queries++;
-            touched.insert(be);
         }
         for (Backend* be : touched) backendFlush(be);
         if (n < kBatch) return;
     }
+}
+
+void Balancer::handleUdpQuery(const uint8_t* dns, size_t dnsLen,
+                              const struct sockaddr_storage& src,
+                              int64_t expiry,
+                              std::set<Backend*>& touched) {
+    Backend* be = pickBackendFast(src);
+    udpQueries_++;
+    if (be == nullptr) {
+        drops_++;
+        return;
+    }
+    /* load shedding: a backend this far behind will answer past any
+     * client deadline — drop now (clients retry) rather than queue
+     * into a multi-ms tail */
+    if (be->pendingCount > kMaxPending ||
+        be->out.size() - be->outOff > (4u << 20)) {
+        drops_++;
+        return;
+    }
+    uint32_t reqId = be->nextReq++;
+    PendingSlot* pr = be->slotFor(reqId);
+    if (pr->reqId == 0) be->pendingCount++;
+    else be->overwrites++;
+    pr->reqId = reqId;
+    pr->expiresAtMs = (uint32_t)expiry;
+    pr->tcp = false;
+    pr->tcpFd = -1;
+    /* frame header assembled on the stack: one append for the 30-byte
+     * head + one for the DNS payload (was 7 string appends per query
+     * — 13% of user CPU in _M_append) */
+    uint32_t plen = (uint32_t)(bsock::kQueryHeadLen + dnsLen);
+    uint8_t head[bsock::kHeaderLen + bsock::kQueryHeadLen];
+    head[0] = bsock::kMagic;
+    head[1] = bsock::FRAME_QUERY;
+    /* bsock1 integers are little-endian (protocol.hpp) */
+    head[2] = (uint8_t)plen;
+    head[3] = (uint8_t)(plen >> 8);
+    head[4] = (uint8_t)(plen >> 16);
+    head[5] = (uint8_t)(plen >> 24);
+    head[6] = (uint8_t)reqId;
+    head[7] = (uint8_t)(reqId >> 8);
+    head[8] = (uint8_t)(reqId >> 16);
+    head[9] = (uint8_t)(reqId >> 24);
+    head[11] = 0;  // udp
+    if (src.ss_family == AF_INET) {
+        const auto* sa = (const struct sockaddr_in*)&src;
+        uint16_t p = ntohs(sa->sin_port);
+        head[10] = 4;
+        head[12] = (uint8_t)p;
+        head[13] = (uint8_t)(p >> 8);
+        memcpy(head + 14, &sa->sin_addr, 4);
+        memset(head + 18, 0, 12);
+        pr->family = 4;
+        pr->srcPort = p;
+        memcpy(pr->addr, &sa->sin_addr, 4);
+    } else {
+        const auto* sa = (const struct sockaddr_in6*)&src;
+        uint16_t p = ntohs(sa->sin6_port);
+        head[10] = 6;
+        head[12] = (uint8_t)p;
+        head[13] = (uint8_t)(p >> 8);
+        memcpy(head + 14, &sa->sin6_addr, 16);
+        pr->family = 6;
+        pr->srcPort = p;
+        memcpy(pr->addr, &sa->sin6_addr, 16);
+    }
+    std::string& o = be->out;
+    o.append((const char*)head, sizeof(head));
+    o.append((const char*)dns, dnsLen);
+    be->queries++;
+    touched.insert(be);
 }
 
 void Balancer::onTcpAccept() {

@@ -29,6 +29,7 @@
 #include <signal.h>
 #include <arpa/inet.h>
 #include <netinet/in.h>
+#include <netinet/udp.h>
 #include <poll.h>
 #include <sys/socket.h>
 #include <unistd.h>
@@ -39,6 +40,7 @@
 #include <fstream>
 #include <array>
 #include <random>
+#include <map>
 #include <sstream>
 #include <string>
 #include <thread>
@@ -104,10 +106,18 @@ struct Config {
     std::string bindBase;
     int timeoutMs = 2000;
     bool rd = false;  // set RD (exercises recursion paths)
+    bool gso = true;  // UDP_SEGMENT batching (auto-fallback)
 };
+
+/* wire indices grouped by identical wire size: a GSO super-packet
+ * must carry equal-size segments, so each burst draws all its names
+ * from ONE class (class picked per burst weighted by class size, so
+ * the long-run query mix is unchanged) */
+using SizeClasses = std::vector<std::vector<size_t>>;
 
 void worker(const Config& cfg, int tid,
             const std::vector<std::vector<uint8_t>>& wires,
+            const SizeClasses& classes,
             ThreadResult* out, std::atomic<bool>* abort) {
     const int K = cfg.socksPerThread < 1 ? 1 : cfg.socksPerThread;
     std::vector<int> fds;
@@ -186,14 +196,69 @@ void worker(const Config& cfg, int tid,
         return a < left ? a : left;
     };
 
-    /* Send queries for free slots, grouped per socket (slot % nSock)
-     * so each socket's burst goes out in one sendmmsg. */
+    /* Send queries for free slots, grouped per socket (slot % nSock).
+     * GSO path: the whole burst is equal-size segments of one
+     * super-packet => ONE trip through the kernel's send path instead
+     * of one per query (the chain is loopback-kernel bound under the
+     * bench box's CPU quota). Falls back to sendmmsg when UDP_SEGMENT
+     * is unavailable. */
+    bool gsoOk = cfg.gso;
+    std::vector<uint8_t> gsoBuf(2048 * 64);
     auto batchSend = [&](std::vector<int>& slots) {
         uint64_t budget = allowance();
         size_t used = 0;
         for (int k = 0; k < nSock && budget > 0; ++k) {
+            /* burst draws names from one size class (GSO needs equal
+             * segments); class picked by a weighted die so the
+             * aggregate mix still matches the names file */
+            const std::vector<size_t>* cls = nullptr;
+            size_t wireSize = 0;
+            if (gsoOk) {
+                size_t ni = rng() % wires.size();
+                for (const auto& c : classes)
+                    if (wires[c[0]].size() == wires[ni].size()) {
+                        cls = &c;
+                        break;
+                    }
+                wireSize = wires[(*cls)[0]].size();
+            }
             int nTx = 0;
+            size_t gsoLen = 0;
             auto flush = [&]() {
+                if (nTx == 0) return;
+                if (gsoOk) {
+                    struct msghdr mh {};
+                    struct iovec iov {gsoBuf.data(), gsoLen};
+                    char cbuf[CMSG_SPACE(sizeof(uint16_t))] = {0};
+                    mh.msg_iov = &iov;
+                    mh.msg_iovlen = 1;
+                    if (nTx > 1) {
+                        mh.msg_control = cbuf;
+                        mh.msg_controllen = sizeof(cbuf);
+                        struct cmsghdr* cm = CMSG_FIRSTHDR(&mh);
+                        cm->cmsg_level = SOL_UDP;
+                        cm->cmsg_type = UDP_SEGMENT;
+                        cm->cmsg_len = CMSG_LEN(sizeof(uint16_t));
+                        uint16_t seg = (uint16_t)wireSize;
+                        memcpy(CMSG_DATA(cm), &seg, sizeof(seg));
+                    }
+                    ssize_t rv = sendmsg(fds[k], &mh, 0);
+                    if (rv < 0 && (errno == EINVAL || errno == EIO ||
+                                   errno == ENOTSUP)) {
+                        /* kernel without UDP GSO: permanent fallback
+                         * (burst is re-sent as discrete packets) */
+                        gsoOk = false;
+                        for (int m = 0; m < nTx; ++m) {
+                            ssize_t r2 = send(
+                                fds[k], gsoBuf.data() + m * wireSize,
+                                wireSize, 0);
+                            (void)r2;
+                        }
+                    }
+                    gsoLen = 0;
+                    nTx = 0;
+                    return;
+                }
                 int done = 0;
                 while (done < nTx) {
                     int rv = sendmmsg(fds[k], txHdrs.data() + done,
@@ -203,21 +268,32 @@ void worker(const Config& cfg, int tid,
                 }
                 nTx = 0;
             };
+            /* <= 32 segments per GSO packet (kernel caps at 64) */
+            const int maxBurst = gsoOk ? 32 : kRxBatch;
             for (size_t si = 0; si < slots.size() && budget > 0; ++si) {
                 int slot = slots[si];
                 if (slot < 0 || slot % nSock != k) continue;
-                size_t ni = rng() % wires.size();
+                size_t ni = cls != nullptr
+                                ? (*cls)[rng() % cls->size()]
+                                : rng() % wires.size();
                 const auto& w = wires[ni];
                 slotSeq[slot]++;
                 uint16_t qid = (uint16_t)((slot & 0xFF) |
                                           ((slotSeq[slot] & 0xFF) << 8));
-                memcpy(txBufs[nTx].data(), w.data(), w.size());
-                txBufs[nTx][0] = (uint8_t)(qid >> 8);
-                txBufs[nTx][1] = (uint8_t)qid;
-                txIovs[nTx] = {txBufs[nTx].data(), w.size()};
-                memset(&txHdrs[nTx], 0, sizeof(txHdrs[nTx]));
-                txHdrs[nTx].msg_hdr.msg_iov = &txIovs[nTx];
-                txHdrs[nTx].msg_hdr.msg_iovlen = 1;
+                uint8_t* dst;
+                if (gsoOk) {
+                    dst = gsoBuf.data() + gsoLen;
+                    gsoLen += w.size();
+                } else {
+                    dst = txBufs[nTx].data();
+                    txIovs[nTx] = {dst, w.size()};
+                    memset(&txHdrs[nTx], 0, sizeof(txHdrs[nTx]));
+                    txHdrs[nTx].msg_hdr.msg_iov = &txIovs[nTx];
+                    txHdrs[nTx].msg_hdr.msg_iovlen = 1;
+                }
+                memcpy(dst, w.data(), w.size());
+                dst[0] = (uint8_t)(qid >> 8);
+                dst[1] = (uint8_t)qid;
                 sentAt[slot] = nowUs();
                 slotName[slot] = ni;
                 out->sent++;
@@ -225,7 +301,7 @@ void worker(const Config& cfg, int tid,
                 budget--;
                 used++;
                 slots[si] = -1;  // consumed
-                if (++nTx == kRxBatch) flush();
+                if (++nTx == maxBurst) flush();
             }
             flush();
         }
@@ -341,9 +417,10 @@ int main(int argc, char** argv) {
     signal(SIGPIPE, SIG_IGN);
     Config cfg;
     int c;
-    while ((c = getopt(argc, argv, "hs:p:n:c:t:P:r:f:B:T:R")) != -1) {
+    while ((c = getopt(argc, argv, "hs:p:n:c:t:P:r:f:B:T:Rg")) != -1) {
         switch (c) {
         case 'R': cfg.rd = true; break;
+        case 'g': cfg.gso = false; break;
         case 's': cfg.server = optarg; break;
         case 'p': cfg.port = (uint16_t)atoi(optarg); break;
         case 'n': cfg.queries = strtoull(optarg, nullptr, 10); break;
@@ -360,7 +437,8 @@ int main(int argc, char** argv) {
                     "usage: dnsblast -s server -p port -n queries "
                     "[-c window] [-t threads] [-P socks/thread] "
                     "[-r offered-qps] [-f names-file] "
-                    "[-B bind-base-ip] [-T timeout-ms] [-R]\n");
+                    "[-B bind-base-ip] [-T timeout-ms] [-R] "
+                    "[-g no-gso]\n");
             return c == 'h' ? 0 : 1;
         }
     }
@@ -396,6 +474,15 @@ int main(int argc, char** argv) {
         wires.push_back(q.encode(0));
     }
 
+    /* group wires by size for the GSO burst classes */
+    SizeClasses classes;
+    {
+        std::map<size_t, std::vector<size_t>> bySize;
+        for (size_t i = 0; i < wires.size(); ++i)
+            bySize[wires[i].size()].push_back(i);
+        for (auto& [sz, v] : bySize) classes.push_back(std::move(v));
+    }
+
     std::vector<ThreadResult> results(cfg.threads);
     std::atomic<bool> abort{false};
     uint64_t perThread = cfg.queries / cfg.threads;
@@ -405,9 +492,10 @@ int main(int argc, char** argv) {
     for (int i = 0; i < cfg.threads; ++i) {
         Config tcfg = cfg;
         tcfg.queries = perThread;
-        threads.emplace_back([tcfg, i, &wires, &results, &abort]() {
-            worker(tcfg, i, wires, &results[i], &abort);
-        });
+        threads.emplace_back(
+            [tcfg, i, &wires, &classes, &results, &abort]() {
+                worker(tcfg, i, wires, classes, &results[i], &abort);
+            });
     }
     for (auto& t : threads) t.join();
     int64_t elapsedUs = nowUs() - t0;
