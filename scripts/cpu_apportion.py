@@ -9,6 +9,7 @@ usage: cpu_apportion.py [--procs 8] [--workers 8] [--threads 6]
 """
 import argparse
 import json
+import socket
 import subprocess
 import sys
 import tempfile
@@ -38,6 +39,7 @@ def main():
     ap.add_argument("--workers", type=int, default=8)
     ap.add_argument("--threads", type=int, default=6)
     ap.add_argument("--seconds", type=float, default=6.0)
+    ap.add_argument("--no-gso", action="store_true")
     args = ap.parse_args()
 
     from binder_amd.harness import free_port, NativeZkd
@@ -54,12 +56,19 @@ def main():
                                workers=args.workers)
     bench.wait_balancer_ready(port, args.procs, tmp)
 
-    blast = subprocess.Popen(
-        [str(REPO / "bin" / "dnsblast"), "-s", "127.0.0.1",
-         "-p", str(port), "-n", "100000000", "-c", "256",
-         "-t", str(args.threads), "-P", "8", "-f", str(names_file),
-         "-B", "127.0.1.1", "-T", "10000"],
-        stdout=subprocess.DEVNULL)
+    cmd = [str(REPO / "bin" / "dnsblast"), "-s", "127.0.0.1",
+           "-p", str(port), "-n", "100000000", "-c", "256",
+           "-t", str(args.threads), "-P", "8", "-f", str(names_file),
+           "-B", "127.0.1.1", "-T", "10000"]
+    if args.no_gso:
+        cmd.append("-g")
+    blast = subprocess.Popen(cmd, stdout=subprocess.DEVNULL)
+
+    def balstat():
+        with socket.socket(socket.AF_UNIX) as s:
+            s.settimeout(2)
+            s.connect(str(tmp / "stats.sock"))
+            return json.loads(s.recv(1 << 20).decode())
     try:
         time.sleep(2.0)  # ramp
         pids = {"dnsblast": [blast.pid], "balancer": [bal.pid],
@@ -67,11 +76,16 @@ def main():
                 "binderd": [b.proc.pid for b in backends]}
         t0 = time.time()
         before = {k: [cpu_of(p) for p in v] for k, v in pids.items()}
+        st0 = balstat()
         time.sleep(args.seconds)
         dt = time.time() - t0
         after = {k: [cpu_of(p) for p in v] for k, v in pids.items()}
+        st1 = balstat()
         out = {"window_s": round(dt, 2), "procs": args.procs,
-               "workers": args.workers, "threads": args.threads}
+               "workers": args.workers, "threads": args.threads,
+               "gso": not args.no_gso,
+               "qps": round((st1["udp_replies"] -
+                             st0["udp_replies"]) / dt)}
         total = 0.0
         for k in pids:
             cores = sum((a - b) for a, b in
