@@ -89,7 +89,7 @@ def main():
             return
         # quota-aware sweep: restart the balancer per worker count
         nonlocal_best = {"qps": 0}
-        for w in (6, 8, 12):
+        for w in (12, 16, 20, 24):
             bal.terminate()
             bal.wait(timeout=5)
             port2 = free_port()
@@ -113,7 +113,7 @@ def main():
                 return r
 
             run2(f"w{w}-ramp", 8, 256, 8, 0)
-            for t in (6, 8, 10):
+            for t in (10, 12, 14):
                 r = run2(f"w{w}-t{t}", t, 256, 8, 0)
                 if r["qps"] > nonlocal_best["qps"]:
                     nonlocal_best = {"qps": r["qps"], "w": w, "t": t}
