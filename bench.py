@@ -222,10 +222,12 @@ def main():
     # class (16-core quota; scripts/paced_probe.py --sweep, committed
     # under profiles/).
     if ncpu >= 12:
-        # probe-swept best on the 16-core-quota box class: w8/t6 won
-        # at every N (profiles/, sweep2_n*.jsonl)
-        workers = max(2, min(8, ncpu // 2))            # 8 @ 16 cpus
-        threads = max(4, min(6, (ncpu * 3) // 8))      # 6 @ 16 cpus
+        # probe-swept best on the 16-core-quota box class with the
+        # GSO/GRO ingress (profiles/, sweep3/sweep4_n8.jsonl): the
+        # surface is flat at ~4.3M across w12-24 x t10-14; w12/t10 is
+        # the lowest-footprint point on the plateau
+        workers = max(2, min(12, (ncpu * 3) // 4))     # 12 @ 16 cpus
+        threads = max(4, min(10, (ncpu * 5) // 8))     # 10 @ 16 cpus
         window = 256
         socks = 8  # flows = threads*socks >> balancer reuseport shards
         # experiment overrides (profiling/tuning only)
