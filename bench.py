@@ -228,7 +228,10 @@ def main():
         # the lowest-footprint point on the plateau
         workers = max(2, min(12, (ncpu * 3) // 4))     # 12 @ 16 cpus
         threads = max(4, min(10, (ncpu * 5) // 8))     # 10 @ 16 cpus
-        window = 256
+        # deep windows so the paced generator can absorb RTT
+        # excursions + sleep jitter without delivery deficit (the
+        # calibration requires >=97% delivery)
+        window = 512
         socks = 8  # flows = threads*socks >> balancer reuseport shards
         # experiment overrides (profiling/tuning only)
         workers = int(os.environ.get("BENCH_WORKERS", workers))

@@ -151,7 +151,14 @@ void worker(const Config& cfg, int tid,
     if (fds.empty()) return;
     const int nSock = (int)fds.size();
 
-    const int W = cfg.window;
+    /* qid layout: slot in the low bits (window-sized), per-slot
+     * sequence in the rest — W up to 1024 with >=6 stale-check bits */
+    int W = cfg.window;
+    if (W > 1024) W = 1024;
+    int slotBits = 1;
+    while ((1 << slotBits) < W) slotBits++;
+    const uint16_t slotMask = (uint16_t)((1 << slotBits) - 1);
+    const uint16_t seqMask = (uint16_t)(0xFFFF >> slotBits);
     std::vector<int64_t> sentAt(W, 0);       // 0 = slot idle
     std::vector<uint16_t> slotSeq(W, 0);
     std::vector<size_t> slotName(W, 0);
@@ -278,8 +285,9 @@ void worker(const Config& cfg, int tid,
                                 : rng() % wires.size();
                 const auto& w = wires[ni];
                 slotSeq[slot]++;
-                uint16_t qid = (uint16_t)((slot & 0xFF) |
-                                          ((slotSeq[slot] & 0xFF) << 8));
+                uint16_t qid = (uint16_t)(
+                    (slot & slotMask) |
+                    ((slotSeq[slot] & seqMask) << slotBits));
                 uint8_t* dst;
                 if (gsoOk) {
                     dst = gsoBuf.data() + gsoLen;
@@ -366,10 +374,10 @@ void worker(const Config& cfg, int tid,
                         const uint8_t* rb = rxBufs[i].data();
                         if (rxHdrs[i].msg_len < 12) continue;
                         uint16_t qid = (uint16_t)((rb[0] << 8) | rb[1]);
-                        int slot = qid & 0xFF;
-                        uint8_t seq = (uint8_t)(qid >> 8);
+                        int slot = qid & slotMask;
+                        uint16_t seq = (uint16_t)(qid >> slotBits);
                         if (slot >= W || sentAt[slot] == 0 ||
-                            (uint8_t)(slotSeq[slot] & 0xFF) != seq)
+                            (uint16_t)(slotSeq[slot] & seqMask) != seq)
                             continue;  // stale/duplicate
                         int64_t lat = now - sentAt[slot];
                         out->latBuckets[latBucket2(lat)]++;
