@@ -186,9 +186,10 @@ void worker(const Config& cfg, int tid,
     freeSlots.reserve(W);
 
     /* How many more queries may launch right now (rate pacing).
-     * Launches are quantized (min 8 or 100 us worth) so the spinning
-     * paced loop amortizes sendmmsg instead of degenerating into one
-     * syscall per packet at high per-thread rates. */
+     * Launches are quantized to full GSO bursts (32 segments, or
+     * 300 us of tokens at low rates) — smaller quanta shrink the
+     * super-packets and the per-packet kernel cost comes back. */
+    constexpr uint64_t kQuantum = 32;
     int64_t lastLaunchUs = 0;
     auto allowance = [&]() -> uint64_t {
         if (launched >= target) return 0;
@@ -198,7 +199,8 @@ void worker(const Config& cfg, int tid,
         uint64_t paced = (uint64_t)((double)(now - tStart) * rate / 1e6);
         if (paced <= launched) return 0;
         uint64_t a = paced - launched;
-        if (a < left && a < 8 && now - lastLaunchUs < 100) return 0;
+        if (a < left && a < kQuantum && now - lastLaunchUs < 300)
+            return 0;
         lastLaunchUs = now;
         return a < left ? a : left;
     };
@@ -345,10 +347,10 @@ void worker(const Config& cfg, int tid,
         if (rate > 0) {
             uint64_t backlog = allowance();
             long waitNs;
-            if (backlog >= 8) {
+            if (backlog >= 32) {
                 waitNs = 0;
             } else {
-                double need = (double)(8 - backlog) / rate * 1e9;
+                double need = (double)(32 - backlog) / rate * 1e9;
                 waitNs = need < 250000 ? (long)need : 250000;
             }
             struct timespec ts {0, waitNs};
