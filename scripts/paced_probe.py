@@ -19,6 +19,17 @@ sys.path.insert(0, str(REPO))
 import bench  # noqa: E402  (reuse the harness pieces)
 
 
+def cpustat():
+    try:
+        d = {}
+        for line in open("/sys/fs/cgroup/cpu.stat"):
+            k, v = line.split()
+            d[k] = int(v)
+        return d
+    except OSError:
+        return {}
+
+
 def balstat(path):
     with socket.socket(socket.AF_UNIX) as s:
         s.settimeout(2)
@@ -89,7 +100,7 @@ def main():
             return
         # quota-aware sweep: restart the balancer per worker count
         nonlocal_best = {"qps": 0}
-        for w in (12, 16, 20, 24):
+        for w in (8, 12):
             bal.terminate()
             bal.wait(timeout=5)
             port2 = free_port()
@@ -100,23 +111,33 @@ def main():
 
             def run2(tag, threads, window, socks, rate):
                 b0 = balstat(stats_path)
+                c0 = cpustat()
                 r = bench.run_blast(port, q, names_file, threads,
                                     window, socks=socks, rate=rate)
+                c1 = cpustat()
                 b1 = balstat(stats_path)
                 out = {"tag": tag, "workers": w, "threads": threads,
                        "window": window, "socks": socks, "rate": rate,
                        "qps": r["qps"], "p50_us": r["p50_us"],
                        "p90_us": r["p90_us"], "p99_us": r["p99_us"],
                        "timeouts": r["timeouts"],
+                       "throttled": c1.get("nr_throttled", 0) -
+                       c0.get("nr_throttled", 0),
+                       "throttled_ms": (c1.get("throttled_usec", 0) -
+                                        c0.get("throttled_usec", 0))
+                       // 1000,
                        "bal": stat_delta(b0, b1)}
                 print(json.dumps(out), flush=True)
                 return r
 
-            run2(f"w{w}-ramp", 8, 256, 8, 0)
-            for t in (10, 12, 14):
-                r = run2(f"w{w}-t{t}", t, 256, 8, 0)
+            run2(f"w{w}-ramp", 10, 512, 8, 0)
+            cap = run2(f"w{w}-closed", 10, 512, 8, 0)["qps"]
+            for frac in (0.55, 0.75, 0.9):
+                r = run2(f"w{w}-paced{frac}", 10, 512, 8,
+                         int(cap * frac))
                 if r["qps"] > nonlocal_best["qps"]:
-                    nonlocal_best = {"qps": r["qps"], "w": w, "t": t}
+                    nonlocal_best = {"qps": r["qps"], "w": w,
+                                     "frac": frac}
         print(json.dumps({"best": nonlocal_best}), flush=True)
     finally:
         bal.terminate()
