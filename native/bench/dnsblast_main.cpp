@@ -40,6 +40,7 @@
 #include <fstream>
 #include <array>
 #include <random>
+#include <algorithm>
 #include <map>
 #include <sstream>
 #include <string>
@@ -61,6 +62,10 @@ struct NameEntry {
 constexpr int kLatBuckets = 512;
 
 struct ThreadResult {
+    int64_t tSetupUs = 0;   // thread start -> sockets ready
+    int64_t tFirstUs = 0;   // thread start -> first launch
+    int64_t tLastUs = 0;    // thread start -> last launch
+    int64_t tDoneUs = 0;    // thread start -> completion
     uint64_t sent = 0;
     uint64_t received = 0;
     uint64_t timeouts = 0;
@@ -119,6 +124,7 @@ void worker(const Config& cfg, int tid,
             const std::vector<std::vector<uint8_t>>& wires,
             const SizeClasses& classes,
             ThreadResult* out, std::atomic<bool>* abort) {
+    const int64_t tEnter = nowUs();
     const int K = cfg.socksPerThread < 1 ? 1 : cfg.socksPerThread;
     std::vector<int> fds;
     for (int k = 0; k < K; ++k) {
@@ -172,6 +178,7 @@ void worker(const Config& cfg, int tid,
                             ? cfg.rateQps / (double)cfg.threads
                             : 0;
     const int64_t tStart = nowUs();
+    out->tSetupUs = tStart - tEnter;
 
     /* batched RX (recvmmsg) and TX (sendmmsg): syscall count, not
      * packet handling, bounds the generator at high QPS */
@@ -305,6 +312,9 @@ void worker(const Config& cfg, int tid,
                 dst[0] = (uint8_t)(qid >> 8);
                 dst[1] = (uint8_t)qid;
                 sentAt[slot] = nowUs();
+                if (out->tFirstUs == 0)
+                    out->tFirstUs = sentAt[slot] - tEnter;
+                out->tLastUs = sentAt[slot] - tEnter;
                 slotName[slot] = ni;
                 out->sent++;
                 launched++;
@@ -416,6 +426,7 @@ void worker(const Config& cfg, int tid,
             else freeSlots.clear();
         }
     }
+    out->tDoneUs = nowUs() - tEnter;
     for (int fd : fds) close(fd);
 }
 
@@ -533,6 +544,19 @@ int main(int argc, char** argv) {
         }
         return bucketMidUs(kLatBuckets - 1);
     };
+
+    int64_t maxSetup = 0, maxFirst = 0, maxLast = 0, maxDone = 0;
+    for (const auto& r : results) {
+        maxSetup = std::max(maxSetup, r.tSetupUs);
+        maxFirst = std::max(maxFirst, r.tFirstUs);
+        maxLast = std::max(maxLast, r.tLastUs);
+        maxDone = std::max(maxDone, r.tDoneUs);
+    }
+    fprintf(stderr,
+            "# phases(max over threads, ms): setup %.1f first %.1f "
+            "last %.1f done %.1f wall %.1f\n",
+            maxSetup / 1e3, maxFirst / 1e3, maxLast / 1e3,
+            maxDone / 1e3, elapsedUs / 1e3);
 
     double secs = (double)elapsedUs / 1e6;
     double qps = secs > 0 ? (double)total.received / secs : 0;
