@@ -150,6 +150,8 @@ def run_blast(port, queries, names_file, threads, window, socks=8,
     if rate > 0:
         cmd += ["-r", str(int(rate))]
     out = subprocess.run(cmd, capture_output=True, text=True, check=True)
+    if out.stderr:
+        sys.stderr.write(out.stderr)  # phase diagnostics from dnsblast
     return json.loads(out.stdout.strip().splitlines()[-1])
 
 
