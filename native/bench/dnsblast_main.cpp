@@ -44,6 +44,7 @@
 #include <map>
 #include <sstream>
 #include <string>
+#include <chrono>
 #include <thread>
 #include <vector>
 
@@ -123,7 +124,8 @@ using SizeClasses = std::vector<std::vector<size_t>>;
 void worker(const Config& cfg, int tid,
             const std::vector<std::vector<uint8_t>>& wires,
             const SizeClasses& classes,
-            ThreadResult* out, std::atomic<bool>* abort) {
+            ThreadResult* out, std::atomic<bool>* abort,
+            std::atomic<int>* ready, std::atomic<bool>* go) {
     const int64_t tEnter = nowUs();
     const int K = cfg.socksPerThread < 1 ? 1 : cfg.socksPerThread;
     std::vector<int> fds;
@@ -177,8 +179,16 @@ void worker(const Config& cfg, int tid,
     const double rate = cfg.rateQps > 0
                             ? cfg.rateQps / (double)cfg.threads
                             : 0;
+    /* Start barrier: setup (sockets, buffers) measured at 100-170 ms
+     * on the bench boxes; without the barrier it lands inside the
+     * measured window and reads as a throughput deficit. All threads
+     * report ready, the main thread starts the clock, then everyone
+     * begins pacing together. */
+    out->tSetupUs = nowUs() - tEnter;
+    ready->fetch_add(1);
+    while (!go->load(std::memory_order_acquire) && !abort->load())
+        std::this_thread::sleep_for(std::chrono::microseconds(100));
     const int64_t tStart = nowUs();
-    out->tSetupUs = tStart - tEnter;
 
     /* batched RX (recvmmsg) and TX (sendmmsg): syscall count, not
      * packet handling, bounds the generator at high QPS */
@@ -506,18 +516,26 @@ int main(int argc, char** argv) {
 
     std::vector<ThreadResult> results(cfg.threads);
     std::atomic<bool> abort{false};
+    std::atomic<int> ready{0};
+    std::atomic<bool> go{false};
     uint64_t perThread = cfg.queries / cfg.threads;
 
-    int64_t t0 = nowUs();
     std::vector<std::thread> threads;
     for (int i = 0; i < cfg.threads; ++i) {
         Config tcfg = cfg;
         tcfg.queries = perThread;
         threads.emplace_back(
-            [tcfg, i, &wires, &classes, &results, &abort]() {
-                worker(tcfg, i, wires, classes, &results[i], &abort);
+            [tcfg, i, &wires, &classes, &results, &abort, &ready,
+             &go]() {
+                worker(tcfg, i, wires, classes, &results[i], &abort,
+                       &ready, &go);
             });
     }
+    /* start the clock only when every thread is set up */
+    while (ready.load() < cfg.threads)
+        std::this_thread::sleep_for(std::chrono::milliseconds(1));
+    int64_t t0 = nowUs();
+    go.store(true, std::memory_order_release);
     for (auto& t : threads) t.join();
     int64_t elapsedUs = nowUs() - t0;
 
