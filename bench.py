@@ -151,12 +151,42 @@ def run_blast(port, queries, names_file, threads, window, socks=8,
         cmd += ["-r", str(int(rate))]
     out = subprocess.run(cmd, capture_output=True, text=True, check=True)
     if out.stderr:
-        sys.stderr.write(out.stderr)  # phase diagnostics from dnsblast
+        sys.stderr.write(out.stderr)  # diagnostics from dnsblast
     return json.loads(out.stdout.strip().splitlines()[-1])
 
 
-def calibrate_rate(port, names_file, threads, window, socks, capacity,
-                   slo_us):
+class BlastDaemon:
+    """Persistent dnsblast (-D): the generator's threads/sockets/wires
+    survive across steps, so the timed region contains only query
+    work (process + socket setup measured 100-400 ms per invocation
+    on the bench boxes)."""
+
+    def __init__(self, port, names_file, threads, window, socks):
+        self.proc = subprocess.Popen(
+            [str(REPO / "bin" / "dnsblast"), "-D", "-s", "127.0.0.1",
+             "-p", str(port), "-c", str(window), "-t", str(threads),
+             "-P", str(socks), "-f", str(names_file),
+             "-B", "127.0.1.1", "-T", "10000"],
+            stdin=subprocess.PIPE, stdout=subprocess.PIPE, text=True,
+            bufsize=1)
+
+    def step(self, queries, rate=0):
+        self.proc.stdin.write(f"RUN {int(queries)} {int(rate)}\n")
+        self.proc.stdin.flush()
+        line = self.proc.stdout.readline()
+        if not line:
+            raise RuntimeError("dnsblast daemon died")
+        return json.loads(line)
+
+    def close(self):
+        try:
+            self.proc.stdin.close()
+            self.proc.wait(timeout=10)
+        except (OSError, subprocess.TimeoutExpired):
+            self.proc.kill()
+
+
+def calibrate_rate(blast, capacity, slo_us):
     """qps@SLO discovery (untimed): walk a descending rate ladder from
     the closed-loop capacity; the first offered rate that sustains
     p99 <= SLO with zero timeouts and full delivery is the operating
@@ -166,8 +196,7 @@ def calibrate_rate(port, names_file, threads, window, socks, capacity,
     for frac in (1.0, 0.95, 0.9, 0.85, 0.8, 0.72, 0.64, 0.55):
         rate = capacity * frac
         q = max(200_000, int(rate * 1.2))
-        r = run_blast(port, q, names_file, threads, window, socks,
-                      rate=rate)
+        r = blast.step(q, rate=rate)
         ok = (r["timeouts"] == 0 and r["p99_us"] <= slo_us and
               r["qps"] >= 0.97 * rate)
         log(f"calibrate {frac:.2f}x: offered {rate:.0f} -> "
@@ -285,22 +314,22 @@ def main():
             # measured +11% at N=8 vs a single warmup step
             log(f"balancer ready on :{bal_port}; ramp + warmup "
                 f"{args.warmup} x {q_step} queries")
+            blast = BlastDaemon(bal_port, names_file, threads,
+                                window, socks)
+            stack.append(blast.close)
             capacity = 0.0
             for _ in range(2):
-                r = run_blast(bal_port, q_step, names_file, threads,
-                              window, socks)
+                r = blast.step(q_step)
                 capacity = max(capacity, r["qps"])
             rate = 0
             if not closed_loop:
                 # qps@SLO protocol: fixed offered rate for the timed
                 # steps, discovered against the SLO (untimed)
-                rate = calibrate_rate(bal_port, names_file, threads,
-                                      window, socks, capacity, slo_us)
+                rate = calibrate_rate(blast, capacity, slo_us)
                 log(f"operating point: {rate:.0f} qps offered "
                     f"(capacity {capacity:.0f}, SLO p99<={slo_us}us)")
             for _ in range(args.warmup):
-                run_blast(bal_port, q_step, names_file, threads, window,
-                          socks, rate=rate)
+                blast.step(q_step, rate=rate)
 
         churn_stop = None
         if rank == 0 and args.churn_qps > 0:
@@ -335,8 +364,7 @@ def main():
         last = None
         if rank == 0:
             for s in range(args.steps):
-                last = run_blast(bal_port, q_step, names_file, threads,
-                                 window, socks, rate=rate)
+                last = blast.step(q_step, rate=rate)
                 log(f"step {s + 1}/{args.steps}: "
                     f"{last['qps']:.0f} qps, p99 {last['p99_us']}us")
         barrier()

@@ -6,22 +6,30 @@
  * spread over K sockets (-P), cycling through a name list loaded from
  * a file. Query wire images are prebuilt; per-send we patch only the
  * DNS id. Latencies land in log-spaced microsecond buckets for
- * p50/p99 extraction. Output: ONE JSON line on stdout.
+ * p50/p99 extraction. Output: ONE JSON line per run on stdout.
  *
- * Multiple sockets per thread matter against a SO_REUSEPORT-sharded
- * balancer: the kernel hashes FLOWS (4-tuples) across shards, so the
- * flow count must comfortably exceed the shard count or shard load is
- * balls-in-bins uneven and the busiest shard sets the ceiling.
+ * Design notes (each measured on the quota-capped bench boxes;
+ * profiles/SCALING.md round 2):
+ *  - multiple sockets per thread (-P): the balancer's SO_REUSEPORT
+ *    shards hash FLOWS; the flow count must comfortably exceed the
+ *    shard count or shard load is balls-in-bins uneven;
+ *  - every burst is one UDP_SEGMENT super-packet (equal segment sizes
+ *    via per-burst wire size classes, class picked by a weighted die
+ *    so the aggregate mix is unchanged) - one kernel loopback
+ *    traversal per burst instead of per query; -g disables;
+ *  - -r <qps> paces a fixed offered rate (32-packet token quanta,
+ *    sub-ms sleeps): the "qps at SLO" protocol, far less noisy than
+ *    the chaotic closed-loop saturation equilibrium;
+ *  - -D daemon mode keeps the generator (threads, sockets, wires)
+ *    alive across runs, reading "RUN <queries> <rate>" lines on
+ *    stdin and answering one JSON line each: thread/socket setup is
+ *    100-170 ms on the bench boxes and must not be charged to the
+ *    timed steps; a start barrier excludes it in one-shot mode too.
  *
- * -r <qps> switches from closed-loop saturation to a fixed offered
- * rate (token-paced per thread): the measurement protocol is then
- * "qps at SLO" — the caller binary-searches the max rate whose p99
- * stays under its SLO, which is far less noisy step-to-step than
- * running at the chaotic saturation equilibrium.
- *
- * usage: dnsblast -s <server-ip> -p <port> -n <queries> [-c window]
+ * usage: dnsblast -s <server-ip> -p <port> [-n <queries>] [-c window]
  *        [-t threads] [-P socks/thread] [-r offered-qps]
- *        [-f names-file] [-B bind-ip-base] [-T timeout-ms] [-R]
+ *        [-f names-file] [-B bind-ip-base] [-T timeout-ms] [-R] [-g]
+ *        [-D]
  *   names-file: lines of "<name> <qtype>"; default a single test name.
  *   -B 127.0.0.x base: thread i binds source ip base+i (gives the
  *      balancer distinct remotes so per-IP affinity spreads load).
@@ -34,17 +42,17 @@
 #include <sys/socket.h>
 #include <unistd.h>
 
+#include <algorithm>
+#include <array>
 #include <atomic>
+#include <chrono>
 #include <cstdio>
 #include <cstring>
 #include <fstream>
-#include <array>
-#include <random>
-#include <algorithm>
 #include <map>
+#include <random>
 #include <sstream>
 #include <string>
-#include <chrono>
 #include <thread>
 #include <vector>
 
@@ -63,10 +71,6 @@ struct NameEntry {
 constexpr int kLatBuckets = 512;
 
 struct ThreadResult {
-    int64_t tSetupUs = 0;   // thread start -> sockets ready
-    int64_t tFirstUs = 0;   // thread start -> first launch
-    int64_t tLastUs = 0;    // thread start -> last launch
-    int64_t tDoneUs = 0;    // thread start -> completion
     uint64_t sent = 0;
     uint64_t received = 0;
     uint64_t timeouts = 0;
@@ -74,6 +78,12 @@ struct ThreadResult {
     uint64_t rcodeOther = 0;
     uint64_t answers = 0;
     std::vector<uint64_t> latBuckets = std::vector<uint64_t>(kLatBuckets);
+
+    void reset() {
+        sent = received = timeouts = 0;
+        rcodeNoerror = rcodeOther = answers = 0;
+        std::fill(latBuckets.begin(), latBuckets.end(), 0);
+    }
 };
 
 /* latency mapping: buckets 0..255: 1us each; 256..495: 16us each
@@ -111,22 +121,29 @@ struct Config {
     std::string namesFile;
     std::string bindBase;
     int timeoutMs = 2000;
-    bool rd = false;  // set RD (exercises recursion paths)
-    bool gso = true;  // UDP_SEGMENT batching (auto-fallback)
+    bool rd = false;   // set RD (exercises recursion paths)
+    bool gso = true;   // UDP_SEGMENT batching (auto-fallback)
+    bool daemon = false;
 };
 
-/* wire indices grouped by identical wire size: a GSO super-packet
- * must carry equal-size segments, so each burst draws all its names
- * from ONE class (class picked per burst weighted by class size, so
- * the long-run query mix is unchanged) */
+/* wire indices grouped by identical wire size (GSO super-packets
+ * need equal-size segments) */
 using SizeClasses = std::vector<std::vector<size_t>>;
+
+/* run control shared between the command loop and the workers */
+struct RunCtl {
+    std::atomic<uint64_t> gen{0};      // bumped per RUN
+    std::atomic<int> setup{0};         // workers finished setup
+    std::atomic<int> running{0};       // workers still in a run
+    std::atomic<bool> shutdown{false};
+    uint64_t queriesPerThread = 0;
+    double rateQps = 0;                // total offered rate
+};
 
 void worker(const Config& cfg, int tid,
             const std::vector<std::vector<uint8_t>>& wires,
-            const SizeClasses& classes,
-            ThreadResult* out, std::atomic<bool>* abort,
-            std::atomic<int>* ready, std::atomic<bool>* go) {
-    const int64_t tEnter = nowUs();
+            const SizeClasses& classes, ThreadResult* out,
+            RunCtl* ctl) {
     const int K = cfg.socksPerThread < 1 ? 1 : cfg.socksPerThread;
     std::vector<int> fds;
     for (int k = 0; k < K; ++k) {
@@ -156,11 +173,14 @@ void worker(const Config& cfg, int tid,
         }
         fds.push_back(fd);
     }
-    if (fds.empty()) return;
+    if (fds.empty()) {
+        ctl->setup.fetch_add(1);
+        return;
+    }
     const int nSock = (int)fds.size();
 
     /* qid layout: slot in the low bits (window-sized), per-slot
-     * sequence in the rest — W up to 1024 with >=6 stale-check bits */
+     * sequence in the rest - W up to 1024 with >=6 stale-check bits */
     int W = cfg.window;
     if (W > 1024) W = 1024;
     int slotBits = 1;
@@ -169,28 +189,10 @@ void worker(const Config& cfg, int tid,
     const uint16_t seqMask = (uint16_t)(0xFFFF >> slotBits);
     std::vector<int64_t> sentAt(W, 0);       // 0 = slot idle
     std::vector<uint16_t> slotSeq(W, 0);
-    std::vector<size_t> slotName(W, 0);
     std::mt19937 rng(12345 + tid);
-
-    uint64_t target = cfg.queries;
-    uint64_t launched = 0, completed = 0;
     const int64_t timeoutUs = (int64_t)cfg.timeoutMs * 1000;
-    /* fixed-rate pacing: this thread's share of the offered rate */
-    const double rate = cfg.rateQps > 0
-                            ? cfg.rateQps / (double)cfg.threads
-                            : 0;
-    /* Start barrier: setup (sockets, buffers) measured at 100-170 ms
-     * on the bench boxes; without the barrier it lands inside the
-     * measured window and reads as a throughput deficit. All threads
-     * report ready, the main thread starts the clock, then everyone
-     * begins pacing together. */
-    out->tSetupUs = nowUs() - tEnter;
-    ready->fetch_add(1);
-    while (!go->load(std::memory_order_acquire) && !abort->load())
-        std::this_thread::sleep_for(std::chrono::microseconds(100));
-    const int64_t tStart = nowUs();
 
-    /* batched RX (recvmmsg) and TX (sendmmsg): syscall count, not
+    /* batched RX (recvmmsg) and TX (sendmmsg/GSO): syscall count, not
      * packet handling, bounds the generator at high QPS */
     constexpr int kRxBatch = 64;
     std::vector<std::array<uint8_t, 2048>> rxBufs(kRxBatch);
@@ -201,242 +203,270 @@ void worker(const Config& cfg, int tid,
     std::vector<struct iovec> txIovs(kRxBatch);
     std::vector<int> freeSlots;
     freeSlots.reserve(W);
-
-    /* How many more queries may launch right now (rate pacing).
-     * Launches are quantized to full GSO bursts (32 segments, or
-     * 300 us of tokens at low rates) — smaller quanta shrink the
-     * super-packets and the per-packet kernel cost comes back. */
-    constexpr uint64_t kQuantum = 32;
-    int64_t lastLaunchUs = 0;
-    auto allowance = [&]() -> uint64_t {
-        if (launched >= target) return 0;
-        uint64_t left = target - launched;
-        if (rate <= 0) return left;
-        int64_t now = nowUs();
-        uint64_t paced = (uint64_t)((double)(now - tStart) * rate / 1e6);
-        if (paced <= launched) return 0;
-        uint64_t a = paced - launched;
-        if (a < left && a < kQuantum && now - lastLaunchUs < 300)
-            return 0;
-        lastLaunchUs = now;
-        return a < left ? a : left;
-    };
-
-    /* Send queries for free slots, grouped per socket (slot % nSock).
-     * GSO path: the whole burst is equal-size segments of one
-     * super-packet => ONE trip through the kernel's send path instead
-     * of one per query (the chain is loopback-kernel bound under the
-     * bench box's CPU quota). Falls back to sendmmsg when UDP_SEGMENT
-     * is unavailable. */
     bool gsoOk = cfg.gso;
     std::vector<uint8_t> gsoBuf(2048 * 64);
-    auto batchSend = [&](std::vector<int>& slots) {
-        uint64_t budget = allowance();
-        size_t used = 0;
-        for (int k = 0; k < nSock && budget > 0; ++k) {
-            /* burst draws names from one size class (GSO needs equal
-             * segments); class picked by a weighted die so the
-             * aggregate mix still matches the names file */
-            const std::vector<size_t>* cls = nullptr;
-            size_t wireSize = 0;
-            if (gsoOk) {
-                size_t ni = rng() % wires.size();
-                for (const auto& c : classes)
-                    if (wires[c[0]].size() == wires[ni].size()) {
-                        cls = &c;
-                        break;
-                    }
-                wireSize = wires[(*cls)[0]].size();
-            }
-            int nTx = 0;
-            size_t gsoLen = 0;
-            auto flush = [&]() {
-                if (nTx == 0) return;
-                if (gsoOk) {
-                    struct msghdr mh {};
-                    struct iovec iov {gsoBuf.data(), gsoLen};
-                    char cbuf[CMSG_SPACE(sizeof(uint16_t))] = {0};
-                    mh.msg_iov = &iov;
-                    mh.msg_iovlen = 1;
-                    if (nTx > 1) {
-                        mh.msg_control = cbuf;
-                        mh.msg_controllen = sizeof(cbuf);
-                        struct cmsghdr* cm = CMSG_FIRSTHDR(&mh);
-                        cm->cmsg_level = SOL_UDP;
-                        cm->cmsg_type = UDP_SEGMENT;
-                        cm->cmsg_len = CMSG_LEN(sizeof(uint16_t));
-                        uint16_t seg = (uint16_t)wireSize;
-                        memcpy(CMSG_DATA(cm), &seg, sizeof(seg));
-                    }
-                    ssize_t rv = sendmsg(fds[k], &mh, 0);
-                    if (rv < 0 && (errno == EINVAL || errno == EIO ||
-                                   errno == ENOTSUP)) {
-                        /* kernel without UDP GSO: permanent fallback
-                         * (burst is re-sent as discrete packets) */
-                        gsoOk = false;
-                        for (int m = 0; m < nTx; ++m) {
-                            ssize_t r2 = send(
-                                fds[k], gsoBuf.data() + m * wireSize,
-                                wireSize, 0);
-                            (void)r2;
-                        }
-                    }
-                    gsoLen = 0;
-                    nTx = 0;
-                    return;
-                }
-                int done = 0;
-                while (done < nTx) {
-                    int rv = sendmmsg(fds[k], txHdrs.data() + done,
-                                      nTx - done, 0);
-                    if (rv <= 0) break;
-                    done += rv;
-                }
-                nTx = 0;
-            };
-            /* <= 32 segments per GSO packet (kernel caps at 64) */
-            const int maxBurst = gsoOk ? 32 : kRxBatch;
-            for (size_t si = 0; si < slots.size() && budget > 0; ++si) {
-                int slot = slots[si];
-                if (slot < 0 || slot % nSock != k) continue;
-                size_t ni = cls != nullptr
-                                ? (*cls)[rng() % cls->size()]
-                                : rng() % wires.size();
-                const auto& w = wires[ni];
-                slotSeq[slot]++;
-                uint16_t qid = (uint16_t)(
-                    (slot & slotMask) |
-                    ((slotSeq[slot] & seqMask) << slotBits));
-                uint8_t* dst;
-                if (gsoOk) {
-                    dst = gsoBuf.data() + gsoLen;
-                    gsoLen += w.size();
-                } else {
-                    dst = txBufs[nTx].data();
-                    txIovs[nTx] = {dst, w.size()};
-                    memset(&txHdrs[nTx], 0, sizeof(txHdrs[nTx]));
-                    txHdrs[nTx].msg_hdr.msg_iov = &txIovs[nTx];
-                    txHdrs[nTx].msg_hdr.msg_iovlen = 1;
-                }
-                memcpy(dst, w.data(), w.size());
-                dst[0] = (uint8_t)(qid >> 8);
-                dst[1] = (uint8_t)qid;
-                sentAt[slot] = nowUs();
-                if (out->tFirstUs == 0)
-                    out->tFirstUs = sentAt[slot] - tEnter;
-                out->tLastUs = sentAt[slot] - tEnter;
-                slotName[slot] = ni;
-                out->sent++;
-                launched++;
-                budget--;
-                used++;
-                slots[si] = -1;  // consumed
-                if (++nTx == maxBurst) flush();
-            }
-            flush();
-        }
-        if (used == slots.size() || budget == 0) {
-            /* compact: drop consumed entries */
-            size_t w = 0;
-            for (size_t i = 0; i < slots.size(); ++i)
-                if (slots[i] >= 0) slots[w++] = slots[i];
-            slots.resize(w);
-        } else {
-            slots.clear();
-        }
-    };
-
-    /* prime the window */
-    for (int s = 0; s < W && (uint64_t)s < target; ++s)
-        freeSlots.push_back(s);
-    batchSend(freeSlots);
-
     std::vector<struct pollfd> pfds(nSock);
     for (int k = 0; k < nSock; ++k) pfds[k] = {fds[k], POLLIN, 0};
-    int64_t lastSweep = nowUs();
-    while (completed < target && !abort->load()) {
-        /* Paced mode: wait only until the next 8-packet launch
-         * quantum accrues (clamped to 250 us). The box runs under a
-         * CPU quota, so busy-spinning generator threads steal quota
-         * from the server chain and trigger CFS throttling — whose
-         * whole-group freezes are exactly the multi-ms p99 artifacts
-         * the paced protocol exists to avoid. Sub-quantum sleeps keep
-         * launches smooth at microsecond granularity without burning
-         * idle cycles. */
-        int rv;
-        if (rate > 0) {
-            uint64_t backlog = allowance();
-            long waitNs;
-            if (backlog >= 32) {
-                waitNs = 0;
-            } else {
-                double need = (double)(32 - backlog) / rate * 1e9;
-                waitNs = need < 250000 ? (long)need : 250000;
-            }
-            struct timespec ts {0, waitNs};
-            rv = ppoll(pfds.data(), (nfds_t)nSock, &ts, nullptr);
-        } else {
-            rv = poll(pfds.data(), (nfds_t)nSock, 50);
+
+    ctl->setup.fetch_add(1);
+
+    uint64_t seenGen = 0;
+    while (!ctl->shutdown.load(std::memory_order_acquire)) {
+        /* wait for the next run (or shutdown) */
+        if (ctl->gen.load(std::memory_order_acquire) == seenGen) {
+            std::this_thread::sleep_for(std::chrono::microseconds(200));
+            continue;
         }
-        if (rv > 0) {
-            for (int k = 0; k < nSock; ++k) {
-                if (!(pfds[k].revents & POLLIN)) continue;
-                while (true) {
-                    for (int i = 0; i < kRxBatch; ++i) {
-                        rxIovs[i] = {rxBufs[i].data(), rxBufs[i].size()};
-                        memset(&rxHdrs[i], 0, sizeof(rxHdrs[i]));
-                        rxHdrs[i].msg_hdr.msg_iov = &rxIovs[i];
-                        rxHdrs[i].msg_hdr.msg_iovlen = 1;
-                    }
-                    int nr = recvmmsg(fds[k], rxHdrs.data(), kRxBatch,
-                                      MSG_DONTWAIT, nullptr);
-                    if (nr <= 0) break;
-                    int64_t now = nowUs();
-                    for (int i = 0; i < nr; ++i) {
-                        const uint8_t* rb = rxBufs[i].data();
-                        if (rxHdrs[i].msg_len < 12) continue;
-                        uint16_t qid = (uint16_t)((rb[0] << 8) | rb[1]);
-                        int slot = qid & slotMask;
-                        uint16_t seq = (uint16_t)(qid >> slotBits);
-                        if (slot >= W || sentAt[slot] == 0 ||
-                            (uint16_t)(slotSeq[slot] & seqMask) != seq)
-                            continue;  // stale/duplicate
-                        int64_t lat = now - sentAt[slot];
-                        out->latBuckets[latBucket2(lat)]++;
-                        out->received++;
-                        uint8_t rcode = rb[3] & 0x0F;
-                        if (rcode == 0)
-                            out->rcodeNoerror++;
-                        else
-                            out->rcodeOther++;
-                        out->answers +=
-                            (uint64_t)((rb[6] << 8) | rb[7]);
-                        sentAt[slot] = 0;
-                        completed++;
-                        freeSlots.push_back(slot);
-                    }
-                    if (nr < kRxBatch) break;
+        seenGen = ctl->gen.load(std::memory_order_acquire);
+
+        const uint64_t target = ctl->queriesPerThread;
+        const double rate = ctl->rateQps > 0
+                                ? ctl->rateQps / (double)cfg.threads
+                                : 0;
+        uint64_t launched = 0, completed = 0;
+        std::fill(sentAt.begin(), sentAt.end(), 0);
+        freeSlots.clear();
+
+        /* drain stragglers from a previous run so they cannot collide
+         * with fresh slots */
+        for (int k = 0; k < nSock; ++k) {
+            uint8_t junk[2048];
+            while (recv(fds[k], junk, sizeof(junk), MSG_DONTWAIT) > 0) {
+            }
+        }
+
+        const int64_t tStart = nowUs();
+
+        /* How many more queries may launch right now (rate pacing).
+         * Launches are quantized to full GSO bursts (32 segments, or
+         * 300 us of tokens at low rates) - smaller quanta shrink the
+         * super-packets and the per-packet kernel cost comes back. */
+        constexpr uint64_t kQuantum = 32;
+        int64_t lastLaunchUs = 0;
+        auto allowance = [&]() -> uint64_t {
+            if (launched >= target) return 0;
+            uint64_t left = target - launched;
+            if (rate <= 0) return left;
+            int64_t now = nowUs();
+            uint64_t paced =
+                (uint64_t)((double)(now - tStart) * rate / 1e6);
+            if (paced <= launched) return 0;
+            uint64_t a = paced - launched;
+            if (a < left && a < kQuantum && now - lastLaunchUs < 300)
+                return 0;
+            lastLaunchUs = now;
+            return a < left ? a : left;
+        };
+
+        /* Send queries for free slots, grouped per socket
+         * (slot % nSock). GSO path: the whole burst is equal-size
+         * segments of one super-packet => ONE trip through the
+         * kernel's send path instead of one per query. */
+        auto batchSend = [&](std::vector<int>& slots) {
+            uint64_t budget = allowance();
+            size_t used = 0;
+            for (int k = 0; k < nSock && budget > 0; ++k) {
+                const std::vector<size_t>* cls = nullptr;
+                size_t wireSize = 0;
+                if (gsoOk) {
+                    size_t ni = rng() % wires.size();
+                    for (const auto& c : classes)
+                        if (wires[c[0]].size() == wires[ni].size()) {
+                            cls = &c;
+                            break;
+                        }
+                    wireSize = wires[(*cls)[0]].size();
                 }
+                int nTx = 0;
+                size_t gsoLen = 0;
+                auto flush = [&]() {
+                    if (nTx == 0) return;
+                    if (gsoOk) {
+                        struct msghdr mh {};
+                        struct iovec iov {gsoBuf.data(), gsoLen};
+                        char cbuf[CMSG_SPACE(sizeof(uint16_t))] = {0};
+                        mh.msg_iov = &iov;
+                        mh.msg_iovlen = 1;
+                        if (nTx > 1) {
+                            mh.msg_control = cbuf;
+                            mh.msg_controllen = sizeof(cbuf);
+                            struct cmsghdr* cm = CMSG_FIRSTHDR(&mh);
+                            cm->cmsg_level = SOL_UDP;
+                            cm->cmsg_type = UDP_SEGMENT;
+                            cm->cmsg_len = CMSG_LEN(sizeof(uint16_t));
+                            uint16_t seg = (uint16_t)wireSize;
+                            memcpy(CMSG_DATA(cm), &seg, sizeof(seg));
+                        }
+                        ssize_t rv = sendmsg(fds[k], &mh, 0);
+                        if (rv < 0 &&
+                            (errno == EINVAL || errno == EIO ||
+                             errno == ENOTSUP)) {
+                            /* kernel without UDP GSO: permanent
+                             * fallback; re-send as discrete packets */
+                            gsoOk = false;
+                            for (int m = 0; m < nTx; ++m) {
+                                ssize_t r2 = send(
+                                    fds[k],
+                                    gsoBuf.data() + m * wireSize,
+                                    wireSize, 0);
+                                (void)r2;
+                            }
+                        }
+                        gsoLen = 0;
+                        nTx = 0;
+                        return;
+                    }
+                    int done = 0;
+                    while (done < nTx) {
+                        int rv = sendmmsg(fds[k], txHdrs.data() + done,
+                                          nTx - done, 0);
+                        if (rv <= 0) break;
+                        done += rv;
+                    }
+                    nTx = 0;
+                };
+                /* <= 32 segments per GSO packet (kernel caps at 64) */
+                const int maxBurst = gsoOk ? 32 : kRxBatch;
+                for (size_t si = 0;
+                     si < slots.size() && budget > 0; ++si) {
+                    int slot = slots[si];
+                    if (slot < 0 || slot % nSock != k) continue;
+                    size_t ni = cls != nullptr
+                                    ? (*cls)[rng() % cls->size()]
+                                    : rng() % wires.size();
+                    const auto& w = wires[ni];
+                    slotSeq[slot]++;
+                    uint16_t qid = (uint16_t)(
+                        (slot & slotMask) |
+                        ((slotSeq[slot] & seqMask) << slotBits));
+                    uint8_t* dst;
+                    if (gsoOk) {
+                        dst = gsoBuf.data() + gsoLen;
+                        gsoLen += w.size();
+                    } else {
+                        dst = txBufs[nTx].data();
+                        txIovs[nTx] = {dst, w.size()};
+                        memset(&txHdrs[nTx], 0, sizeof(txHdrs[nTx]));
+                        txHdrs[nTx].msg_hdr.msg_iov = &txIovs[nTx];
+                        txHdrs[nTx].msg_hdr.msg_iovlen = 1;
+                    }
+                    memcpy(dst, w.data(), w.size());
+                    dst[0] = (uint8_t)(qid >> 8);
+                    dst[1] = (uint8_t)qid;
+                    sentAt[slot] = nowUs();
+                    out->sent++;
+                    launched++;
+                    budget--;
+                    used++;
+                    slots[si] = -1;  // consumed
+                    if (++nTx == maxBurst) flush();
+                }
+                flush();
             }
-        }
-        if (launched < target) batchSend(freeSlots);
-        else freeSlots.clear();
-        int64_t now = nowUs();
-        if (now - lastSweep > 100000) {  // sweep timeouts every 100ms
-            lastSweep = now;
-            for (int s = 0; s < W; ++s) {
-                if (sentAt[s] != 0 && now - sentAt[s] > timeoutUs) {
-                    out->timeouts++;
-                    completed++;
-                    sentAt[s] = 0;
-                    if (launched < target) freeSlots.push_back(s);
+            if (used == slots.size() || budget == 0) {
+                /* compact: drop consumed entries */
+                size_t w = 0;
+                for (size_t i = 0; i < slots.size(); ++i)
+                    if (slots[i] >= 0) slots[w++] = slots[i];
+                slots.resize(w);
+            } else {
+                slots.clear();
+            }
+        };
+
+        /* prime the window */
+        for (int s = 0; s < W && (uint64_t)s < target; ++s)
+            freeSlots.push_back(s);
+        batchSend(freeSlots);
+
+        int64_t lastSweep = nowUs();
+        while (completed < target && !ctl->shutdown.load()) {
+            /* Paced mode: wait only until the next launch quantum
+             * accrues. The box runs under a CPU quota, so spinning
+             * generator threads steal quota from the server chain and
+             * CFS throttling puts whole-group freezes straight into
+             * p99; sub-quantum sleeps keep launches smooth without
+             * burning idle cycles. */
+            int rv;
+            if (rate > 0) {
+                uint64_t backlog = allowance();
+                long waitNs;
+                if (backlog >= kQuantum) {
+                    waitNs = 0;
+                } else {
+                    double need =
+                        (double)(kQuantum - backlog) / rate * 1e9;
+                    waitNs = need < 250000 ? (long)need : 250000;
+                }
+                struct timespec ts {0, waitNs};
+                rv = ppoll(pfds.data(), (nfds_t)nSock, &ts, nullptr);
+            } else {
+                rv = poll(pfds.data(), (nfds_t)nSock, 50);
+            }
+            if (rv > 0) {
+                for (int k = 0; k < nSock; ++k) {
+                    if (!(pfds[k].revents & POLLIN)) continue;
+                    while (true) {
+                        for (int i = 0; i < kRxBatch; ++i) {
+                            rxIovs[i] = {rxBufs[i].data(),
+                                         rxBufs[i].size()};
+                            memset(&rxHdrs[i], 0, sizeof(rxHdrs[i]));
+                            rxHdrs[i].msg_hdr.msg_iov = &rxIovs[i];
+                            rxHdrs[i].msg_hdr.msg_iovlen = 1;
+                        }
+                        int nr = recvmmsg(fds[k], rxHdrs.data(),
+                                          kRxBatch, MSG_DONTWAIT,
+                                          nullptr);
+                        if (nr <= 0) break;
+                        int64_t now = nowUs();
+                        for (int i = 0; i < nr; ++i) {
+                            const uint8_t* rb = rxBufs[i].data();
+                            if (rxHdrs[i].msg_len < 12) continue;
+                            uint16_t qid =
+                                (uint16_t)((rb[0] << 8) | rb[1]);
+                            int slot = qid & slotMask;
+                            uint16_t seq = (uint16_t)(qid >> slotBits);
+                            if (slot >= W || sentAt[slot] == 0 ||
+                                (uint16_t)(slotSeq[slot] & seqMask) !=
+                                    seq)
+                                continue;  // stale/duplicate
+                            int64_t lat = now - sentAt[slot];
+                            out->latBuckets[latBucket2(lat)]++;
+                            out->received++;
+                            uint8_t rcode = rb[3] & 0x0F;
+                            if (rcode == 0)
+                                out->rcodeNoerror++;
+                            else
+                                out->rcodeOther++;
+                            out->answers +=
+                                (uint64_t)((rb[6] << 8) | rb[7]);
+                            sentAt[slot] = 0;
+                            completed++;
+                            freeSlots.push_back(slot);
+                        }
+                        if (nr < kRxBatch) break;
+                    }
                 }
             }
             if (launched < target) batchSend(freeSlots);
             else freeSlots.clear();
+            int64_t now = nowUs();
+            if (now - lastSweep > 100000) {  // timeout sweep / 100ms
+                lastSweep = now;
+                for (int s = 0; s < W; ++s) {
+                    if (sentAt[s] != 0 && now - sentAt[s] > timeoutUs) {
+                        out->timeouts++;
+                        completed++;
+                        sentAt[s] = 0;
+                        if (launched < target) freeSlots.push_back(s);
+                    }
+                }
+                if (launched < target) batchSend(freeSlots);
+                else freeSlots.clear();
+            }
         }
+        ctl->running.fetch_sub(1);
     }
-    out->tDoneUs = nowUs() - tEnter;
     for (int fd : fds) close(fd);
 }
 
@@ -448,10 +478,11 @@ int main(int argc, char** argv) {
     signal(SIGPIPE, SIG_IGN);
     Config cfg;
     int c;
-    while ((c = getopt(argc, argv, "hs:p:n:c:t:P:r:f:B:T:Rg")) != -1) {
+    while ((c = getopt(argc, argv, "hs:p:n:c:t:P:r:f:B:T:RgD")) != -1) {
         switch (c) {
         case 'R': cfg.rd = true; break;
         case 'g': cfg.gso = false; break;
+        case 'D': cfg.daemon = true; break;
         case 's': cfg.server = optarg; break;
         case 'p': cfg.port = (uint16_t)atoi(optarg); break;
         case 'n': cfg.queries = strtoull(optarg, nullptr, 10); break;
@@ -465,11 +496,11 @@ int main(int argc, char** argv) {
         case 'h':
         default:
             fprintf(stderr,
-                    "usage: dnsblast -s server -p port -n queries "
+                    "usage: dnsblast -s server -p port [-n queries] "
                     "[-c window] [-t threads] [-P socks/thread] "
                     "[-r offered-qps] [-f names-file] "
                     "[-B bind-base-ip] [-T timeout-ms] [-R] "
-                    "[-g no-gso]\n");
+                    "[-g no-gso] [-D daemon]\n");
             return c == 'h' ? 0 : 1;
         }
     }
@@ -515,79 +546,83 @@ int main(int argc, char** argv) {
     }
 
     std::vector<ThreadResult> results(cfg.threads);
-    std::atomic<bool> abort{false};
-    std::atomic<int> ready{0};
-    std::atomic<bool> go{false};
-    uint64_t perThread = cfg.queries / cfg.threads;
-
+    RunCtl ctl;
     std::vector<std::thread> threads;
     for (int i = 0; i < cfg.threads; ++i) {
-        Config tcfg = cfg;
-        tcfg.queries = perThread;
         threads.emplace_back(
-            [tcfg, i, &wires, &classes, &results, &abort, &ready,
-             &go]() {
-                worker(tcfg, i, wires, classes, &results[i], &abort,
-                       &ready, &go);
+            [&cfg, i, &wires, &classes, &results, &ctl]() {
+                worker(cfg, i, wires, classes, &results[i], &ctl);
             });
     }
-    /* start the clock only when every thread is set up */
-    while (ready.load() < cfg.threads)
+    /* barrier: socket/buffer setup must not be charged to any run */
+    while (ctl.setup.load() < cfg.threads)
         std::this_thread::sleep_for(std::chrono::milliseconds(1));
-    int64_t t0 = nowUs();
-    go.store(true, std::memory_order_release);
-    for (auto& t : threads) t.join();
-    int64_t elapsedUs = nowUs() - t0;
 
-    ThreadResult total;
-    for (const auto& r : results) {
-        total.sent += r.sent;
-        total.received += r.received;
-        total.timeouts += r.timeouts;
-        total.rcodeNoerror += r.rcodeNoerror;
-        total.rcodeOther += r.rcodeOther;
-        total.answers += r.answers;
-        for (int b = 0; b < kLatBuckets; ++b)
-            total.latBuckets[b] += r.latBuckets[b];
-    }
+    auto doRun = [&](uint64_t queries, double rate) {
+        for (auto& r : results) r.reset();
+        ctl.queriesPerThread = queries / (uint64_t)cfg.threads;
+        ctl.rateQps = rate;
+        ctl.running.store(cfg.threads);
+        int64_t t0 = nowUs();
+        ctl.gen.fetch_add(1, std::memory_order_release);
+        while (ctl.running.load() > 0)
+            std::this_thread::sleep_for(std::chrono::microseconds(200));
+        int64_t elapsedUs = nowUs() - t0;
 
-    auto pct = [&](double q) {
-        uint64_t n = total.received;
-        if (n == 0) return 0.0;
-        uint64_t want = (uint64_t)(q * (double)n);
-        uint64_t cum = 0;
-        for (int b = 0; b < kLatBuckets; ++b) {
-            cum += total.latBuckets[b];
-            if (cum > want) return bucketMidUs(b);
+        ThreadResult total;
+        for (const auto& r : results) {
+            total.sent += r.sent;
+            total.received += r.received;
+            total.timeouts += r.timeouts;
+            total.rcodeNoerror += r.rcodeNoerror;
+            total.rcodeOther += r.rcodeOther;
+            total.answers += r.answers;
+            for (int b = 0; b < kLatBuckets; ++b)
+                total.latBuckets[b] += r.latBuckets[b];
         }
-        return bucketMidUs(kLatBuckets - 1);
+        auto pct = [&](double q) {
+            uint64_t n = total.received;
+            if (n == 0) return 0.0;
+            uint64_t want = (uint64_t)(q * (double)n);
+            uint64_t cum = 0;
+            for (int b = 0; b < kLatBuckets; ++b) {
+                cum += total.latBuckets[b];
+                if (cum > want) return bucketMidUs(b);
+            }
+            return bucketMidUs(kLatBuckets - 1);
+        };
+        double secs = (double)elapsedUs / 1e6;
+        double qps = secs > 0 ? (double)total.received / secs : 0;
+        printf("{\"sent\": %llu, \"received\": %llu, "
+               "\"timeouts\": %llu, \"noerror\": %llu, "
+               "\"other_rcode\": %llu, \"answers\": %llu, "
+               "\"elapsed_s\": %.6f, \"qps\": %.1f, "
+               "\"p50_us\": %.1f, \"p90_us\": %.1f, "
+               "\"p99_us\": %.1f}\n",
+               (unsigned long long)total.sent,
+               (unsigned long long)total.received,
+               (unsigned long long)total.timeouts,
+               (unsigned long long)total.rcodeNoerror,
+               (unsigned long long)total.rcodeOther,
+               (unsigned long long)total.answers, secs, qps,
+               pct(0.50), pct(0.90), pct(0.99));
+        fflush(stdout);
     };
 
-    int64_t maxSetup = 0, maxFirst = 0, maxLast = 0, maxDone = 0;
-    for (const auto& r : results) {
-        maxSetup = std::max(maxSetup, r.tSetupUs);
-        maxFirst = std::max(maxFirst, r.tFirstUs);
-        maxLast = std::max(maxLast, r.tLastUs);
-        maxDone = std::max(maxDone, r.tDoneUs);
+    if (cfg.daemon) {
+        /* "RUN <queries> <rate>" per line; EOF = shutdown */
+        char line[256];
+        while (fgets(line, sizeof(line), stdin) != nullptr) {
+            unsigned long long q = 0;
+            double r = 0;
+            if (sscanf(line, "RUN %llu %lf", &q, &r) >= 1 && q > 0)
+                doRun(q, r);
+        }
+    } else {
+        doRun(cfg.queries, cfg.rateQps);
     }
-    fprintf(stderr,
-            "# phases(max over threads, ms): setup %.1f first %.1f "
-            "last %.1f done %.1f wall %.1f\n",
-            maxSetup / 1e3, maxFirst / 1e3, maxLast / 1e3,
-            maxDone / 1e3, elapsedUs / 1e3);
 
-    double secs = (double)elapsedUs / 1e6;
-    double qps = secs > 0 ? (double)total.received / secs : 0;
-    printf("{\"sent\": %llu, \"received\": %llu, \"timeouts\": %llu, "
-           "\"noerror\": %llu, \"other_rcode\": %llu, \"answers\": %llu, "
-           "\"elapsed_s\": %.6f, \"qps\": %.1f, "
-           "\"p50_us\": %.1f, \"p90_us\": %.1f, \"p99_us\": %.1f}\n",
-           (unsigned long long)total.sent,
-           (unsigned long long)total.received,
-           (unsigned long long)total.timeouts,
-           (unsigned long long)total.rcodeNoerror,
-           (unsigned long long)total.rcodeOther,
-           (unsigned long long)total.answers, secs, qps, pct(0.50),
-           pct(0.90), pct(0.99));
+    ctl.shutdown.store(true, std::memory_order_release);
+    for (auto& t : threads) t.join();
     return 0;
 }
