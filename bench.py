@@ -187,24 +187,37 @@ class BlastDaemon:
 
 
 def calibrate_rate(blast, capacity, slo_us):
-    """qps@SLO discovery (untimed): walk a descending rate ladder from
-    the closed-loop capacity; the first offered rate that sustains
-    p99 <= SLO with zero timeouts and full delivery is the operating
-    point. Fixed-rate steps at that point are comparable round over
-    round, unlike the chaotic closed-loop saturation equilibrium
-    (profiles/SCALING.md round-1: steps swung +/-40% at saturation)."""
-    for frac in (1.0, 0.95, 0.9, 0.85, 0.8, 0.72, 0.64, 0.55):
-        rate = capacity * frac
-        q = max(200_000, int(rate * 1.2))
+    """qps@SLO discovery (untimed): binary-search the highest offered
+    rate that sustains p99 <= SLO with zero timeouts and >=97%
+    delivery. Near the edge single probes are flaky (rare ~10 ms
+    stall events land in p99 or miss it), so a rate must pass TWO
+    consecutive 2-second probes, and the chosen point keeps a 3%
+    margin + one final confirmation. Fixed-rate steps at that point
+    are comparable round over round, unlike the chaotic closed-loop
+    saturation equilibrium (profiles/SCALING.md)."""
+    def probe(rate, tag):
+        q = max(400_000, int(rate * 2.0))
         r = blast.step(q, rate=rate)
-        ok = (r["timeouts"] == 0 and r["p99_us"] <= slo_us and
-              r["qps"] >= 0.97 * rate)
-        log(f"calibrate {frac:.2f}x: offered {rate:.0f} -> "
-            f"{r['qps']:.0f} qps, p99 {r['p99_us']}us, "
-            f"timeouts {r['timeouts']} {'OK' if ok else 'over SLO'}")
-        if ok:
-            return rate
-    return capacity * 0.5
+        good = (r["timeouts"] == 0 and r["p99_us"] <= slo_us and
+                r["qps"] >= 0.97 * rate)
+        log(f"calibrate {tag}: offered {rate:.0f} -> "
+            f"{r['qps']:.0f} qps, p99 {r['p99_us']}us "
+            f"{'OK' if good else 'FAIL'}")
+        return good
+
+    lo, hi = 0.45 * capacity, 1.02 * capacity
+    best = 0.0
+    for i in range(5):
+        mid = (lo + hi) / 2
+        if probe(mid, f"bisect{i}a") and probe(mid, f"bisect{i}b"):
+            best = mid
+            lo = mid
+        else:
+            hi = mid
+    rate = 0.97 * (best if best > 0 else lo)
+    if not probe(rate, "confirm"):
+        rate *= 0.93
+    return rate
 
 
 def main():
