@@ -36,6 +36,7 @@
 #include <cstring>
 #include <map>
 #include <unordered_map>
+#include <chrono>
 #include <memory>
 #include <unordered_map>
 #include <mutex>
@@ -91,6 +92,64 @@ struct StatsHub {
     }
 };
 StatsHub g_stats;
+
+/*
+ * Shared socket-directory scanner: opendir/stat on the serving loop
+ * measured as rare ~10 ms stalls on the bench boxes' filesystem
+ * (profiles stall1.jsonl: stalls vanish when the tick slows), and
+ * every worker scanning the same directory each tick multiplies the
+ * exposure. One helper thread scans; workers read the snapshot.
+ */
+struct DirScanner {
+    std::mutex m;
+    std::set<std::string> paths;
+    std::atomic<uint64_t> gen{0};
+    std::thread thread;
+    std::atomic<bool> stop{false};
+
+    void start(std::string dir, int intervalMs) {
+        thread = std::thread([this, dir = std::move(dir),
+                              intervalMs]() {
+            while (!stop.load()) {
+                std::set<std::string> seen;
+                DIR* d = opendir(dir.c_str());
+                if (d != nullptr) {
+                    struct dirent* ent;
+                    while ((ent = readdir(d)) != nullptr) {
+                        std::string name = ent->d_name;
+                        if (name == "." || name == "..") continue;
+                        std::string full = dir + "/" + name;
+                        struct stat st;
+                        if (stat(full.c_str(), &st) == 0 &&
+                            S_ISSOCK(st.st_mode))
+                            seen.insert(full);
+                    }
+                    closedir(d);
+                }
+                {
+                    std::lock_guard<std::mutex> g(m);
+                    if (paths != seen) {
+                        paths = std::move(seen);
+                    }
+                }
+                gen.fetch_add(1);
+                for (int i = 0; i < intervalMs / 10 && !stop.load();
+                     ++i)
+                    std::this_thread::sleep_for(
+                        std::chrono::milliseconds(10));
+            }
+        });
+    }
+    std::set<std::string> snapshot() {
+        std::lock_guard<std::mutex> g(m);
+        return paths;
+    }
+    void shutdown() {
+        stop.store(true);
+        if (thread.joinable()) thread.join();
+    }
+};
+DirScanner g_scanner;
 
 /*
  * In-flight request table: a flat power-of-two ring indexed by
@@ -319,15 +378,17 @@ bool Balancer::start() {
     }
 
     rescan();
-    std::function<void()> tick = [this, &tick]() {};
-    /* periodic rescans + sweeps */
+    /* periodic rescans + sweeps, staggered across workers so the
+     * ticks never stall every shard in the same instant */
     auto schedule = std::make_shared<std::function<void()>>();
     *schedule = [this, schedule]() {
         rescan();
         sweep();
         loop_->addTimer(rescanMs_, *schedule);
     };
-    loop_->addTimer(rescanMs_, *schedule);
+    int64_t offset =
+        nWorkers_ > 0 ? (rescanMs_ * workerId_) / nWorkers_ : 0;
+    loop_->addTimer(rescanMs_ + offset, *schedule);
     log_.info({{"port", Json((int)port_)}, {"dir", Json(sockDir_)}},
               "balancer started");
     return true;
@@ -345,18 +406,11 @@ void Balancer::stop() {
 }
 
 void Balancer::rescan() {
-    std::set<std::string> seen;
-    DIR* d = opendir(sockDir_.c_str());
-    if (d != nullptr) {
-        struct dirent* ent;
-        while ((ent = readdir(d)) != nullptr) {
-            std::string name = ent->d_name;
-            if (name == "." || name == "..") continue;
-            std::string full = sockDir_ + "/" + name;
-            struct stat st;
-            if (stat(full.c_str(), &st) != 0 || !S_ISSOCK(st.st_mode))
-                continue;
-            seen.insert(full);
+    /* no filesystem access on the serving loop: the shared scanner
+     * thread maintains the socket set */
+    std::set<std::string> seen = g_scanner.snapshot();
+    {
+        for (const std::string& full : seen) {
             if (backends_.count(full) == 0) {
                 auto be = std::make_shared<Backend>();
                 be->id = nextBackendId_++;
@@ -372,7 +426,6 @@ void Balancer::rescan() {
                 if (be->fd < 0) connectBackend(be);  // retry
             }
         }
-        closedir(d);
     }
     /* removed sockets => drain (main.js:181-193 unlink-on-SIGTERM) */
     std::vector<std::string> gone;
@@ -1148,6 +1201,11 @@ int main(int argc, char** argv) {
      * affinity holds within a worker; -w 1 (default) preserves the
      * reference's exact single-process per-IP affinity semantics.
      */
+    g_scanner.start(dir, rescanMs < 500 ? rescanMs : 500);
+    /* give the first scan a moment so startup discovery is immediate */
+    while (g_scanner.gen.load() == 0)
+        std::this_thread::sleep_for(std::chrono::milliseconds(2));
+
     std::vector<std::unique_ptr<EventLoop>> loops;
     std::vector<std::unique_ptr<Balancer>> bals;
     std::vector<std::thread> threads;
@@ -1170,5 +1228,6 @@ int main(int argc, char** argv) {
     for (auto& l : loops) l->stop();
     for (auto& t : threads) t.join();
     for (auto& b : bals) b->stop();
+    g_scanner.shutdown();
     return 0;
 }
