@@ -283,6 +283,7 @@ class Balancer {
     std::unordered_map<uint64_t, int> remotesFast_;
     std::map<int, std::shared_ptr<TcpClient>> tcpClients_;
     int nextBackendId_ = 1;
+    int salvageDepth_ = 0;  /* bounds backendDown->salvage recursion */
     uint64_t udpQueries_ = 0, udpReplies_ = 0, drops_ = 0;
 
     static constexpr size_t kMaxRemotes = 262144;
@@ -472,6 +473,12 @@ void Balancer::backendDown(Backend* be) {
     }
     be->ok = false;
     be->in.clear();
+    /* Salvage: frames we queued but never flushed into the dead
+     * connection can be re-dispatched to a healthy backend instead of
+     * silently dropped — the client's in-flight query then survives a
+     * backend crash (a query already inside the dead socket's buffer
+     * is genuinely lost; the client retries). */
+    std::string unsent = be->out.substr(be->outOff);
     be->out.clear();
     be->outOff = 0;
     be->writeBlocked = false;
@@ -491,6 +498,50 @@ void Balancer::backendDown(Backend* be) {
     }
     g_pins.add(be->path, -(int)be->remotes);
     be->remotes = 0;
+
+    if (!unsent.empty() && salvageDepth_ < 4) {
+        salvageDepth_++;
+        std::set<Backend*> touched;
+        int64_t expiry = monotonicMillis() + kReplyTtlMs;
+        size_t off = 0;
+        while (unsent.size() - off >= bsock::kHeaderLen) {
+            const uint8_t* h = (const uint8_t*)unsent.data() + off;
+            uint32_t plen = bsock::getU32(h + 2);
+            if (h[0] != bsock::kMagic ||
+                plen > bsock::kMaxPayload ||
+                unsent.size() - off < bsock::kHeaderLen + plen)
+                break;
+            const uint8_t* payload = h + bsock::kHeaderLen;
+            /* re-dispatch UDP QUERY frames (layout: protocol.hpp);
+             * PINGs are dropped, TCP queries have their reply route
+             * in the dead backend's pending ring so the client
+             * retries those */
+            if (h[1] == bsock::FRAME_QUERY &&
+                plen >= bsock::kQueryHeadLen && payload[5] == 0) {
+                struct sockaddr_storage src {};
+                uint16_t port =
+                    (uint16_t)(payload[6] | (payload[7] << 8));
+                if (payload[4] == 4) {
+                    auto* sa = (struct sockaddr_in*)&src;
+                    sa->sin_family = AF_INET;
+                    sa->sin_port = htons(port);
+                    memcpy(&sa->sin_addr, payload + 8, 4);
+                } else {
+                    auto* sa = (struct sockaddr_in6*)&src;
+                    sa->sin6_family = AF_INET6;
+                    sa->sin6_port = htons(port);
+                    memcpy(&sa->sin6_addr, payload + 8, 16);
+                }
+                handleUdpQuery(payload + bsock::kQueryHeadLen,
+                               plen - bsock::kQueryHeadLen, src,
+                               expiry, touched);
+            }
+            off += bsock::kHeaderLen + plen;
+        }
+        for (Backend* tb : touched)
+            if (tb != be) backendFlush(tb);
+        salvageDepth_--;
+    }
 }
 
 void Balancer::backendFlush(Backend* be) {
