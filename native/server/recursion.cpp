@@ -29,6 +29,26 @@ Recursion::Recursion(EventLoop* loop, Logger log, RecursionOptions opts,
 Recursion::~Recursion() {
     if (ldapThread_.joinable()) ldapThread_.join();
     if (refreshTimer_) loop_->cancelTimer(refreshTimer_);
+    for (int fd : poolFds_)
+        if (fd >= 0) {
+            loop_->delFd(fd);
+            close(fd);
+        }
+}
+
+bool Recursion::ensurePool() {
+    if (!poolFds_.empty()) return true;
+    poolFds_.reserve(kPoolSize);
+    poolPending_.resize(kPoolSize);
+    for (int i = 0; i < kPoolSize; ++i) {
+        int fd = socket(AF_INET, SOCK_DGRAM | SOCK_NONBLOCK |
+                        SOCK_CLOEXEC, 0);
+        if (fd < 0) break;
+        poolFds_.push_back(fd);
+        loop_->addFd(fd, EPOLLIN,
+                     [this, i](uint32_t) { onPoolReadable(i); });
+    }
+    return !poolFds_.empty();
 }
 
 void Recursion::emitReady() {
@@ -367,23 +387,29 @@ void Recursion::resolve(const Message& query, Message& resp,
         refuse();
         return;
     }
-    if (activeLookups_ >= 1024) {
-        /* bound per-lookup fd usage: shed load (best effort) */
+    if (activeLookups_ >= 4096 || !ensurePool()) {
+        /* bound in-flight lookups: shed load (best effort) */
         refuse();
         return;
     }
 
     auto up = std::make_shared<Upstream>();
-    /* Fresh ephemeral-port socket per lookup: the kernel assigns a
-     * random source port, so a forged reply must hit both that port
-     * and the random qid, and pass the source/question checks below. */
-    up->fd = socket(AF_INET, SOCK_DGRAM | SOCK_NONBLOCK | SOCK_CLOEXEC, 0);
-    if (up->fd < 0) {
+    /* random socket from the outgoing pool + random non-colliding qid
+     * (the port+qid pair is what an off-path forger must guess) */
+    up->poolIdx = (int)(rng_() % poolFds_.size());
+    up->fd = poolFds_[up->poolIdx];
+    auto& pending = poolPending_[up->poolIdx];
+    uint16_t qid = 0;
+    for (int tries = 0; tries < 16; ++tries) {
+        qid = (uint16_t)(rng_() & 0xffff);
+        if (qid != 0 && pending.count(qid) == 0) break;
+        qid = 0;
+    }
+    if (qid == 0) {
         refuse();
         return;
     }
-    up->qid = (uint16_t)(rng_() & 0xffff);
-    if (up->qid == 0) up->qid = 1;
+    up->qid = qid;
     up->hosts = std::move(filtered);
     up->maxConcurrency = isPtr ? 100 : 2;  // recursion.js:64-78
     up->resp = &resp;
@@ -399,8 +425,7 @@ void Recursion::resolve(const Message& query, Message& resp,
     up->wire = out.encode(0);
 
     activeLookups_++;
-    loop_->addFd(up->fd, EPOLLIN,
-                 [this, up](uint32_t) { onSockReadable(up); });
+    poolPending_[up->poolIdx][up->qid] = up;
 
     up->timeoutTimer = loop_->addTimer(kUpstreamTimeoutMs, [this, up]() {
         up->timeoutTimer = 0;
@@ -433,14 +458,23 @@ void Recursion::sendNext(const std::shared_ptr<Upstream>& up) {
     if (up->inFlight == 0) finish(up, nullptr);
 }
 
-void Recursion::onSockReadable(const std::shared_ptr<Upstream>& up) {
+void Recursion::onPoolReadable(int idx) {
     uint8_t buf[4096];
-    while (!up->finished) {
+    int fd = poolFds_[idx];
+    auto& pending = poolPending_[idx];
+    while (true) {
         struct sockaddr_in src {};
         socklen_t slen = sizeof(src);
-        ssize_t nr = recvfrom(up->fd, buf, sizeof(buf), 0,
+        ssize_t nr = recvfrom(fd, buf, sizeof(buf), 0,
                               (struct sockaddr*)&src, &slen);
         if (nr <= 0) return;
+        if (nr < 12) continue;
+        /* match the lookup by qid on THIS socket */
+        uint16_t qid = (uint16_t)(((uint16_t)buf[0] << 8) | buf[1]);
+        auto it = pending.find(qid);
+        if (it == pending.end()) continue;
+        auto up = it->second;
+        if (up->finished) continue;
         /* Source must be an upstream we actually queried, replying
          * from the DNS port we sent to. */
         if (src.sin_family != AF_INET ||
@@ -484,10 +518,9 @@ void Recursion::finish(const std::shared_ptr<Upstream>& up,
         loop_->cancelTimer(up->timeoutTimer);
         up->timeoutTimer = 0;
     }
-    if (up->fd >= 0) {
-        loop_->delFd(up->fd);
-        close(up->fd);
-        up->fd = -1;
+    if (up->poolIdx >= 0) {
+        poolPending_[up->poolIdx].erase(up->qid);
+        up->poolIdx = -1;
         activeLookups_--;
     }
 

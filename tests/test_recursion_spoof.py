@@ -170,7 +170,9 @@ def test_question_mismatch_ignored(make_binder):
 
 def test_qids_randomized_and_ports_ephemeral(make_binder):
     """Outgoing qids must not be the old deterministic 1,2,3,... walk,
-    and each lookup must use its own ephemeral-port socket."""
+    and lookups must spread across the outgoing socket pool's
+    unpredictable source ports (port+qid is what a forger must
+    guess)."""
     def script(q, addr, sock):
         up.answer(q, addr, sock)
 
@@ -184,8 +186,9 @@ def test_qids_randomized_and_ports_ephemeral(make_binder):
         ports = [s[1] for s in up.seen]
         assert len(qids) == 4
         assert qids != [1, 2, 3, 4], "qids are the deterministic walk"
-        # fresh socket per lookup => kernel-assigned source ports; all
-        # four colliding on one port would mean a shared socket
+        # pool-drawn sockets => multiple kernel-assigned source ports
+        # in play; all four landing on one port would mean a single
+        # fixed socket (the original vulnerability)
         assert len(set(ports)) >= 2, f"one shared source port: {ports}"
     finally:
         up.close()
