@@ -487,13 +487,19 @@ void Recursion::onPoolReadable(int idx) {
         auto msg = Message::decode(buf, (size_t)nr);
         if (!msg || !msg->header.qr) continue;
         if (msg->header.id != up->qid) continue;
-        /* The echoed question must match what we asked. */
+        /* The echoed question must match what we asked
+         * (case-insensitive, compared in place). */
         if (msg->questions.size() != 1) continue;
-        std::string qn = msg->questions[0].name;
-        toLowerAscii(qn);
-        std::string expect = up->question.name;
-        toLowerAscii(expect);
-        if (qn != expect ||
+        const std::string& qn = msg->questions[0].name;
+        const std::string& expect = up->question.name;
+        bool sameName = qn.size() == expect.size();
+        for (size_t i = 0; sameName && i < qn.size(); ++i) {
+            char a = qn[i], b = expect[i];
+            if (a >= 'A' && a <= 'Z') a = (char)(a + 32);
+            if (b >= 'A' && b <= 'Z') b = (char)(b + 32);
+            sameName = a == b;
+        }
+        if (!sameName ||
             msg->questions[0].qtype != up->question.qtype ||
             msg->questions[0].qclass != up->question.qclass)
             continue;
