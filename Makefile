@@ -103,6 +103,7 @@ release: all
 	rm -rf dist/binder-amd
 	mkdir -p dist/binder-amd/bin dist/binder-amd/etc
 	cp bin/binderd bin/binder-balancer bin/binder-adjust \
+	    bin/zkd \
 	    bin/binder-supervisor bin/zklogcat bin/dnsblast bin/zktool \
 	    dist/binder-amd/bin/
 	cp -r deploy dist/binder-amd/
