@@ -32,6 +32,7 @@
 #include <sys/un.h>
 #include <unistd.h>
 
+#include <algorithm>
 #include <cstdlib>
 #include <cstring>
 #include <map>
@@ -284,6 +285,7 @@ class Balancer {
     std::map<int, std::shared_ptr<TcpClient>> tcpClients_;
     int nextBackendId_ = 1;
     int salvageDepth_ = 0;  /* bounds backendDown->salvage recursion */
+    bool replyGso_ = true;  /* UDP_SEGMENT reply runs (auto-fallback) */
     uint64_t udpQueries_ = 0, udpReplies_ = 0, drops_ = 0;
 
     static constexpr size_t kMaxRemotes = 262144;
@@ -603,17 +605,108 @@ void Balancer::onBackendEvent(std::shared_ptr<Backend> be, uint32_t ev) {
     }
 
     /* Walk complete frames without per-frame erase; UDP replies are
-     * batched into sendmmsg (a syscall per reply was the balancer's
-     * top cost at high QPS). */
+     * batched (a syscall per reply was the balancer's top cost at
+     * high QPS), and — because worker->backend affinity slices give
+     * each worker few flows — consecutive same-destination equal-size
+     * replies form runs that go out as ONE UDP_SEGMENT super-packet
+     * (multi-iov, no copy): the reply hop's kernel cost is per RUN,
+     * not per packet. Falls back to sendmmsg when GSO is refused. */
     constexpr int kReplyBatch = kBatch;
     struct mmsghdr* rh = replyHdrs_;
     struct iovec* riov = replyIovs_;
     struct sockaddr_storage* raddr = replyAddrs_;
     int nReply = 0;
+    int order[kReplyBatch];
+    auto sameDest = [&](int a, int b) {
+        if (raddr[a].ss_family != raddr[b].ss_family) return false;
+        if (raddr[a].ss_family == AF_INET) {
+            auto* x = (const struct sockaddr_in*)&raddr[a];
+            auto* y = (const struct sockaddr_in*)&raddr[b];
+            return x->sin_port == y->sin_port &&
+                   x->sin_addr.s_addr == y->sin_addr.s_addr;
+        }
+        auto* x = (const struct sockaddr_in6*)&raddr[a];
+        auto* y = (const struct sockaddr_in6*)&raddr[b];
+        return x->sin6_port == y->sin6_port &&
+               memcmp(&x->sin6_addr, &y->sin6_addr, 16) == 0;
+    };
     auto flushReplies = [&]() {
+        if (nReply == 0) return;
+        for (int i = 0; i < nReply; ++i) order[i] = i;
+        std::sort(order, order + nReply, [&](int a, int b) {
+            int c = memcmp(&raddr[a], &raddr[b], sizeof(raddr[a]));
+            if (c != 0) return c < 0;
+            return riov[a].iov_len > riov[b].iov_len;
+        });
+        int i = 0;
+        int nSingle = 0;
+        static thread_local std::vector<struct iovec> runIovs;
+        while (i < nReply) {
+            /* run: same destination, equal sizes (a shorter one may
+             * close the run — UDP_SEGMENT's trailing segment) */
+            int j = i + 1;
+            size_t seg = riov[order[i]].iov_len;
+            while (j < nReply && j - i < 48 &&
+                   sameDest(order[i], order[j]) &&
+                   (riov[order[j]].iov_len == seg ||
+                    (riov[order[j]].iov_len < seg &&
+                     (j + 1 == nReply ||
+                      !sameDest(order[i], order[j + 1])))))
+                ++j;
+            if (replyGso_ && j - i >= 2) {
+                runIovs.clear();
+                for (int k = i; k < j; ++k)
+                    runIovs.push_back(riov[order[k]]);
+                struct msghdr mh {};
+                mh.msg_name = &raddr[order[i]];
+                mh.msg_namelen =
+                    raddr[order[i]].ss_family == AF_INET
+                        ? sizeof(struct sockaddr_in)
+                        : sizeof(struct sockaddr_in6);
+                mh.msg_iov = runIovs.data();
+                mh.msg_iovlen = runIovs.size();
+                char cbuf[CMSG_SPACE(sizeof(uint16_t))] = {0};
+                mh.msg_control = cbuf;
+                mh.msg_controllen = sizeof(cbuf);
+                struct cmsghdr* cm = CMSG_FIRSTHDR(&mh);
+                cm->cmsg_level = SOL_UDP;
+                cm->cmsg_type = UDP_SEGMENT;
+                cm->cmsg_len = CMSG_LEN(sizeof(uint16_t));
+                uint16_t gso = (uint16_t)seg;
+                memcpy(CMSG_DATA(cm), &gso, sizeof(gso));
+                if (sendmsg(udpFd_, &mh, 0) < 0 &&
+                    (errno == EINVAL || errno == EIO ||
+                     errno == ENOTSUP)) {
+                    replyGso_ = false;  /* fall back permanently */
+                    for (int k = i; k < j; ++k) {
+                        struct msghdr m1 {};
+                        m1.msg_name = mh.msg_name;
+                        m1.msg_namelen = mh.msg_namelen;
+                        m1.msg_iov = &riov[order[k]];
+                        m1.msg_iovlen = 1;
+                        ssize_t r1 = sendmsg(udpFd_, &m1, 0);
+                        (void)r1;
+                    }
+                }
+            } else {
+                /* singles re-packed into a sendmmsg batch */
+                for (int k = i; k < j; ++k) {
+                    memset(&rh[nSingle], 0, sizeof(rh[nSingle]));
+                    rh[nSingle].msg_hdr.msg_iov = &riov[order[k]];
+                    rh[nSingle].msg_hdr.msg_iovlen = 1;
+                    rh[nSingle].msg_hdr.msg_name = &raddr[order[k]];
+                    rh[nSingle].msg_hdr.msg_namelen =
+                        raddr[order[k]].ss_family == AF_INET
+                            ? sizeof(struct sockaddr_in)
+                            : sizeof(struct sockaddr_in6);
+                    nSingle++;
+                }
+            }
+            i = j;
+        }
         int sent = 0;
-        while (sent < nReply) {
-            int rv = sendmmsg(udpFd_, rh + sent, nReply - sent, 0);
+        while (sent < nSingle) {
+            int rv = sendmmsg(udpFd_, rh + sent, nSingle - sent, 0);
             if (rv <= 0) break;
             sent += rv;
         }
@@ -661,7 +754,9 @@ void Balancer::onBackendEvent(std::shared_ptr<Backend> be, uint32_t ev) {
                     socklen_t slen;
                     if (pr->family == 4) {
                         auto* sa = (struct sockaddr_in*)&raddr[nReply];
-                        memset(sa, 0, sizeof(*sa));
+                        /* zero the FULL storage: the reply sorter
+                         * memcmps whole sockaddr_storage entries */
+                        memset(&raddr[nReply], 0, sizeof(raddr[0]));
                         sa->sin_family = AF_INET;
                         sa->sin_port = htons(pr->srcPort);
                         memcpy(&sa->sin_addr, pr->addr, 4);
@@ -669,18 +764,14 @@ void Balancer::onBackendEvent(std::shared_ptr<Backend> be, uint32_t ev) {
                     } else {
                         auto* sa =
                             (struct sockaddr_in6*)&raddr[nReply];
-                        memset(sa, 0, sizeof(*sa));
+                        memset(&raddr[nReply], 0, sizeof(raddr[0]));
                         sa->sin6_family = AF_INET6;
                         sa->sin6_port = htons(pr->srcPort);
                         memcpy(&sa->sin6_addr, pr->addr, 16);
                         slen = sizeof(*sa);
                     }
+                    (void)slen;
                     riov[nReply] = {const_cast<uint8_t*>(dns), dnsLen};
-                    memset(&rh[nReply], 0, sizeof(rh[nReply]));
-                    rh[nReply].msg_hdr.msg_iov = &riov[nReply];
-                    rh[nReply].msg_hdr.msg_iovlen = 1;
-                    rh[nReply].msg_hdr.msg_name = &raddr[nReply];
-                    rh[nReply].msg_hdr.msg_namelen = slen;
                     nReply++;
                 }
                 be->replies++;

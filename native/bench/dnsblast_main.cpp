@@ -152,6 +152,10 @@ void worker(const Config& cfg, int tid,
         int sz = 4 << 20;
         setsockopt(fd, SOL_SOCKET, SO_RCVBUF, &sz, sizeof(sz));
         setsockopt(fd, SOL_SOCKET, SO_SNDBUF, &sz, sizeof(sz));
+        /* receive balancer reply runs coalesced (one kernel
+         * traversal per run); harmless if unsupported */
+        int one = 1;
+        setsockopt(fd, SOL_UDP, UDP_GRO, &one, sizeof(one));
         if (!cfg.bindBase.empty()) {
             /* bind distinct loopback source ip per thread: base + tid */
             struct in_addr base;
@@ -195,7 +199,11 @@ void worker(const Config& cfg, int tid,
     /* batched RX (recvmmsg) and TX (sendmmsg/GSO): syscall count, not
      * packet handling, bounds the generator at high QPS */
     constexpr int kRxBatch = 64;
-    std::vector<std::array<uint8_t, 2048>> rxBufs(kRxBatch);
+    /* 16 KB per rx message: with UDP_GRO one message may be a
+     * coalesced run of up to 48 replies */
+    std::vector<std::array<uint8_t, 16384>> rxBufs(kRxBatch);
+    std::vector<std::array<char, CMSG_SPACE(sizeof(uint16_t))>>
+        rxCtrl(kRxBatch);
     std::vector<struct mmsghdr> rxHdrs(kRxBatch);
     std::vector<struct iovec> rxIovs(kRxBatch);
     std::vector<std::array<uint8_t, 2048>> txBufs(kRxBatch);
@@ -413,6 +421,10 @@ void worker(const Config& cfg, int tid,
                             memset(&rxHdrs[i], 0, sizeof(rxHdrs[i]));
                             rxHdrs[i].msg_hdr.msg_iov = &rxIovs[i];
                             rxHdrs[i].msg_hdr.msg_iovlen = 1;
+                            rxHdrs[i].msg_hdr.msg_control =
+                                rxCtrl[i].data();
+                            rxHdrs[i].msg_hdr.msg_controllen =
+                                rxCtrl[i].size();
                         }
                         int nr = recvmmsg(fds[k], rxHdrs.data(),
                                           kRxBatch, MSG_DONTWAIT,
@@ -420,8 +432,29 @@ void worker(const Config& cfg, int tid,
                         if (nr <= 0) break;
                         int64_t now = nowUs();
                         for (int i = 0; i < nr; ++i) {
-                            const uint8_t* rb = rxBufs[i].data();
-                            if (rxHdrs[i].msg_len < 12) continue;
+                            /* GRO: split a coalesced message into its
+                             * equal-size segments (cmsg carries the
+                             * segment size) */
+                            size_t seg = rxHdrs[i].msg_len;
+                            for (struct cmsghdr* cm = CMSG_FIRSTHDR(
+                                     &rxHdrs[i].msg_hdr);
+                                 cm != nullptr;
+                                 cm = CMSG_NXTHDR(&rxHdrs[i].msg_hdr,
+                                                  cm)) {
+                                if (cm->cmsg_level == SOL_UDP &&
+                                    cm->cmsg_type == UDP_GRO) {
+                                    uint16_t g;
+                                    memcpy(&g, CMSG_DATA(cm),
+                                           sizeof(g));
+                                    if (g > 0) seg = g;
+                                }
+                            }
+                            size_t total = rxHdrs[i].msg_len;
+                            for (size_t off = 0; off < total;
+                                 off += seg) {
+                            const uint8_t* rb = rxBufs[i].data() + off;
+                            size_t len = std::min(seg, total - off);
+                            if (len < 12) continue;
                             uint16_t qid =
                                 (uint16_t)((rb[0] << 8) | rb[1]);
                             int slot = qid & slotMask;
@@ -443,6 +476,7 @@ void worker(const Config& cfg, int tid,
                             sentAt[slot] = 0;
                             completed++;
                             freeSlots.push_back(slot);
+                            }
                         }
                         if (nr < kRxBatch) break;
                     }
