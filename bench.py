@@ -209,7 +209,13 @@ def calibrate_rate(blast, capacity, slo_us):
     best = 0.0
     for i in range(5):
         mid = (lo + hi) / 2
-        if probe(mid, f"bisect{i}a") and probe(mid, f"bisect{i}b"):
+        a = probe(mid, f"bisect{i}a")
+        b = probe(mid, f"bisect{i}b") if a else False
+        if a and not b:
+            # one transient flake must not collapse the search:
+            # 2-of-3 consecutive probes decide the level
+            b = probe(mid, f"bisect{i}c")
+        if a and b:
             best = mid
             lo = mid
         else:
