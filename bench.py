@@ -434,6 +434,11 @@ def main():
                     "p50_us": last["p50_us"] if last else None,
                     "p99_us": last["p99_us"] if last else None,
                     "timeouts_last_step": last["timeouts"] if last else None,
+                    # honesty guard: every counted reply must be a real
+                    # NOERROR answer
+                    "noerror_frac_last_step":
+                        (round(last["noerror"] / max(1, last["received"]),
+                               4) if last else None),
                 },
             }
             print(json.dumps(result), flush=True)
