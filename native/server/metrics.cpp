@@ -16,6 +16,13 @@ Histogram::Histogram()
     : bounds_({0.0001, 0.00025, 0.0005, 0.001, 0.0025, 0.005, 0.01, 0.025,
                0.05, 0.1, 0.25, 0.5, 1, 2.5, 5, 10}) {}
 
+Histogram::Series& Histogram::seriesRef(const std::string& labels) {
+    Series& s = series_[labels];
+    if (s.bucketCounts.empty())
+        s.bucketCounts.resize(bounds_.size() + 1, 0);
+    return s;
+}
+
 void Histogram::observe(const std::string& labels, double v) {
     Series& s = series_[labels];
     if (s.bucketCounts.empty()) s.bucketCounts.resize(bounds_.size() + 1, 0);

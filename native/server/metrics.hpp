@@ -24,6 +24,9 @@ class Counter {
     void increment(const std::string& labels, uint64_t by = 1) {
         vals_[labels] += by;
     }
+    /* hot-path form: the returned slot is stable (map node), so
+     * callers cache it once and bump it without a string lookup */
+    uint64_t& slot(const std::string& labels) { return vals_[labels]; }
     void set(const std::string& labels, uint64_t v) {  // gauges
         vals_[labels] = v;
     }
@@ -37,6 +40,18 @@ class Histogram {
   public:
     Histogram();
     void observe(const std::string& labels, double v);
+
+    struct Series;
+    /* hot-path form: resolve the labelled series once (stable map
+     * node), then observeFast per event without a string lookup */
+    Series& seriesRef(const std::string& labels);
+    void observeFast(Series& s, double v) {
+        size_t i = 0;
+        while (i < bounds_.size() && v > bounds_[i]) ++i;
+        s.bucketCounts[i]++;
+        s.sum += v;
+        s.count++;
+    }
 
     struct Series {
         std::vector<uint64_t> bucketCounts;

@@ -118,6 +118,14 @@ class DnsServer {
     std::string_view srvSvc_, srvProto_;   /* scratch, loop thread only */
     std::vector<uint32_t> shuffleIdx_;
     std::mt19937 rng_{std::random_device{}()};
+    void initFastMetricSlots();
+    /* cached hot-path metric slots (stable map nodes) */
+    uint64_t* fpCntA_ = nullptr;
+    uint64_t* fpCntSrv_ = nullptr;
+    Histogram::Series* fpLatA_ = nullptr;
+    Histogram::Series* fpLatSrv_ = nullptr;
+    Histogram::Series* fpSizeA_ = nullptr;
+    Histogram::Series* fpSizeSrv_ = nullptr;
     std::string logFields_, logScratch_;   /* afterQuery log buffers */
 
   public:
