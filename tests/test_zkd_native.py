@@ -331,3 +331,58 @@ def test_binderd_survives_zkd_restart(tmp_path):
         if b:
             b.stop()
         z.stop()
+
+
+@pytest.mark.timeout(120)
+def test_mirror_converges_under_zkd_churn(tmp_path):
+    """Sustained mutations against the native registry: the binderd
+    mirror must converge to the final addresses (the stubzk churn
+    suite's semantics, against zkd)."""
+    z = NativeZkd().start()
+    b = None
+    try:
+        c = ZkConn("127.0.0.1", z.port)
+        c.mkdirp("/com/foo")
+        n_hosts = 300
+        for i in range(n_hosts):
+            c.create(f"/com/foo/h{i}", json.dumps(
+                {"type": "host",
+                 "host": {"address": f"10.1.{i // 250}.{i % 250}"}}
+            ).encode())
+
+        b = BinderProcess(dns_domain="foo.com", store="zk",
+                          zk_host="127.0.0.1", zk_port=z.port,
+                          workdir=tmp_path)
+        b.start()
+        b.wait_ready(f"h{n_hosts - 1}.foo.com", timeout=30)
+
+        # churn: rewrite every host several times, last write wins
+        import random
+        rng = random.Random(3)
+        final = {}
+        for _ in range(4 * n_hosts):
+            i = rng.randrange(n_hosts)
+            addr = f"10.9.{rng.randrange(250)}.{rng.randrange(1, 250)}"
+            c.set(f"/com/foo/h{i}", json.dumps(
+                {"type": "host", "host": {"address": addr}}).encode())
+            final[i] = addr
+
+        deadline = time.time() + 30
+        pendingkeys = sorted(final)
+        while time.time() < deadline and pendingkeys:
+            still = []
+            for i in pendingkeys:
+                r = b.dig(f"h{i}.foo.com")
+                if not (r.status == "NOERROR" and
+                        r.answers[0]["address"] == final[i]):
+                    still.append(i)
+            pendingkeys = still
+            if pendingkeys:
+                time.sleep(0.3)
+        assert not pendingkeys, \
+            f"{len(pendingkeys)} hosts never converged"
+        c.close()
+    finally:
+        if b:
+            b.stop()
+        z.stop()
