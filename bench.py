@@ -341,7 +341,10 @@ def main():
                 r = blast.step(q_step)
                 capacity = max(capacity, r["qps"])
             rate = 0
-            if not closed_loop:
+            fixed = float(os.environ.get("BENCH_FIXED_RATE", "0"))
+            if fixed > 0:
+                rate = fixed  # diagnostics: skip calibration
+            elif not closed_loop:
                 # qps@SLO protocol: fixed offered rate for the timed
                 # steps, discovered against the SLO (untimed)
                 rate = calibrate_rate(blast, capacity, slo_us)
