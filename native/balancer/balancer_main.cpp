@@ -591,11 +591,17 @@ void Balancer::onBackendEvent(std::shared_ptr<Backend> be, uint32_t ev) {
     if (ev & EPOLLOUT) backendFlush(be.get());
     if (!(ev & EPOLLIN)) return;
 
+    /* bounded batch per event (level-triggered epoll re-fires):
+     * draining a many-MB reply backlog in one go stalls the worker's
+     * other duties for milliseconds */
+    constexpr size_t kMaxReadPerEvent = 256 * 1024;
+    size_t got = 0;
     char buf[65536];
-    while (be->fd >= 0) {
+    while (be->fd >= 0 && got < kMaxReadPerEvent) {
         ssize_t nr = read(be->fd, buf, sizeof(buf));
         if (nr > 0) {
             be->in.append(buf, (size_t)nr);
+            got += (size_t)nr;
             continue;
         }
         if (nr < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) break;

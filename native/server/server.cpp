@@ -1014,11 +1014,21 @@ void DnsServer::onBalConn(BalConn* c, uint32_t events) {
     if (events & EPOLLOUT) balFlush(c);
     if (c->closed || !(events & EPOLLIN)) return;
 
+    /* Bounded batch per epoll event: the GSO-era balancer can queue
+     * megabytes behind one connection, and draining it all here would
+     * process tens of thousands of queries (10+ ms) while the other
+     * workers' connections wait - measured as bimodal multi-ms p99 at
+     * N=1. Level-triggered epoll re-fires for the remainder, so
+     * capping the read keeps per-event latency bounded and service
+     * across connections fair. */
+    constexpr size_t kMaxReadPerEvent = 128 * 1024;
+    size_t got = 0;
     char buf[16384];
-    while (true) {
+    while (got < kMaxReadPerEvent) {
         ssize_t nr = read(c->fd, buf, sizeof(buf));
         if (nr > 0) {
             c->in.append(buf, (size_t)nr);
+            got += (size_t)nr;
             continue;
         }
         if (nr < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) break;
