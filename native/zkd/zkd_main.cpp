@@ -491,24 +491,25 @@ void Zkd::reapEphemerals(int64_t sessionId) {
 
 void Zkd::sweepSessions() {
     int64_t now = monotonicMillis();
-    /* a session with a live connection is refreshed by its packets;
-     * one without must re-attach within its negotiated timeout */
-    std::set<int64_t> attached;
-    for (auto& [fd, c] : conns_)
-        if (!c->closed && c->sessionId) attached.insert(c->sessionId);
+    /* Real-ZK semantics: liveness comes from PACKETS (pings), not
+     * from the TCP connection existing — a silent client's session
+     * expires even while its socket stays open. lastSeenMs is
+     * refreshed in op()/handshake() only. */
     std::vector<int64_t> dead;
-    for (auto& [sid, s] : sessions_) {
-        if (attached.count(sid)) {
-            s.lastSeenMs = now;
-            continue;
-        }
+    for (auto& [sid, s] : sessions_)
         if (now - s.lastSeenMs > s.timeoutMs) dead.push_back(sid);
-    }
     for (int64_t sid : dead) {
         log_.info({{"session", Json((int64_t)sid)}},
                   "session expired; reaping ephemerals");
         sessions_.erase(sid);
         reapEphemerals(sid);
+        /* drop the session's connections: their next op would act on
+         * an expired session */
+        std::vector<Conn*> doomed;
+        for (auto& [fd, c] : conns_)
+            if (!c->closed && c->sessionId == sid)
+                doomed.push_back(c.get());
+        for (Conn* c : doomed) closeConn(c);
     }
 }
 

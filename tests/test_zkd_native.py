@@ -136,16 +136,35 @@ def test_session_timeout_reaps_ephemerals_and_fires_watches():
         assert stat["ephemeralOwner"] == sid_a
 
         # A's connection dies without close; after the session timeout
-        # the ephemeral must be reaped and B's watch must fire DELETED
+        # the ephemeral must be reaped and B's watch must fire
+        # DELETED. B keeps ITS session alive by pinging (liveness is
+        # packet-based, like real ZooKeeper), so reads interleave ping
+        # replies (xid -2) with the notification (xid -1).
         a.close()
-        b.settimeout(5)
-        body = read_packet(b)  # blocks until the notification
+        b.settimeout(0.3)
+        deadline = time.time() + 8
+        body = None
+        while time.time() < deadline:
+            b.sendall(req(-2, 11))  # ping
+            try:
+                pkt = read_packet(b)
+            except (TimeoutError, socket.timeout):
+                continue
+            if struct.unpack(">i", pkt[:4])[0] == -1:
+                body = pkt
+                break
+        assert body is not None, "notification never arrived"
         xid, zxid, err, rest = parse_reply_header(body)
         assert xid == -1
         ev_type, ev_state = struct.unpack(">ii", rest[:8])
         assert ev_type == 2  # NodeDeleted
+        b.settimeout(5)
         b.sendall(req(2, 3, jstr("/eph") + b"\x00"))
-        xid, zxid, err, rest = parse_reply_header(read_packet(b))
+        while True:
+            pkt = read_packet(b)
+            if struct.unpack(">i", pkt[:4])[0] == 2:
+                break
+        xid, zxid, err, rest = parse_reply_header(pkt)
         assert err == -101  # ZNONODE
         b.close()
     finally:
