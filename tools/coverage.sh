@@ -18,6 +18,7 @@ for src in native/dns/codec native/engine/engine native/engine/store \
            native/server/recursion native/server/ldap \
            native/server/metrics native/balancer/balancer_main \
            native/adjust/supervisor_main native/adjust/adjust_main \
+           native/zkd/zkd_main \
            native/zklog/zklogcat_main; do
   pct=$(gcov -n -o "$(dirname "$src")" "$OLDPWD/${src}.cpp" 2>/dev/null |
         grep -A1 "File.*${src##*/}" | grep "Lines executed" | head -1)
