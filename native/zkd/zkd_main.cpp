@@ -385,19 +385,24 @@ void Zkd::fire(const std::string& path, int32_t ev, bool child,
     w.i32(ev);
     w.i32(STATE_SYNC_CONNECTED);
     w.str(path);
+    /* collect first: sendRaw -> flushConn may closeConn, which
+     * erases from conns_ and would invalidate this iteration */
+    std::vector<std::shared_ptr<Conn>> fired;
     for (auto& [fd, c] : conns_) {
         if (c->closed || !c->handshaken) continue;
-        bool fired = false;
+        bool hit = false;
         if (child) {
-            fired = c->childWatches.erase(path) > 0;
+            hit = c->childWatches.erase(path) > 0;
         } else {
-            if (!existsOnly) fired = c->dataWatches.erase(path) > 0;
-            if (c->existsWatches.erase(path) > 0) fired = true;
+            if (!existsOnly) hit = c->dataWatches.erase(path) > 0;
+            if (c->existsWatches.erase(path) > 0) hit = true;
         }
-        if (fired) {
-            sendRaw(c.get(), w.buf);
-            watchesFired_++;
-        }
+        if (hit) fired.push_back(c);
+    }
+    for (auto& c : fired) {
+        if (c->closed) continue;
+        sendRaw(c.get(), w.buf);
+        watchesFired_++;
     }
 }
 
