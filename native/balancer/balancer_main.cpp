@@ -652,7 +652,12 @@ void Balancer::onBackendEvent(std::shared_ptr<Backend> be, uint32_t ev) {
              * close the run — UDP_SEGMENT's trailing segment) */
             int j = i + 1;
             size_t seg = riov[order[i]].iov_len;
-            while (j < nReply && j - i < 48 &&
+            /* one UDP datagram caps at ~64KB: bound the run so large
+             * replies can never push the super-packet past it */
+            size_t maxRun = seg > 0 ? 60000 / seg : 1;
+            if (maxRun > 48) maxRun = 48;
+            if (maxRun < 1) maxRun = 1;
+            while (j < nReply && (size_t)(j - i) < maxRun &&
                    sameDest(order[i], order[j]) &&
                    (riov[order[j]].iov_len == seg ||
                     (riov[order[j]].iov_len < seg &&
@@ -680,10 +685,12 @@ void Balancer::onBackendEvent(std::shared_ptr<Backend> be, uint32_t ev) {
                 cm->cmsg_len = CMSG_LEN(sizeof(uint16_t));
                 uint16_t gso = (uint16_t)seg;
                 memcpy(CMSG_DATA(cm), &gso, sizeof(gso));
-                if (sendmsg(udpFd_, &mh, 0) < 0 &&
-                    (errno == EINVAL || errno == EIO ||
-                     errno == ENOTSUP)) {
-                    replyGso_ = false;  /* fall back permanently */
+                if (sendmsg(udpFd_, &mh, 0) < 0) {
+                    if (errno == EINVAL || errno == EIO ||
+                        errno == ENOTSUP)
+                        replyGso_ = false;  /* disable permanently */
+                    /* ANY failure: the whole run would be lost —
+                     * resend its replies individually */
                     for (int k = i; k < j; ++k) {
                         struct msghdr m1 {};
                         m1.msg_name = mh.msg_name;

@@ -306,12 +306,12 @@ void worker(const Config& cfg, int tid,
                             memcpy(CMSG_DATA(cm), &seg, sizeof(seg));
                         }
                         ssize_t rv = sendmsg(fds[k], &mh, 0);
-                        if (rv < 0 &&
-                            (errno == EINVAL || errno == EIO ||
-                             errno == ENOTSUP)) {
-                            /* kernel without UDP GSO: permanent
-                             * fallback; re-send as discrete packets */
-                            gsoOk = false;
+                        if (rv < 0) {
+                            if (errno == EINVAL || errno == EIO ||
+                                errno == ENOTSUP)
+                                gsoOk = false;  /* no kernel GSO */
+                            /* any failure: re-send the burst as
+                             * discrete packets so nothing is lost */
                             for (int m = 0; m < nTx; ++m) {
                                 ssize_t r2 = send(
                                     fds[k],
@@ -333,8 +333,13 @@ void worker(const Config& cfg, int tid,
                     }
                     nTx = 0;
                 };
-                /* <= 32 segments per GSO packet (kernel caps at 64) */
-                const int maxBurst = gsoOk ? 32 : kRxBatch;
+                /* <= 32 segments per GSO packet (kernel caps at 64),
+                 * and never past the ~64KB UDP datagram limit */
+                int maxBurst = gsoOk ? 32 : kRxBatch;
+                if (gsoOk && wireSize > 0 &&
+                    (size_t)maxBurst * wireSize > 60000)
+                    maxBurst = (int)(60000 / wireSize);
+                if (maxBurst < 1) maxBurst = 1;
                 for (size_t si = 0;
                      si < slots.size() && budget > 0; ++si) {
                     int slot = slots[si];
