@@ -384,3 +384,80 @@ def test_pending_ring_overwrites_counted(tmp_path):
         bsock.close()
         for c in conns:
             c.close()
+
+
+def test_large_reply_runs_through_balancer(tmp_path):
+    """Reply-run GSO with big responses: hundreds of in-flight EDNS
+    SRV queries from ONE client socket produce same-destination runs
+    of ~1.3KB replies — exercising the super-packet size cap and the
+    any-failure resend path. Every reply must come back NOERROR with
+    the full answer set (nothing silently lost)."""
+    from binder_amd import require_native
+    from binder_amd.harness import BinderProcess
+    n = require_native()
+
+    members = 16
+    tree = {"foo.com": None,
+            "big.foo.com": {"type": "service",
+                            "service": {"srvce": "_x", "proto": "_tcp",
+                                        "port": 80, "ttl": 60}}}
+    for i in range(members):
+        tree[f"m{i:02d}.big.foo.com"] = {
+            "type": "rr_host", "rr_host": {"address": f"10.4.0.{i+1}"}}
+    store = tmp_path / "tree.json"
+    store.write_text(json.dumps(tree))
+    sockdir = tmp_path / "socks"
+    sockdir.mkdir()
+    b = BinderProcess(store=f"file:{store}", workdir=tmp_path,
+                      balancer_socket=str(sockdir / "b0"))
+    b.start()
+    port = free_port()
+    bal = subprocess.Popen(
+        [str(BALANCERD), "-p", str(port), "-H", "127.0.0.1",
+         "-s", str(sockdir), "-r", "100"],
+        env=dict(os.environ, LOG_LEVEL="warn"))
+    try:
+        deadline = time.time() + 10
+        while time.time() < deadline:
+            try:
+                r = dig("big.foo.com", port=port, timeout=0.3)
+                if r.status == "NOERROR":
+                    break
+            except OSError:
+                pass
+            time.sleep(0.1)
+
+        total = 300
+        s = socket.socket(socket.AF_INET, socket.SOCK_DGRAM)
+        s.setsockopt(socket.SOL_SOCKET, socket.SO_RCVBUF, 8 << 20)
+        s.settimeout(5)
+        s.connect(("127.0.0.1", port))
+        # EDNS queries (1400B payload) so the multi-answer SRV
+        # response is not TC-truncated
+        for i in range(total):
+            wire = n.encode_message(
+                {"id": i, "questions": [
+                    {"name": "_x._tcp.big.foo.com", "type": "SRV"}],
+                 "additionals": [{"type": "OPT", "udp_size": 1400}]})
+            s.send(wire)
+        got = {}
+        deadline = time.time() + 10
+        while len(got) < total and time.time() < deadline:
+            try:
+                data = s.recv(65535)
+            except socket.timeout:
+                break
+            m = n.decode_message(data)
+            if m is None:
+                continue
+            got[m["id"]] = m
+        assert len(got) >= int(total * 0.99), \
+            f"lost replies: {total - len(got)} of {total}"
+        for m in got.values():
+            assert m["rcode"] == "NOERROR"
+            assert len(m["answers"]) == members, len(m["answers"])
+        s.close()
+    finally:
+        bal.terminate()
+        bal.wait(timeout=5)
+        b.stop()
