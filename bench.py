@@ -9,13 +9,18 @@ Topology per run (all on one node; "GPU" count N maps to N binderd
 server processes, the reference's only parallelism axis — SURVEY.md
 §2.3: replicated processes behind one balancer, no sharding):
 
-    StubZk (10k records) <-- N x binderd (zk mirror) <-- binder-balancer
-                                                            ^
-                dnsblast (C++ load generator, 4N source IPs)+
+    zkd (native registry, 10k records)
+      <-- N x binderd (zk mirror) <-- binder-balancer
+                                         ^
+         dnsblast daemon (C++ load generator, GSO bursts, paced)
 
-A "step" is a fixed batch of Q = 200k*N queries (per-process work fixed
-as N grows => weak scaling). Queries are a uniform A+SRV mix over the
-tree. Note: this workload has no tensor compute — the reference is a
+Protocol: fixed-rate qps@SLO — untimed calibration binary-searches the
+highest offered rate sustaining p99 <= 2 ms with zero timeouts and
+>=97% delivery (2-of-3 probes per level + margin), then the timed
+steps run at that rate on a persistent generator, so a "step" is a
+fixed batch of Q = 1M*N queries of pure query work (weak scaling).
+Queries are a uniform A+SRV mix over the tree; replies are verified
+NOERROR (noerror_frac reported). Note: this workload has no tensor compute — the reference is a
 Node.js DNS server (BASELINE.json "north_star" records the tier
 mismatch); torch is used only for the multi-rank barrier contract.
 """
