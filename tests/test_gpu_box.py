@@ -88,3 +88,46 @@ def test_bench_entrypoint_quick():
     line = json.loads(out.stdout.strip().splitlines()[-1])
     assert line["metric"] == "dns_queries_per_sec"
     assert line["value"] > 1000
+
+
+def test_native_zkd_registry_chain(tmp_path):
+    """The native registry on the real box: zkd-backed mirror serves,
+    survives a zkd hard-kill + restart on the same data, and new
+    writes propagate (exercises bin/zkd in the recorded gpu tier)."""
+    from binder_amd.harness import NativeZkd
+    from binder_amd.zkclient import ZkConn
+
+    d = tmp_path / "data"
+    z = NativeZkd(data_dir=str(d)).start()
+    port = z.port
+    b = None
+    try:
+        c = ZkConn("127.0.0.1", z.port)
+        c.mkdirp("/com/foo")
+        c.create("/com/foo/web", json.dumps(
+            {"type": "host", "host": {"address": "10.0.0.9"}}).encode())
+        c.close()
+        b = BinderProcess(dns_domain="foo.com", store="zk",
+                          zk_host="127.0.0.1", zk_port=z.port,
+                          workdir=tmp_path)
+        b.start()
+        r = b.wait_ready("web.foo.com", timeout=30)
+        assert r.answers[0]["address"] == "10.0.0.9"
+
+        z.proc.kill()
+        z.proc.wait()
+        assert b.dig("web.foo.com").answers[0]["address"] == "10.0.0.9"
+
+        z = NativeZkd(port=port, data_dir=str(d)).start()
+        assert z.nodes_restored == 3
+        c = ZkConn("127.0.0.1", z.port)
+        c.create("/com/foo/neu", json.dumps(
+            {"type": "host",
+             "host": {"address": "10.0.0.10"}}).encode())
+        r = b.wait_ready("neu.foo.com", timeout=40)
+        assert r.answers[0]["address"] == "10.0.0.10"
+        c.close()
+    finally:
+        if b:
+            b.stop()
+        z.stop()
