@@ -62,3 +62,38 @@ def test_bench_under_torchrun_gloo_world2():
     assert d["n_gpus"] == 2
     assert d["config"]["queries_per_step"] == 30000
     assert d["value"] > 0
+
+
+def test_calibration_converges_and_tolerates_flakes():
+    """calibrate_rate must converge near the true SLO threshold and a
+    single flaky probe per level must not collapse the search (the
+    2-of-3 tiebreak)."""
+    import random
+
+    import bench
+
+    true_limit = 3_000_000
+    rng = random.Random(42)
+
+    class FakeBlast:
+        def __init__(self, flake_prob=0.0):
+            self.flake_prob = flake_prob
+
+        def step(self, queries, rate=0):
+            ok = rate <= true_limit
+            if ok and rng.random() < self.flake_prob:
+                ok = False  # transient stall lands in p99
+            return {"qps": rate if ok else rate * 0.9,
+                    "p99_us": 500 if ok else 9000,
+                    "timeouts": 0, "received": queries,
+                    "noerror": queries}
+
+    # clean probes: land within [85%, 100%] of the true limit
+    r = bench.calibrate_rate(FakeBlast(), capacity=4_000_000,
+                             slo_us=2000)
+    assert 0.85 * true_limit <= r <= true_limit, r
+
+    # 20% flake probability: still within [70%, 100%]
+    r = bench.calibrate_rate(FakeBlast(flake_prob=0.2),
+                             capacity=4_000_000, slo_us=2000)
+    assert 0.70 * true_limit <= r <= true_limit, r
