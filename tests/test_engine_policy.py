@@ -438,3 +438,16 @@ def test_host_non_ipv4_address_empty_noerror():
         r = e.query(name, "A")
         assert r["rcode"] == "NOERROR", name
         assert r["answers"] == [], name
+
+
+def test_resolver_typed_record_not_served_as_host(eng):
+    """The typed resolver-registry records ({"type":"resolver",...},
+    recursion.hpp schema) are registry data, not DNS answers: an A
+    query for one must get the unknown-type treatment (NOERROR, no
+    answers — server.js:419-424), never the resolver's address."""
+    put(eng, "r1.bar.foo.com", {
+        "type": "resolver",
+        "resolver": {"datacenter": "dc2", "address": "10.9.9.9"}})
+    r = eng.query("r1.bar.foo.com", "A")
+    assert r["rcode"] == "NOERROR"
+    assert r["answers"] == []
