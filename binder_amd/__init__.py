@@ -11,7 +11,7 @@ reference is a Node.js event loop (no GPU / tensor work exists there —
 BASELINE.json "north_star"); the rebuild is native C++ end to end.
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
 
 from pathlib import Path
 

@@ -39,7 +39,7 @@
 
 using namespace bamd;
 
-static const char* kVersion = "binder-amd 0.1.0";
+static const char* kVersion = "binder-amd 0.2.0";
 
 static void usage(const char* name) {
     fprintf(stderr,
