@@ -303,3 +303,86 @@ def test_client_handles_golden_expiry_response(tmp_path):
     finally:
         proc.kill()
         srv.close()
+
+
+def test_server_exists_getchildren_and_error_codes(zk):
+    """Remaining spec vectors: exists (op 3, Stat-only response),
+    plain getChildren (op 8, no trailing Stat), ZNODEEXISTS (-110) on
+    duplicate create, ZNONODE (-101) on exists/delete of a missing
+    node, ZNOTEMPTY (-111) on deleting a parent."""
+    s, *_ = connect_raw(zk)
+    try:
+        s.sendall(req(1, 1, jstr("/p") + jstr(b"d") + ACL_OPEN +
+                      be32(0)))
+        assert parse_reply_header(read_packet(s))[2] == 0
+        s.sendall(req(2, 1, jstr("/p/k") + jstr(b"") + ACL_OPEN +
+                      be32(0)))
+        assert parse_reply_header(read_packet(s))[2] == 0
+
+        # exists: ReplyHeader + bare Stat (68 bytes, no data buffer)
+        s.sendall(req(3, 3, jstr("/p") + b"\x00"))
+        xid, zxid, err, rest = parse_reply_header(read_packet(s))
+        assert (xid, err) == (3, 0)
+        stat, tail = parse_stat(rest)
+        assert tail == b""
+        assert stat["numChildren"] == 1
+        assert stat["dataLength"] == 1
+
+        # exists on a missing node => ZNONODE, empty body
+        s.sendall(req(4, 3, jstr("/nope") + b"\x00"))
+        xid, zxid, err, rest = parse_reply_header(read_packet(s))
+        assert (xid, err) == (4, -101)
+        assert rest == b""
+
+        # plain getChildren (op 8): vector<string> only, NO Stat
+        s.sendall(req(5, 8, jstr("/p") + b"\x00"))
+        xid, zxid, err, rest = parse_reply_header(read_packet(s))
+        assert (xid, err) == (5, 0)
+        (count,) = struct.unpack(">i", rest[:4])
+        assert count == 1
+        (klen,) = struct.unpack(">i", rest[4:8])
+        assert rest[8:8 + klen] == b"k"
+        assert rest[8 + klen:] == b"", "op 8 must not append a Stat"
+
+        # duplicate create => ZNODEEXISTS
+        s.sendall(req(6, 1, jstr("/p") + jstr(b"x") + ACL_OPEN +
+                      be32(0)))
+        assert parse_reply_header(read_packet(s))[2] == -110
+
+        # delete non-empty parent => ZNOTEMPTY
+        s.sendall(req(7, 2, jstr("/p") + be32(-1)))
+        assert parse_reply_header(read_packet(s))[2] == -111
+
+        # delete missing => ZNONODE
+        s.sendall(req(8, 2, jstr("/gone") + be32(-1)))
+        assert parse_reply_header(read_packet(s))[2] == -101
+    finally:
+        s.close()
+
+
+def test_server_child_watch_fires_on_create(zk):
+    """getChildren with watch=1: a subsequent child create must push
+    ReplyHeader{xid=-1} + WatcherEvent{NodeChildrenChanged=4,
+    SyncConnected=3, parent path}."""
+    a, *_ = connect_raw(zk)
+    b, *_ = connect_raw(zk)
+    try:
+        a.sendall(req(1, 1, jstr("/cw") + jstr(b"") + ACL_OPEN +
+                      be32(0)))
+        read_packet(a)
+        a.sendall(req(2, 12, jstr("/cw") + b"\x01"))  # watch
+        read_packet(a)
+        b.sendall(req(1, 1, jstr("/cw/kid") + jstr(b"") + ACL_OPEN +
+                      be32(0)))
+        read_packet(b)
+        body = read_packet(a)
+        xid, zxid, err, rest = parse_reply_header(body)
+        assert (xid, err) == (-1, 0)
+        ev_type, ev_state = struct.unpack(">ii", rest[:8])
+        (plen,) = struct.unpack(">i", rest[8:12])
+        assert ev_type == 4       # NodeChildrenChanged
+        assert ev_state == 3      # SyncConnected
+        assert rest[12:12 + plen] == b"/cw"
+    finally:
+        a.close()
+        b.close()
