@@ -8,6 +8,7 @@ unlink (SIGTERM), and the stats endpoint (balstat replacement).
 import json
 import os
 import socket
+import struct
 import subprocess
 import time
 from pathlib import Path
@@ -461,3 +462,46 @@ def test_large_reply_runs_through_balancer(tmp_path):
         bal.terminate()
         bal.wait(timeout=5)
         b.stop()
+
+
+def test_gso_coalesced_ingress_with_garbage_segment(cluster):
+    """A client can ship several datagrams as one UDP_SEGMENT
+    super-packet; the balancer (UDP_GRO) must split and forward each
+    segment as its own query — including surviving a garbage segment,
+    which binderd simply drops as malformed."""
+    from binder_amd import require_native
+    n = require_native()
+    wires = []
+    for i in range(3):
+        w = bytearray(n.encode_message(
+            {"id": 700 + i,
+             "questions": [{"name": "web.foo.com", "type": "A"}]}))
+        wires.append(bytes(w))
+    seg = len(wires[0])
+    assert all(len(w) == seg for w in wires)
+    # 3 valid queries + one garbage segment of the same size
+    blob = b"".join(wires) + b"\xde\xad" * (seg // 2) + \
+        (b"\x00" if seg % 2 else b"")
+    s = socket.socket(socket.AF_INET, socket.SOCK_DGRAM)
+    s.settimeout(3)
+    s.connect(("127.0.0.1", cluster["port"]))
+    UDP_SEGMENT = 103
+    try:
+        s.sendmsg([blob], [(socket.IPPROTO_UDP, UDP_SEGMENT,
+                            struct.pack("H", seg))])
+    except OSError:
+        pytest.skip("kernel without UDP_SEGMENT")
+    got = set()
+    deadline = time.time() + 3
+    while len(got) < 3 and time.time() < deadline:
+        try:
+            data = s.recv(4096)
+        except socket.timeout:
+            break
+        m = n.decode_message(data)
+        if m and m["rcode"] == "NOERROR":
+            got.add(m["id"])
+    assert got == {700, 701, 702}, got
+    # chain still healthy
+    assert dig("web.foo.com", port=cluster["port"]).status == "NOERROR"
+    s.close()
