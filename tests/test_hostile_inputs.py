@@ -246,3 +246,59 @@ def test_balancer_sockdir_junk_and_disappearance(balancer, tmp_path):
             pass
     else:
         pytest.fail("balancer never resumed after dir returned")
+
+
+def test_balancer_rogue_backend_garbage_frames(balancer, tmp_path):
+    """A 'backend' that accepts the balancer's connection and then
+    speaks garbage (bad magic, oversized lengths, random bytes) must
+    be dropped as a bad peer while service continues via the healthy
+    backend."""
+    import threading
+
+    rogue_path = balancer["sockdir"] / "b7"
+    lsock = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+    lsock.bind(str(rogue_path))
+    lsock.listen(4)
+    stop = threading.Event()
+
+    def rogue():
+        lsock.settimeout(0.3)
+        payloads = [b"\xff\xff\xff\xff\xff\xff" * 10,
+                    b"\xb5\x02\xff\xff\xff\x7f" + b"x" * 64,
+                    os.urandom(512)]
+        i = 0
+        while not stop.is_set():
+            try:
+                c, _ = lsock.accept()
+            except socket.timeout:
+                continue
+            try:
+                c.sendall(payloads[i % len(payloads)])
+                i += 1
+                time.sleep(0.1)
+                c.close()
+            except OSError:
+                pass
+
+    t = threading.Thread(target=rogue, daemon=True)
+    t.start()
+    try:
+        # give the balancer a few rescan cycles to meet the rogue
+        deadline = time.time() + 4
+        ok = 0
+        while time.time() < deadline:
+            try:
+                r = dig("web.foo.com", "A", server="127.0.0.1",
+                        port=balancer["port"], timeout=2)
+                if r.status == "NOERROR":
+                    ok += 1
+            except OSError:
+                pass
+            time.sleep(0.1)
+        assert ok >= 20, f"service degraded next to rogue backend ({ok})"
+        assert balancer["bal"].poll() is None
+    finally:
+        stop.set()
+        t.join()
+        lsock.close()
+        rogue_path.unlink(missing_ok=True)
