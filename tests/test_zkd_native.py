@@ -115,6 +115,12 @@ def test_compaction_bounds_log(tmp_path):
         c.close()
     finally:
         z3.stop()
+    # the compacted log remains valid FileTxnLog v2 for zklogcat
+    out = subprocess.run([str(ZKLOGCAT), str(d / "log.1")],
+                         capture_output=True, text=True, check=True)
+    txns = [json.loads(line) for line in out.stdout.splitlines()]
+    assert any(t["type"] == "create" and t["path"] == "/hot"
+               for t in txns)
 
 
 def test_session_timeout_reaps_ephemerals_and_fires_watches():
