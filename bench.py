@@ -48,11 +48,20 @@ def effective_cpus():
     os.cpu_count() oversubscribes the quota and CFS throttling puts
     10-100 ms whole-group freezes straight into p99 (measured,
     profiles/SCALING.md round 2)."""
-    try:
+    try:  # cgroup v2
         parts = Path("/sys/fs/cgroup/cpu.max").read_text().split()
         if parts[0] != "max":
             return max(2, int(parts[0]) // int(parts[1]))
     except (OSError, ValueError, IndexError):
+        pass
+    try:  # cgroup v1
+        q = int(Path("/sys/fs/cgroup/cpu/cpu.cfs_quota_us")
+                .read_text())
+        p = int(Path("/sys/fs/cgroup/cpu/cpu.cfs_period_us")
+                .read_text())
+        if q > 0 and p > 0:
+            return max(2, q // p)
+    except (OSError, ValueError):
         pass
     return os.cpu_count() or 8
 
